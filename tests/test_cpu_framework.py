@@ -1,0 +1,220 @@
+"""CPU plumbing tests: model contract, optimizer, recorder, checkpoints,
+conv-plan table construction.  (Multi-process rule tests live in
+test_rules_cpu.py; GPU kernel numerics in test_kernels_gpu.py.)"""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from theanompi_amd.lib.exchanger_strategy import flatten_params
+from theanompi_amd.lib.opt import ParamRec, SGDMomentum
+from theanompi_amd.lib.recorder import Recorder
+from theanompi_amd.lib import helper_funcs
+from theanompi_amd.models.layers import ConvPlan, PIECE
+from theanompi_amd.models.test_model import TestModel
+from theanompi_amd.ops.functional import cpad_for
+
+
+def make_model(**kw):
+    cfg = {"rank": 0, "size": 1, "device": "cpu", "verbose": False}
+    cfg.update(kw)
+    m = TestModel(cfg)
+    m.compile_iter_fns(kw.get("sync_type", "cdd"))
+    return m
+
+
+def test_cpad_for():
+    assert cpad_for(3) == 4
+    assert cpad_for(16) == 16
+    assert cpad_for(17) == 32
+    assert cpad_for(48) == 48
+    assert cpad_for(96) == 96
+
+
+def test_model_contract_surface():
+    m = make_model()
+    for attr in ("params", "vels", "data", "batch_size"):
+        assert hasattr(m, attr)
+    for meth in ("compile_iter_fns", "train_iter", "val_iter",
+                 "adjust_hyperp", "load", "save", "reset_iter"):
+        assert callable(getattr(m, meth))
+    assert len(m.params) == len(m.vels) > 0
+
+
+def test_train_loss_decreases():
+    torch.manual_seed(0)
+    m = make_model(sync_type="grad")
+    losses = []
+    for i in range(30):
+        loss, err = m.train_iter(i % m.data.n_batch_train)
+        m.apply_update()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0], losses
+
+
+def test_cdd_equals_plain_sgd():
+    """cdd phase1+phase2 on one worker == full fused step (reference
+    semantics: with N=1 the exchanged vels are the local vels)."""
+    torch.manual_seed(1)
+    m1 = make_model(sync_type="cdd")
+    torch.manual_seed(1)
+    m2 = make_model(sync_type="grad")
+    for p1, p2 in zip(m1.params, m2.params):
+        assert torch.allclose(p1, p2)
+    for i in range(3):
+        m1.train_iter(i)
+        m1.apply_update()          # phase2 applies vels
+        m2.train_iter(i)
+        m2.apply_update()          # full step from grads
+    for p1, p2 in zip(m1.params, m2.params):
+        assert torch.allclose(p1, p2, atol=1e-6), (p1 - p2).abs().max()
+
+
+def test_val_iter():
+    m = make_model()
+    cost, top1, top5 = m.val_iter(0)
+    assert np.isfinite(cost)
+    assert 0.0 <= top1 <= 1.0 and 0.0 <= top5 <= top1 + 1e-9
+
+
+def test_adjust_hyperp():
+    m = make_model()
+    lr0 = m.opt.lr
+    m.adjust_hyperp(1)
+    assert m.opt.lr == pytest.approx(lr0 * 0.1)
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    torch.manual_seed(2)
+    m = make_model(sync_type="grad")
+    for i in range(2):
+        m.train_iter(i)
+        m.apply_update()
+    m.save(str(tmp_path), epoch=1)
+    # files follow the reference naming pattern
+    assert (tmp_path / "W_0_1.npy").exists()
+    assert (tmp_path / "V_0_1.npy").exists()
+    ref = [p.detach().clone() for p in m.params]
+    refv = [v.clone() for v in m.vels]
+    torch.manual_seed(99)
+    m2 = make_model(sync_type="grad")
+    m2.load(str(tmp_path), epoch=1)
+    for a, b in zip(ref, m2.params):
+        assert torch.allclose(a, b, atol=1e-6)
+    for a, b in zip(refv, m2.vels):
+        assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_recorder_roundtrip(tmp_path):
+    r = Recorder({"rank": 0, "verbose": False})
+    r.start()
+    r.end("calc")
+    r.train_error(1, 2.0, 0.5)
+    r.print_train_info(0, force=True)
+    r.val_error(1, 1.5, 0.4, 0.1)
+    p = tmp_path / "inforec.pkl"
+    r.save(str(p))
+    r2 = Recorder({"verbose": False}).load(str(p))
+    assert r2.train_info == r.train_info
+    assert r2.val_info == r.val_info
+
+
+def test_flatten_params():
+    ts = [torch.randn(7), torch.randn(3, 5), torch.randn(11)]
+    flat, views = flatten_params(ts, bucket_bytes=1 << 30)
+    assert len(flat) == 1
+    for v, t in zip(views, ts):
+        v.copy_(t)
+    off = 0
+    for t in ts:
+        assert torch.equal(flat[0][off:off + t.numel()], t.reshape(-1))
+        off += t.numel()
+
+
+def test_flatten_params_bucketing():
+    ts = [torch.randn(1000) for _ in range(10)]
+    flat, views = flatten_params(ts, bucket_bytes=4 * 2048)
+    assert len(flat) > 1
+    assert sum(f.numel() for f in flat) == 10 * 1000
+
+
+def test_sgd_momentum_semantics():
+    w = torch.zeros(4)
+    w.grad = torch.ones(4)
+    rec = ParamRec(w)
+    opt = SGDMomentum([rec], lr=0.1, momentum=0.9, weight_decay=0.0)
+    opt.step()
+    assert torch.allclose(w, torch.full((4,), -0.1))
+    w.grad = torch.ones(4)
+    opt.step()
+    # v = 0.9*(-0.1) - 0.1 = -0.19 ; w = -0.1 - 0.19 = -0.29
+    assert torch.allclose(w, torch.full((4,), -0.29))
+
+
+# ---------------------------------------------------------------------------
+# conv plan geometry (pure host logic -> CPU-testable)
+# ---------------------------------------------------------------------------
+
+def test_convplan_table_dense():
+    # 3x3 conv, C=32 (one combined span per r of length 96)
+    plan = ConvPlan(2, 8, 8, 32, 64, 3, 3, (1, 1), (1, 1), 1, "cpu")
+    assert plan.Kdf == 3 * 96
+    tab = plan.table.numpy()
+    # piece j inside row r covers elements r*Wp*C + j16
+    Wp = 10
+    for j, off in enumerate(tab):
+        r = (j * PIECE) // 96
+        pos = (j * PIECE) % 96
+        assert off == r * Wp * 32 + pos
+
+
+def test_convplan_table_stem():
+    # 11x11 stride-4 stem, C=4: runs of 44, padded to 48
+    plan = ConvPlan(1, 227, 227, 4, 96, 11, 11, (4, 4), (0, 0), 1, "cpu")
+    assert plan.Kdf == 11 * 48
+    tab = plan.table.numpy()
+    Wp = 227
+    for j, off in enumerate(tab):
+        r = (j * PIECE) // 48
+        pos = (j * PIECE) % 48
+        if pos < 44:
+            assert off == r * Wp * 4 + pos
+        else:
+            assert off == r * Wp * 4  # padding piece points at span start
+
+
+def test_convplan_grouped():
+    # grouped conv: spans per (r, s) of length Cg
+    plan = ConvPlan(1, 27, 27, 96, 256, 5, 5, (1, 1), (2, 2), 2, "cpu")
+    assert plan.Cg == 48
+    assert plan.Kdf == 5 * 5 * 48
+    assert plan.Kgp == 128
+    assert plan.Kphys == 256
+
+
+def test_convplan_dgrad_classes():
+    # stride-2 conv -> 4 dgrad classes
+    plan = ConvPlan(1, 8, 8, 16, 16, 3, 3, (2, 2), (1, 1), 1, "cpu")
+    classes = [p for p in plan.dgrad_plans if p is not None]
+    assert len(classes) == 4
+    for p in classes:
+        assert p["Kdf2"] % PIECE == 0
+
+
+def test_convplan_flat_roundtrip():
+    plan = ConvPlan(1, 8, 8, 16, 32, 3, 3, (1, 1), (1, 1), 1, "cpu")
+    w = torch.randn(32, 16, 3, 3)
+    flat = plan.w_to_flat(w)
+    back = plan.flat_to_w(flat)
+    assert torch.allclose(w, back)
+
+
+def test_convplan_flat_roundtrip_grouped():
+    plan = ConvPlan(1, 8, 8, 32, 24, 3, 3, (1, 1), (1, 1), 2, "cpu")
+    w = torch.randn(24, 16, 3, 3)     # Kg=12 per group, Cg=16
+    flat = plan.w_to_flat(w)
+    assert flat.shape == (2, plan.Kdf, plan.Kgp)
+    back = plan.flat_to_w(flat)
+    assert torch.allclose(w, back)

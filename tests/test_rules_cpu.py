@@ -1,0 +1,117 @@
+"""Multi-process rule tests on CPU (gloo, world_size 2-3) — the
+reference's BASELINE config #1 plumbing mode.  Each test launches real
+worker processes through the rule API, exactly as a user session script
+would."""
+
+import os
+import subprocess
+import sys
+import textwrap
+
+import numpy as np
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _run_session(tmp_path, body, env_extra=None, timeout=240):
+    script = tmp_path / "session.py"
+    script.write_text(textwrap.dedent(body))
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    env.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    env.setdefault("MASTER_ADDR", "127.0.0.1")
+    if env_extra:
+        env.update(env_extra)
+    res = subprocess.run([sys.executable, str(script)], env=env,
+                         capture_output=True, text=True, timeout=timeout)
+    assert res.returncode == 0, \
+        f"stdout:\n{res.stdout[-4000:]}\nstderr:\n{res.stderr[-4000:]}"
+    return res
+
+
+def test_bsp_two_workers_cpu(tmp_path):
+    """2-worker BSP on CPU completes; workers end bit-identical (the
+    deterministic grad-sum check of SURVEY.md §7 phase 0)."""
+    out = str(tmp_path)
+    body = f"""
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import BSP
+    rule = BSP()
+    rule.init(devices=['cpu', 'cpu'],
+              modelfile='theanompi_amd.models.test_model',
+              modelclass='TestModel',
+              epochs=1, sync_type='cdd', exchange_strategy='ar',
+              dump_dir={out!r})
+    codes = rule.wait()
+    assert codes == [0, 0], codes
+    """
+    # patch: workers dump their params when THEANOMPI_AMD_OPT_DUMP_DIR set
+    _run_session(tmp_path, body)
+    a = np.load(os.path.join(out, "params_rank0.npz"))
+    b = np.load(os.path.join(out, "params_rank1.npz"))
+    assert set(a.files) == set(b.files)
+    for k in a.files:
+        np.testing.assert_allclose(a[k], b[k], atol=1e-6)
+
+
+@pytest.mark.parametrize("strategy", ["ar", "asa"])
+def test_bsp_strategies_cpu(tmp_path, strategy):
+    body = f"""
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import BSP
+    rule = BSP()
+    rule.init(devices=['cpu', 'cpu'],
+              modelfile='theanompi_amd.models.test_model',
+              modelclass='TestModel',
+              epochs=1, sync_type='grad', exchange_strategy={strategy!r})
+    assert rule.wait() == [0, 0]
+    """
+    _run_session(tmp_path, body)
+
+
+def test_bsp_avg_mode_cpu(tmp_path):
+    body = """
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import BSP
+    rule = BSP()
+    rule.init(devices=['cpu', 'cpu'],
+              modelfile='theanompi_amd.models.test_model',
+              modelclass='TestModel',
+              epochs=1, sync_type='avg', exchange_strategy='ar')
+    assert rule.wait() == [0, 0]
+    """
+    _run_session(tmp_path, body)
+
+
+def test_easgd_cpu(tmp_path):
+    body = """
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import EASGD
+    rule = EASGD()
+    rule.init(devices=['cpu', 'cpu', 'cpu'],
+              modelfile='theanompi_amd.models.test_model',
+              modelclass='TestModel',
+              tau=2, alpha=0.5, n_total_iters=24)
+    assert rule.wait() == [0, 0, 0]
+    """
+    _run_session(tmp_path, body)
+
+
+def test_gosgd_cpu(tmp_path):
+    body = """
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import GOSGD
+    rule = GOSGD()
+    rule.init(devices=['cpu', 'cpu'],
+              modelfile='theanompi_amd.models.test_model',
+              modelclass='TestModel',
+              epochs=1, p_push=0.5)
+    assert rule.wait() == [0, 0]
+    """
+    _run_session(tmp_path, body)

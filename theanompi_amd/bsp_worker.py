@@ -1,0 +1,91 @@
+"""BSP worker — rebuild of ``theanompi/bsp_worker.py`` [pub]
+(SURVEY.md §2.2, §3.2).  One process per GPU; per iteration:
+train_iter (fwd/bwd + cdd phase 1) -> exchanger.exchange (RCCL over
+xGMI) -> apply_update (cdd phase 2) -> periodic recorder print; per
+epoch: sharded validation with all-reduced stats, adjust_hyperp,
+checkpoint from rank 0.
+"""
+
+import os
+
+import torch
+import torch.distributed as dist
+
+from theanompi_amd.lib.base import GPUProcess, _env_opt
+from theanompi_amd.lib.exchanger import BSP_Exchanger
+from theanompi_amd.lib.recorder import Recorder
+
+
+def run():
+    proc = GPUProcess()
+    device = proc.init_device()
+    proc.init_comm()
+    model = proc.build_model()
+
+    sync_type = _env_opt("sync_type", "cdd")
+    strategy = _env_opt("exchange_strategy",
+                        "rccl_bf16" if proc.on_gpu else "ar")
+    n_epochs = int(_env_opt("epochs", model.n_epochs))
+    save_dir = _env_opt("save_dir", "")
+    save_freq = int(_env_opt("save_freq", 0))
+
+    model.compile_iter_fns(sync_type)
+    config = {"sync_type": sync_type, "exchange_strategy": strategy,
+              "size": proc.size, "rank": proc.rank,
+              "bucket_mb": float(_env_opt("bucket_mb", 64))}
+    exchanger = BSP_Exchanger(config, model.params, model.vels)
+
+    proc.broadcast_params(model)
+    rec = Recorder({"rank": proc.rank, "size": proc.size,
+                    "verbose": proc.verbose,
+                    "print_freq": int(_env_opt("print_freq", 40)),
+                    "sync_fn": (torch.cuda.synchronize if proc.on_gpu
+                                else None)})
+    proc.barrier()
+
+    for epoch in range(n_epochs):
+        rec.start_epoch()
+        n_batches = model.data.n_batch_train
+        for i in range(n_batches):
+            rec.start()
+            model.train_iter(i, rec)
+            if sync_type == "grad":
+                exchanger.set_grad_buffers(model.opt.grads())
+            exchanger.exchange(rec)
+            rec.start()
+            model.apply_update()
+            rec.end("calc")
+            rec.print_train_info(i)
+        rec.print_train_info(n_batches - 1, force=True)
+
+        # sharded validation, stats all-reduced over the control group
+        stats = torch.zeros(4)
+        for j in range(proc.rank, model.data.n_batch_val, proc.size):
+            cost, top1, top5 = model.val_iter(j, rec)
+            stats += torch.tensor([cost, top1, top5, 1.0])
+        if dist.is_initialized():
+            dist.all_reduce(stats, group=proc.ctrl)
+        if stats[3] > 0:
+            rec.val_error(rec.uidx, *(stats[:3] / stats[3]).tolist())
+
+        model.adjust_hyperp(epoch + 1)
+        rec.end_epoch()
+        if save_dir and save_freq and proc.rank == 0 \
+                and (epoch + 1) % save_freq == 0:
+            model.save(save_dir, epoch + 1)
+            rec.save(os.path.join(save_dir, "inforec.pkl"))
+        proc.barrier()
+
+    dump_dir = _env_opt("dump_dir", "")
+    if dump_dir:    # test hook: dump final params per rank
+        import numpy as np
+        arrs = {f"p{i}": p.detach().float().cpu().numpy()
+                for i, p in enumerate(model.params)}
+        np.savez(os.path.join(dump_dir, f"params_rank{proc.rank}.npz"),
+                 **arrs)
+    proc.finalize()
+    return model, rec
+
+
+if __name__ == "__main__":
+    run()

@@ -1,0 +1,108 @@
+"""GOSGD worker — rebuild of ``theanompi/gosgd_worker.py`` [pub]
+(Blot et al. GoSGD): local SGD; each iteration with
+probability p push (w_i, alpha_i/2) to a uniformly drawn peer; merge any
+received pushes  w <- (a*w + a_j*w_j)/(a + a_j),  a += a_j.
+
+Transport: pushes are announced on the TCP control store (non-blocking
+``check`` — gloo cannot poll a pending irecv in this torch build) and the
+payload travels as a matched gloo isend/recv pair from pinned host
+staging, so a push never blocks the sender's GPU stream (reference used
+MPI Isend/Irecv + probe, C5).
+"""
+
+import random
+import time
+
+import torch
+import torch.distributed as dist
+
+from theanompi_amd.lib.base import GPUProcess, _env_opt
+from theanompi_amd.lib.exchanger import GOSGD_Exchanger
+from theanompi_amd.lib.recorder import Recorder
+
+
+def run():
+    proc = GPUProcess()
+    proc.init_device()
+    proc.init_comm()
+    model = proc.build_model()
+    model.compile_iter_fns("avg")
+
+    p_push = float(_env_opt("p_push", 0.05))
+    n_epochs = int(_env_opt("epochs", model.n_epochs))
+    rng = random.Random(1234 + proc.rank)
+
+    proc.broadcast_params(model)
+    exch = GOSGD_Exchanger({}, model.params, group=proc.ctrl)
+    rec = Recorder({"rank": proc.rank, "size": proc.size,
+                    "verbose": proc.verbose,
+                    "print_freq": int(_env_opt("print_freq", 40)),
+                    "sync_fn": (torch.cuda.synchronize if proc.on_gpu
+                                else None)})
+    store = proc.control_store()
+    peers = [r for r in range(proc.size) if r != proc.rank]
+    rbufs = {src: exch.recv_buffer() for src in peers}
+    send_seq = {dst: 0 for dst in peers}    # pushes I sent to dst
+    recv_seq = {src: 0 for src in peers}    # pushes I consumed from src
+    me = proc.rank
+    proc.barrier()
+
+    def drain():
+        """Receive + merge every announced, not-yet-consumed push."""
+        merged = False
+        for src in peers:
+            while store.check([f"go/push/{me}/{src}/{recv_seq[src]}"]):
+                dist.recv(rbufs[src], src=src, group=proc.ctrl,
+                          tag=400 + src)
+                exch.try_merge_buf(rbufs[src])
+                recv_seq[src] += 1
+                merged = True
+        if merged:
+            model.refresh_all()
+
+    uidx = 0
+    for epoch in range(n_epochs):
+        rec.start_epoch()
+        for i in range(model.data.n_batch_train):
+            rec.start()
+            model.train_iter(i, rec)
+            model.apply_update()
+            rec.end("calc")
+            rec.start()
+            drain()
+            if peers and rng.random() < p_push:
+                dst = rng.choice(peers)
+                exch.push(dst, tag=400 + me)       # isend (non-blocking)
+                store.set(f"go/push/{dst}/{me}/{send_seq[dst]}", "1")
+                send_seq[dst] += 1
+            rec.end("comm")
+            uidx += 1
+            rec.print_train_info(uidx)
+        model.adjust_hyperp(epoch + 1)
+        rec.end_epoch()
+
+    # shutdown: finish my last send (receivers keep draining until every
+    # rank has announced done), then drain stragglers until all done.
+    if exch._pending_send is not None:
+        exch._pending_send.wait()
+    store.add("go/done", 1)
+    while int(store.add("go/done", 0)) < proc.size:
+        drain()
+        time.sleep(0.001)
+    drain()                                   # announced-but-unreceived
+    proc.barrier()
+
+    stats = torch.zeros(3)
+    n = 0
+    for j in range(proc.rank, model.data.n_batch_val, proc.size):
+        c, t1, t5 = model.val_iter(j, rec)
+        stats += torch.tensor([c, t1, t5])
+        n += 1
+    if proc.verbose and n:
+        rec.val_error(uidx, *(stats / n).tolist())
+    proc.finalize()
+    return model, rec
+
+
+if __name__ == "__main__":
+    run()

@@ -1,0 +1,113 @@
+"""Optimizer builders — rebuild of ``theanompi/lib/opt.py`` [pub]
+(SURVEY.md §2.3): SGD + momentum (+ weight decay), with the cdd
+split-phase formulation around the BSP comm point:
+
+  phase 1 (pre-exchange):  vel = mu*vel - lr*(grad + wd*w)
+  exchange:                vel <- mean over workers (BSP_Exchanger)
+  phase 2 (post-exchange): w  += vel
+
+'avg'/'grad' modes use the fused single step.  On GPU the update is one
+pass of the fused HIP kernel N12 over each flat fp32 master buffer,
+optionally writing the bf16 working copy in the same pass.
+"""
+
+import torch
+
+from theanompi_amd import ops as _ops
+
+
+class ParamRec:
+    """One trainable: fp32 master + vel (+ optional bf16 mirror)."""
+
+    __slots__ = ("master", "vel", "bf16_out", "name", "lr_mult", "wd_mult")
+
+    def __init__(self, master, name="", lr_mult=1.0, wd_mult=1.0,
+                 bf16_out=None):
+        self.master = master          # fp32 tensor with .grad after bwd
+        self.vel = torch.zeros_like(master.data)
+        self.bf16_out = bf16_out      # bf16 tensor refreshed by the step
+        self.name = name
+        self.lr_mult = lr_mult
+        self.wd_mult = wd_mult
+
+
+class SGDMomentum:
+    """Momentum SGD over ParamRecs, usable under every rule."""
+
+    def __init__(self, recs, lr=0.01, momentum=0.9, weight_decay=0.0):
+        self.recs = list(recs)
+        self.lr = lr
+        self.momentum = momentum
+        self.weight_decay = weight_decay
+        self._gpu = any(r.master.is_cuda for r in self.recs)
+
+    # -- helpers --------------------------------------------------------------
+
+    def zero_grad(self):
+        for r in self.recs:
+            if r.master.grad is not None:
+                r.master.grad = None
+
+    def grads(self):
+        out = []
+        for r in self.recs:
+            g = r.master.grad
+            if g is None:
+                g = torch.zeros_like(r.master.data)
+                r.master.grad = g
+            out.append(g)
+        return out
+
+    def vels(self):
+        return [r.vel for r in self.recs]
+
+    def _kernel_step(self, r, grad, phase):
+        C = _ops.ext()
+        out = r.bf16_out if (r.bf16_out is not None and phase != 1) \
+            else torch.empty(0, dtype=torch.bfloat16,
+                             device=r.master.device)
+        C.sgd_step(r.master.data.view(-1), grad.view(-1), r.vel.view(-1),
+                   out.view(-1),
+                   self.lr * r.lr_mult, self.momentum,
+                   self.weight_decay * r.wd_mult, int(phase))
+
+    def _torch_step(self, r, grad, phase):
+        lr = self.lr * r.lr_mult
+        wd = self.weight_decay * r.wd_mult
+        with torch.no_grad():
+            if phase in (0, 1):
+                g = grad
+                if wd:
+                    g = g.add(r.master.data, alpha=wd)
+                r.vel.mul_(self.momentum).add_(g, alpha=-lr)
+            if phase == 0:
+                r.master.data.add_(r.vel)
+            elif phase == 2:
+                r.master.data.add_(grad)   # grad holds the averaged vels
+            if r.bf16_out is not None and phase != 1:
+                r.bf16_out.copy_(r.master.data)
+
+    def _dispatch(self, r, grad, phase):
+        if r.master.is_cuda:
+            self._kernel_step(r, grad, phase)
+        else:
+            self._torch_step(r, grad, phase)
+
+    # -- public API ------------------------------------------
+
+    def step(self):
+        """Full fused step: v = mu*v - lr*(g + wd*w); w += v."""
+        for r in self.recs:
+            if r.master.grad is not None:
+                self._dispatch(r, r.master.grad, 0)
+
+    def phase1(self):
+        """cdd pre-exchange: compute velocities only."""
+        for r in self.recs:
+            if r.master.grad is not None:
+                self._dispatch(r, r.master.grad, 1)
+
+    def phase2(self):
+        """cdd post-exchange: apply the (exchanged, averaged) velocities."""
+        for r in self.recs:
+            self._dispatch(r, r.vel, 2)

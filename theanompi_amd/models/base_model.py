@@ -1,0 +1,218 @@
+"""Generic implementation of the model contract (SURVEY.md §1 L4->L2).
+
+Concrete zoo models (alex_net.py, googlenet.py, ...) subclass this and
+provide ``build_module()`` + hyperparameters; anything else satisfying
+the duck-type contract also works (the workers only use the contract).
+"""
+
+import torch
+
+from theanompi_amd.lib import helper_funcs
+from theanompi_amd.lib.opt import ParamRec, SGDMomentum
+from theanompi_amd.models import layers as L
+from theanompi_amd.ops import functional as OF
+
+
+class BaseModel:
+    """Implements: params, vels, data, compile_iter_fns, train_iter,
+    val_iter, adjust_hyperp, load, save, reset_iter."""
+
+    # subclass knobs
+    name = "base"
+    batch_size = 128
+    n_epochs = 60
+    lr = 0.01
+    momentum = 0.9
+    weight_decay = 5e-4
+    n_classes = 1000
+
+    def __init__(self, config):
+        self.config = dict(config)
+        self.rank = int(config.get("rank", 0))
+        self.size = int(config.get("size", 1))
+        self.device = torch.device(config.get("device", "cpu"))
+        self.gpu = self.device.type == "cuda"
+        self.verbose = bool(config.get("verbose", self.rank == 0))
+        if "batch_size" in config:
+            self.batch_size = int(config["batch_size"])
+        if "epochs" in config:
+            self.n_epochs = int(config["epochs"])
+        self.current_lr = self.lr
+        self.epoch = 0
+        self.sync_type = "cdd"
+
+        self.module = self.build_module()       # returns an nn.Module
+        self.module.train()
+        self.data = self.build_data()
+        self._build_recs()
+        self.opt = None
+
+    # -- to be provided by subclasses -----------------------------------------
+
+    def build_module(self):
+        raise NotImplementedError
+
+    def build_data(self):
+        raise NotImplementedError
+
+    def forward(self, x):
+        """Return logits [B, n_classes(+pad)]."""
+        return self.module(x)
+
+    # -- parameter bookkeeping -------------------------------------------------
+
+    def _build_recs(self):
+        self.recs = []
+        self._post_step_layers = []
+        for name, mod in self.module.named_modules():
+            if isinstance(mod, (L.Conv2d, L.Linear)):
+                bf16 = getattr(mod, "w_bf16", None)
+                self.recs.append(ParamRec(mod.weight, name + ".W",
+                                          bf16_out=bf16))
+                self.recs.append(ParamRec(mod.bias, name + ".b",
+                                          wd_mult=0.0))
+                if mod.gpu:
+                    self._post_step_layers.append(mod)
+            elif isinstance(mod, L.BatchNorm2d):
+                self.recs.append(ParamRec(mod.gamma, name + ".gamma",
+                                          wd_mult=0.0))
+                self.recs.append(ParamRec(mod.beta, name + ".beta",
+                                          wd_mult=0.0))
+        # reference-parity flat views for the exchangers / checkpoints
+        self.params = [r.master for r in self.recs]
+        self.vels = [r.vel for r in self.recs]
+
+    def _refresh_layers(self):
+        """Repack dgrad weight copies after a weight change (GPU)."""
+        from theanompi_amd import ops as _ops
+        if not self._post_step_layers:
+            return
+        C = _ops.ext()
+        for mod in self._post_step_layers:
+            for m, wdg in zip(mod.plan.build_repack_maps(), mod._wdg):
+                if m is not None:
+                    C.gather_bf16(mod.w_bf16.view(-1), m.view(-1),
+                                  wdg.view(-1))
+
+    def refresh_all(self):
+        """Resync every derived weight copy after an out-of-band change to
+        the fp32 masters (exchange, merge, broadcast, checkpoint load)."""
+        with torch.no_grad():
+            for rec in self.recs:
+                if rec.bf16_out is not None:
+                    rec.bf16_out.copy_(rec.master.data)
+        self._refresh_layers()
+
+    # -- contract: compile ----------------------------------------------------
+
+    def compile_iter_fns(self, sync_type="cdd"):
+        """Reference compiled Theano functions here (minutes); we build the
+        optimizer and warm the kernels (instant)."""
+        self.sync_type = sync_type
+        self.opt = SGDMomentum(self.recs, lr=self.current_lr,
+                               momentum=self.momentum,
+                               weight_decay=self.weight_decay)
+
+    def sync_tensors(self):
+        """What the BSP exchanger moves, by sync mode."""
+        if self.sync_type == "cdd":
+            return self.vels
+        if self.sync_type == "avg":
+            return self.params
+        if self.sync_type == "grad":
+            return self.opt.grads()
+        raise ValueError(self.sync_type)
+
+    # -- contract: train ------------------------------------------------------
+
+    def train_iter(self, count, recorder=None):
+        x, y = self.data.get_train_batch(count)
+        if recorder:
+            recorder.end("wait")
+        loss, err = self._fwd_bwd(x, y)
+        if self.sync_type == "cdd":
+            self.opt.phase1()
+        elif self.sync_type == "avg":
+            self.opt.step()
+        if recorder:
+            recorder.end("calc")
+            recorder.train_error(count, loss, err)
+        return loss, err
+
+    def _fwd_bwd(self, x, y):
+        self.opt.zero_grad()
+        logits = self.forward(x)
+        loss, _probs = OF.softmax_xent(logits, y, self.n_classes)
+        loss.backward()
+        with torch.no_grad():
+            pred = logits[:, :self.n_classes].argmax(dim=1)
+            err = (pred != y).float().mean()
+        return loss.detach(), err
+
+    def apply_update(self):
+        """Called by the worker after the exchange (phase 2 of cdd; no-op
+        weight math for avg; full fused step for grad mode)."""
+        if self.sync_type == "cdd":
+            self.opt.phase2()
+        elif self.sync_type == "grad":
+            self.opt.step()
+        self._refresh_layers()
+
+    # -- contract: validate ---------------------------------------------------
+
+    def val_iter(self, count, recorder=None):
+        self.module.eval()
+        try:
+            with torch.no_grad():
+                x, y = self.data.get_val_batch(count)
+                logits = self.forward(x)[:, :self.n_classes].float()
+                loss, _ = OF.softmax_xent_ref(logits, y, self.n_classes)
+                top5 = logits.topk(5, dim=1).indices
+                top1err = (top5[:, 0] != y).float().mean()
+                top5err = (~(top5 == y[:, None]).any(dim=1)).float().mean()
+        finally:
+            self.module.train()
+        if recorder:
+            recorder.end("calc")
+        return float(loss), float(top1err), float(top5err)
+
+    # -- contract: schedule / checkpoint --------------------------------------
+
+    def adjust_hyperp(self, epoch):
+        """Step LR schedule (reference models stepped /10)."""
+        self.epoch = epoch
+        decays = sum(1 for e in self.lr_steps() if epoch >= e)
+        self.current_lr = self.lr * (0.1 ** decays)
+        if self.opt:
+            self.opt.lr = self.current_lr
+
+    def lr_steps(self):
+        return (20, 40, 60)
+
+    def _logical_param_list(self):
+        out = []
+        for _name, mod in self.module.named_modules():
+            if isinstance(mod, L._Base):
+                out.extend(mod.logical_params())
+        return out
+
+    def save(self, directory, epoch=None):
+        epoch = self.epoch if epoch is None else epoch
+        logical = [g() for (_n, g, _s) in self._logical_param_list()]
+        helper_funcs.save_weights(logical, directory, epoch)
+        helper_funcs.save_momentums(self.vels, directory, epoch)
+
+    def load(self, directory, epoch=None):
+        epoch = self.epoch if epoch is None else epoch
+        plist = self._logical_param_list()
+        import numpy as np
+        import os
+        for i, (_n, _g, setter) in enumerate(plist):
+            arr = np.load(os.path.join(directory, f"W_{i}_{epoch}.npy"))
+            setter(torch.from_numpy(arr))
+        helper_funcs.load_momentums(self.vels, directory, epoch)
+        self._refresh_layers()
+
+    def reset_iter(self, mode):
+        if hasattr(self.data, "reset"):
+            self.data.reset(mode)

@@ -1,0 +1,589 @@
+"""Layer library — rebuild of Theano-MPI's ``theanompi/models/layers2.py``
+[pub] as NHWC modules over the HIP op set.
+
+GPU path (bf16 compute, fp32 master):
+  * Conv2d / Linear keep their master weights as fp32 tensors in the
+    kernels' flat padded layout ([G, Kdf, Kgp] — see ops/hip/conv.hip),
+    refresh a bf16 working copy plus a repacked dgrad copy after every
+    optimizer step, and run the implicit-GEMM HIP kernels through a
+    torch.autograd.Function so fp32 grads land in ``weight.grad``.
+  * Channel dims are physically padded (ops.functional.cpad_for); padded
+    weight entries are zero and stay zero (their activations are zero or
+    their gradients are zero by construction).
+
+CPU path (fp32, plumbing/test mode): same modules dispatch to torch
+reference ops (ops/functional.py *_ref).
+"""
+
+import math
+
+import numpy as np
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from theanompi_amd import ops as _ops
+from theanompi_amd.ops import functional as OF
+from theanompi_amd.ops.functional import cpad_for
+
+# ---------------------------------------------------------------------------
+# gather-table / repack-map builders (host side, run once per layer)
+# ---------------------------------------------------------------------------
+
+PIECE = 16  # flat-k granularity in elements (32 B of bf16)
+
+
+def _build_spans(R, S, Cg, Cphys, combine_sc):
+    """Return (spans, flat_slots, elem2flat) where spans is a list of
+    (dh, dw, c0, length) contiguous runs in the input tensor, covering
+    every (r, s, c) reduction element; flat_slots is the padded flat
+    K-dim; elem2flat maps (r, s, c) -> flat index."""
+    spans = []
+    elem2flat = {}
+    base = 0
+    if combine_sc:
+        # one run per r covering all (s, c): contiguous when Cg == Cphys
+        for r in range(R):
+            length = S * Cg
+            spans.append((r, 0, 0, length, base))
+            for s in range(S):
+                for c in range(Cg):
+                    elem2flat[(r, s, c)] = base + s * Cg + c
+            base += (length + PIECE - 1) // PIECE * PIECE
+    else:
+        for r in range(R):
+            for s in range(S):
+                length = Cg
+                spans.append((r, s, 0, length, base))
+                for c in range(Cg):
+                    elem2flat[(r, s, c)] = base + c
+                base += (length + PIECE - 1) // PIECE * PIECE
+    return spans, base, elem2flat
+
+
+def _spans_to_table(spans, flat_slots, Wp, Cphys):
+    """int32 element-offset per 16-piece from the (group-adjusted) pixel
+    base pointer.  Padding pieces point at the span start (their weights
+    are zero, so the gathered values are irrelevant)."""
+    n_pieces = flat_slots // PIECE
+    table = np.zeros(n_pieces, dtype=np.int32)
+    for (dh, dw, c0, length, base) in spans:
+        run_start = dh * Wp * Cphys + dw * Cphys + c0
+        n_slots = (length + PIECE - 1) // PIECE * PIECE
+        for j in range(n_slots // PIECE):
+            pos = j * PIECE
+            if pos < length:
+                table[(base // PIECE) + j] = run_start + pos
+            else:
+                table[(base // PIECE) + j] = run_start
+    return table
+
+
+class ConvPlan:
+    """All layout/dispatch state for one conv (or linear) layer on GPU.
+
+    Forward GEMM:   y[M=N*P*Q, Kgp]  = gatherA(xp)[M, Kdf] @ Wflat[Kdf, Kgp]
+    Dgrad GEMM:     dx[M=N*H*W, Cgp] = gatherA(dyp)[M, Kdf2] @ Wdg[Kdf2, Cgp]
+                    (per (sh x sw) output class when stride > 1)
+    Wgrad GEMM(TN): dW[Kdf, Kgp]    += gatherA(xp)[Mpix, Kdf]^T @ dy[Mpix, Kgp]
+    """
+
+    def __init__(self, N, H, W, C, K, R, S, stride, pad, groups, device):
+        sh, sw = stride
+        ph, pw = pad
+        self.NHWC = (N, H, W, C)
+        self.R, self.S, self.G = R, S, groups
+        self.stride, self.pad = stride, pad
+        self.Cg = C // groups
+        self.Cphys = C
+        assert C % groups == 0
+        assert self.Cg == cpad_for(self.Cg) or groups == 1, \
+            "per-group channels must be physically padded"
+        self.Kg = K // groups            # logical K per group
+        self.Kgp = cpad_for(self.Kg)
+        self.Kphys = groups * self.Kgp
+        self.P = (H + 2 * ph - R) // sh + 1
+        self.Q = (W + 2 * pw - S) // sw + 1
+        self.Hp, self.Wp = H + 2 * ph, W + 2 * pw
+        self.device = device
+
+        combine = (groups == 1)
+        spans, self.Kdf, self.e2f = _build_spans(R, S, self.Cg, C, combine)
+        tab = _spans_to_table(spans, self.Kdf, self.Wp, C)
+        self.table = torch.from_numpy(tab).to(device)
+
+        # ---- dgrad plans: one per (sh x sw) output class --------------------
+        self.dgrad_plans = []
+        Hd = self.P + 0  # dy logical dims
+        for i in range(sh):
+            for j in range(sw):
+                Hi = (H - 1 - i) // sh + 1 if i < H else 0
+                Wj = (W - 1 - j) // sw + 1 if j < W else 0
+                if Hi <= 0 or Wj <= 0:
+                    continue
+                r_list = [r for r in range(R) if (i + ph - r) % sh == 0]
+                s_list = [s for s in range(S) if (j + pw - s) % sw == 0]
+                dr = {r: (i + ph - r) // sh for r in r_list}
+                ds = {s: (j + pw - s) // sw for s in s_list}
+                # halo needed on dy so all (p' + dr) are in range
+                if not r_list or not s_list:
+                    # this class gets no contribution (possible when
+                    # kernel doesn't cover the class) -> zero dx there
+                    self.dgrad_plans.append(None)
+                    continue
+                halo_t = max(0, -min(dr.values()))
+                halo_b = max(0, (Hi - 1) + max(dr.values()) - (self.P - 1))
+                halo_l = max(0, -min(ds.values()))
+                halo_r = max(0, (Wj - 1) + max(ds.values()) - (self.Q - 1))
+                Qp = self.Q + halo_l + halo_r
+                spans2 = []
+                base = 0
+                e2f2 = {}
+                for r in r_list:
+                    for s in s_list:
+                        spans2.append((dr[r] + halo_t, ds[s] + halo_l, 0,
+                                       self.Kgp, base))
+                        for k in range(self.Kgp):
+                            e2f2[(r, s, k)] = base + k
+                        base += self.Kgp
+                tab2 = _spans_to_table(spans2, base, Qp, self.Kphys)
+                self.dgrad_plans.append({
+                    "cls": (i, j), "Hi": Hi, "Wj": Wj,
+                    "halo": (halo_t, halo_b, halo_l, halo_r),
+                    "Kdf2": base, "e2f2": e2f2,
+                    "table": torch.from_numpy(tab2).to(device),
+                })
+        # repack maps (dest flat [G,Kdf2,Cgp] <- src flat [G,Kdf,Kgp])
+        self._repack_maps = None
+
+    # -- weight layout ------------------------------------------------------
+
+    def flat_shape(self):
+        return (self.G, self.Kdf, self.Kgp)
+
+    def w_to_flat(self, w_kcrs, out=None):
+        """logical [K, Cg, R, S] fp32 -> flat [G, Kdf, Kgp]."""
+        G, Kg, Kgp = self.G, self.Kg, self.Kgp
+        flat = out if out is not None else torch.zeros(
+            self.flat_shape(), dtype=torch.float32, device=w_kcrs.device)
+        flat.zero_()
+        for r in range(self.R):
+            for s in range(self.S):
+                fd0 = self.e2f[(r, s, 0)]
+                for g in range(G):
+                    blk = w_kcrs[g * Kg:(g + 1) * Kg, :, r, s]  # [Kg, Cg]
+                    flat[g, fd0:fd0 + self.Cg, :Kg] = blk.t()
+        return flat
+
+    def flat_to_w(self, flat):
+        G, Kg = self.G, self.Kg
+        w = torch.zeros((G * Kg, self.Cg, self.R, self.S),
+                        dtype=torch.float32, device=flat.device)
+        for r in range(self.R):
+            for s in range(self.S):
+                fd0 = self.e2f[(r, s, 0)]
+                for g in range(G):
+                    w[g * Kg:(g + 1) * Kg, :, r, s] = \
+                        flat[g, fd0:fd0 + self.Cg, :Kg].t()
+        return w
+
+    def build_repack_maps(self):
+        """int32 maps dest-linear -> src-linear (-1 => zero) for each
+        dgrad class; used by _C.gather_bf16 after every optimizer step."""
+        if self._repack_maps is not None:
+            return self._repack_maps
+        maps = []
+        G, Kgp, Kg = self.G, self.Kgp, self.Kg
+        src_strideG = self.Kdf * Kgp
+        for plan in self.dgrad_plans:
+            if plan is None:
+                maps.append(None)
+                continue
+            Kdf2 = plan["Kdf2"]
+            m = np.full((G, Kdf2, self.Cg), -1, dtype=np.int32)
+            car = np.arange(self.Cg, dtype=np.int32)
+            for (r, s, k), fd2 in plan["e2f2"].items():
+                if k >= Kg:
+                    continue  # padded k: stays zero
+                fd = self.e2f[(r, s, 0)]
+                # dest[g, fd2, c] = w_flat[g, fd + c, k]
+                for g in range(G):
+                    m[g, fd2, :] = (g * src_strideG + (fd + car) * Kgp + k)
+            mt = torch.from_numpy(m).to(self.device)
+            maps.append(mt)
+        self._repack_maps = maps
+        return maps
+
+
+# ---------------------------------------------------------------------------
+# GPU autograd Function for conv / linear
+# ---------------------------------------------------------------------------
+
+class _HipConvFn(torch.autograd.Function):
+
+    @staticmethod
+    def forward(ctx, x, w_master, b_master, layer):
+        C = _ops.ext()
+        plan = layer.plan
+        ph, pw = plan.pad
+        xp = C.halo_pad(x, ph, pw, 0.0) if (ph or pw) else x.contiguous()
+        N = x.shape[0]
+        M = N * plan.P * plan.Q
+        y = torch.empty((N, plan.P, plan.Q, plan.Kphys),
+                        dtype=torch.bfloat16, device=x.device)
+        C.conv_fwd(xp, layer.w_bf16, layer.b_f32, plan.table, y,
+                   plan.Wp * plan.Cphys,            # pixel row stride (h)
+                   plan.Cphys,                      # pixel col stride (w)
+                   plan.stride[0], plan.stride[1],
+                   plan.Cg if plan.G > 1 else 0,    # group in-chan offset
+                   bool(layer.relu))
+        ctx.save_for_backward(xp, y if layer.relu else torch.empty(0))
+        ctx.layer = layer
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        C = _ops.ext()
+        layer = ctx.layer
+        plan = layer.plan
+        xp, y = ctx.saved_tensors
+        dy = dy.contiguous()
+        if layer.relu:
+            dym = torch.empty_like(dy)
+            C.relu_bwd(dy, y, dym)
+            dy = dym
+        # bias grad
+        db = torch.zeros_like(layer.b_f32)
+        C.bias_wgrad(dy, db)
+        # weight grad (fp32, flat layout)
+        dw = torch.zeros(plan.flat_shape(), dtype=torch.float32,
+                         device=dy.device)
+        C.conv_wgrad(xp, dy, plan.table, dw,
+                     plan.Wp * plan.Cphys, plan.Cphys,
+                     plan.stride[0], plan.stride[1],
+                     plan.Cg if plan.G > 1 else 0)
+        dx = None
+        if ctx.needs_input_grad[0]:
+            dx = layer._dgrad(dy)
+        return dx, dw, db, None
+
+
+# ---------------------------------------------------------------------------
+# modules
+# ---------------------------------------------------------------------------
+
+class _Base(nn.Module):
+    """Common bits: device/gpu flag + checkpoint hooks."""
+
+    def logical_params(self):
+        """[(name, getter()->np-compatible tensor, setter(tensor))] in the
+        reference's per-layer W-then-b order."""
+        return []
+
+    def post_step(self):
+        """Called after every optimizer step (refresh bf16 copies)."""
+
+
+def _init_tensor(shape, init, fan_in=None, fan_out=None):
+    t = torch.empty(*shape, dtype=torch.float32)
+    kind, *arg = (init if isinstance(init, tuple) else (init,))
+    if kind == "normal":
+        t.normal_(0.0, arg[0] if arg else 0.01)
+    elif kind == "he":
+        t.normal_(0.0, math.sqrt(2.0 / fan_in))
+    elif kind == "xavier":
+        a = math.sqrt(6.0 / (fan_in + fan_out))
+        t.uniform_(-a, a)
+    elif kind == "constant":
+        t.fill_(arg[0] if arg else 0.0)
+    else:
+        raise ValueError(kind)
+    return t
+
+
+class Conv2d(_Base):
+    """NHWC conv (+ optional fused ReLU).
+
+    GPU: implicit-GEMM HIP kernels (ops/hip/conv.hip; reference N1-N3 in
+    SURVEY.md §2.6).  CPU: torch reference.  Weights logical [K, Cg, R, S].
+    """
+
+    def __init__(self, in_shape, K, R, S=None, stride=1, pad=0, groups=1,
+                 relu=False, init=("he",), bias_init=0.0, device="cpu",
+                 batch=None):
+        super().__init__()
+        S = S if S is not None else R
+        stride = (stride, stride) if isinstance(stride, int) else stride
+        pad = (pad, pad) if isinstance(pad, int) else pad
+        H, W, Cin = in_shape            # logical input HWC (phys C on GPU)
+        self.K, self.R, self.S = K, R, S
+        self.stride, self.pad, self.groups = stride, pad, groups
+        self.relu = relu
+        self.device_t = torch.device(device)
+        self.gpu = self.device_t.type == "cuda"
+        self.Cg_logical = Cin // groups
+
+        fan_in = self.Cg_logical * R * S
+        w0 = _init_tensor((K, self.Cg_logical, R, S), init, fan_in,
+                          K * R * S // groups)
+        b0 = torch.full((K,), float(bias_init))
+
+        if self.gpu:
+            Cphys = groups * cpad_for(self.Cg_logical)
+            self.Cphys_in = Cphys
+            self.plan = ConvPlan(batch or 1, H, W, Cphys, K, R, S,
+                                 stride, pad, groups, self.device_t)
+            wl = torch.zeros((K, Cphys // groups, R, S))
+            wl[:, :self.Cg_logical] = w0
+            flat = self.plan.w_to_flat(wl.to(self.device_t))
+            self.weight = nn.Parameter(flat)
+            bp = torch.zeros(self.plan.Kphys, device=self.device_t)
+            for g in range(groups):
+                bp[g * self.plan.Kgp:
+                   g * self.plan.Kgp + self.plan.Kg] = b0[g * self.plan.Kg:
+                                                          (g + 1) * self.plan.Kg]
+            self.bias = nn.Parameter(bp)
+            self.register_buffer("w_bf16",
+                                 flat.to(torch.bfloat16).contiguous())
+            self.b_f32 = self.bias.data  # alias used by kernels
+            maps = self.plan.build_repack_maps()
+            self._wdg = []
+            for m in maps:
+                if m is None:
+                    self._wdg.append(None)
+                else:
+                    self._wdg.append(torch.zeros(m.shape,
+                                                 dtype=torch.bfloat16,
+                                                 device=self.device_t))
+            self.out_shape = (self.plan.P, self.plan.Q, self.plan.Kphys)
+            self.out_logical_c = K
+        else:
+            self.weight = nn.Parameter(w0)
+            self.bias = nn.Parameter(b0)
+            P = (H + 2 * pad[0] - R) // stride[0] + 1
+            Q = (W + 2 * pad[1] - S) // stride[1] + 1
+            self.out_shape = (P, Q, K)
+            self.out_logical_c = K
+
+    def forward(self, x):
+        if self.gpu:
+            return _HipConvFn.apply(x, self.weight, self.bias, self)
+        c_log = self.Cg_logical * self.groups
+        if x.shape[-1] != c_log:
+            x = x[..., :c_log]
+        return OF.conv2d_ref(x, self.weight, self.bias, self.stride,
+                             self.pad, self.groups, self.relu)
+
+    def _dgrad(self, dy):
+        C = _ops.ext()
+        plan = self.plan
+        N = dy.shape[0]
+        H, W = plan.NHWC[1], plan.NHWC[2]
+        dx = torch.zeros((N, H, W, plan.Cphys), dtype=torch.bfloat16,
+                         device=dy.device)
+        for plan2, wdg in zip(plan.dgrad_plans, self._wdg):
+            if plan2 is None:
+                continue
+            ht, hb, hl, hr = plan2["halo"]
+            dyp = C.halo_pad_lrtb(dy, ht, hb, hl, hr, 0.0) \
+                if (ht or hb or hl or hr) else dy
+            i, j = plan2["cls"]
+            Qp = plan.Q + hl + hr
+            C.conv_dgrad(dyp, wdg, plan2["table"], dx,
+                         Qp * plan.Kphys, plan.Kphys,
+                         plan2["Hi"], plan2["Wj"],
+                         i, j, plan.stride[0], plan.stride[1],
+                         plan.Kgp if plan.G > 1 else 0,
+                         plan.Cg)
+        return dx
+
+    def post_step(self):
+        if not self.gpu:
+            return
+        C = _ops.ext()
+        self.w_bf16.copy_(self.weight.data)   # fp32 -> bf16 cast copy
+        self.b_f32 = self.bias.data
+        for m, wdg in zip(self.plan.build_repack_maps(), self._wdg):
+            if m is not None:
+                C.gather_bf16(self.w_bf16.view(-1), m.view(-1),
+                              wdg.view(-1))
+
+    def logical_params(self):
+        if self.gpu:
+            return [
+                ("W", lambda: self.plan.flat_to_w(self.weight.data)
+                                  [:, :self.Cg_logical],
+                 self._set_w),
+                ("b", lambda: self._get_b(), self._set_b),
+            ]
+        return [("W", lambda: self.weight.data,
+                 lambda v: self.weight.data.copy_(v)),
+                ("b", lambda: self.bias.data,
+                 lambda v: self.bias.data.copy_(v))]
+
+    def _get_b(self):
+        plan = self.plan
+        out = torch.empty(self.K, device=self.device_t)
+        for g in range(self.groups):
+            out[g * plan.Kg:(g + 1) * plan.Kg] = \
+                self.bias.data[g * plan.Kgp:g * plan.Kgp + plan.Kg]
+        return out
+
+    def _set_b(self, v):
+        plan = self.plan
+        for g in range(self.groups):
+            self.bias.data[g * plan.Kgp:g * plan.Kgp + plan.Kg] = \
+                v[g * plan.Kg:(g + 1) * plan.Kg]
+        self.post_step()
+
+    def _set_w(self, v):
+        plan = self.plan
+        wl = torch.zeros((self.K, plan.Cg, self.R, self.S))
+        wl[:, :self.Cg_logical] = v.cpu()
+        plan.w_to_flat(wl.to(self.device_t), out=self.weight.data)
+        self.post_step()
+
+
+class Linear(_Base):
+    """FC layer; logical weight [IN, OUT] (Theano layout).  GPU path is a
+    1x1x1 image conv through the same implicit-GEMM kernel (N11)."""
+
+    def __init__(self, n_in, n_out, relu=False, init=("he",),
+                 bias_init=0.0, device="cpu", batch=None):
+        super().__init__()
+        self.n_in, self.n_out = n_in, n_out
+        self.relu = relu
+        self.device_t = torch.device(device)
+        self.gpu = self.device_t.type == "cuda"
+        w0 = _init_tensor((n_in, n_out), init, n_in, n_out)
+        b0 = torch.full((n_out,), float(bias_init))
+        if self.gpu:
+            inp = cpad_for(n_in)
+            self.plan = ConvPlan(batch or 1, 1, 1, inp, n_out, 1, 1,
+                                 (1, 1), (0, 0), 1, self.device_t)
+            flat = torch.zeros(self.plan.flat_shape(), device=self.device_t)
+            flat[0, :n_in, :n_out] = w0.to(self.device_t)
+            self.weight = nn.Parameter(flat)
+            bp = torch.zeros(self.plan.Kphys, device=self.device_t)
+            bp[:n_out] = b0
+            self.bias = nn.Parameter(bp)
+            self.register_buffer("w_bf16",
+                                 flat.to(torch.bfloat16).contiguous())
+            self.b_f32 = self.bias.data
+            maps = self.plan.build_repack_maps()
+            self._wdg = [None if m is None else
+                         torch.zeros(m.shape, dtype=torch.bfloat16,
+                                     device=self.device_t) for m in maps]
+            self.out_features_phys = self.plan.Kphys
+        else:
+            self.weight = nn.Parameter(w0)
+            self.bias = nn.Parameter(b0)
+            self.out_features_phys = n_out
+
+    def forward(self, x):
+        if self.gpu:
+            n = x.shape[0]
+            x4 = x.view(n, 1, 1, -1)
+            y = _HipConvFn.apply(x4, self.weight, self.bias, self)
+            return y.view(n, -1)
+        if x.shape[-1] != self.n_in:
+            x = x[..., :self.n_in]
+        return OF.linear_ref(x, self.weight, self.bias, self.relu)
+
+    _dgrad = Conv2d._dgrad
+    post_step = Conv2d.post_step
+
+    def logical_params(self):
+        if self.gpu:
+            return [("W", lambda: self.weight.data[0, :self.n_in,
+                                                   :self.n_out],
+                     self._set_w),
+                    ("b", lambda: self.bias.data[:self.n_out],
+                     self._set_b)]
+        return [("W", lambda: self.weight.data,
+                 lambda v: self.weight.data.copy_(v)),
+                ("b", lambda: self.bias.data,
+                 lambda v: self.bias.data.copy_(v))]
+
+    def _set_w(self, v):
+        self.weight.data.zero_()
+        self.weight.data[0, :self.n_in, :self.n_out] = v.to(self.device_t)
+        self.post_step()
+
+    def _set_b(self, v):
+        self.bias.data.zero_()
+        self.bias.data[:self.n_out] = v.to(self.device_t)
+        self.post_step()
+
+
+class MaxPool2d(_Base):
+    def __init__(self, kernel, stride, pad=0):
+        super().__init__()
+        self.kernel, self.stride, self.pad = kernel, stride, pad
+
+    def forward(self, x):
+        return OF.maxpool2d(x, self.kernel, self.stride, self.pad)
+
+
+class AvgPool2d(_Base):
+    def __init__(self, kernel, stride, pad=0):
+        super().__init__()
+        self.kernel, self.stride, self.pad = kernel, stride, pad
+
+    def forward(self, x):
+        return OF.avgpool2d(x, self.kernel, self.stride, self.pad)
+
+
+class LRN(_Base):
+    """Cross-map local response normalisation (AlexNet/GoogLeNet stem)."""
+
+    def __init__(self, size=5, alpha=1e-4, beta=0.75, k=1.0):
+        super().__init__()
+        self.size, self.alpha, self.beta, self.k = size, alpha, beta, k
+
+    def forward(self, x):
+        return OF.lrn(x, self.size, self.alpha, self.beta, self.k)
+
+
+class BatchNorm2d(_Base):
+    def __init__(self, c_phys, eps=1e-5, momentum=0.1, relu=False,
+                 device="cpu"):
+        super().__init__()
+        dev = torch.device(device)
+        self.gamma = nn.Parameter(torch.ones(c_phys, device=dev))
+        self.beta = nn.Parameter(torch.zeros(c_phys, device=dev))
+        self.register_buffer("running_mean",
+                             torch.zeros(c_phys, device=dev))
+        self.register_buffer("running_var",
+                             torch.ones(c_phys, device=dev))
+        self.eps, self.momentum, self.relu = eps, momentum, relu
+
+    def forward(self, x):
+        return OF.batchnorm(x, self.gamma, self.beta, self.running_mean,
+                            self.running_var, self.training, self.momentum,
+                            self.eps, self.relu)
+
+    def logical_params(self):
+        return [("gamma", lambda: self.gamma.data,
+                 lambda v: self.gamma.data.copy_(v)),
+                ("beta", lambda: self.beta.data,
+                 lambda v: self.beta.data.copy_(v))]
+
+
+class Dropout(_Base):
+    def __init__(self, p=0.5):
+        super().__init__()
+        self.p = p
+
+    def forward(self, x):
+        return OF.dropout(x, self.p, self.training)
+
+
+class ReLU(_Base):
+    def forward(self, x):
+        return OF.relu(x)
+
+
+class Flatten(_Base):
+    def forward(self, x):
+        return x.reshape(x.shape[0], -1)
