@@ -160,11 +160,14 @@ def test_sgd_momentum_semantics():
 def test_convplan_table_dense():
     # 3x3 conv, C=32 (one combined span per r of length 96)
     plan = ConvPlan(2, 8, 8, 32, 64, 3, 3, (1, 1), (1, 1), 1, "cpu")
-    assert plan.Kdf == 3 * 96
+    assert plan.Kdf == 320          # 3*96 = 288, rounded up to KPAD=64
     tab = plan.table.numpy()
     # piece j inside row r covers elements r*Wp*C + j16
     Wp = 10
     for j, off in enumerate(tab):
+        if j * PIECE >= 288:
+            assert off == 0          # KPAD padding entries
+            continue
         r = (j * PIECE) // 96
         pos = (j * PIECE) % 96
         assert off == r * Wp * 32 + pos
@@ -173,10 +176,13 @@ def test_convplan_table_dense():
 def test_convplan_table_stem():
     # 11x11 stride-4 stem, C=4: runs of 44, padded to 48
     plan = ConvPlan(1, 227, 227, 4, 96, 11, 11, (4, 4), (0, 0), 1, "cpu")
-    assert plan.Kdf == 11 * 48
+    assert plan.Kdf == (11 * 48 + 63) // 64 * 64
     tab = plan.table.numpy()
     Wp = 227
     for j, off in enumerate(tab):
+        if j * PIECE >= 11 * 48:
+            assert off == 0          # KPAD padding entries
+            continue
         r = (j * PIECE) // 48
         pos = (j * PIECE) % 48
         if pos < 44:
@@ -189,7 +195,7 @@ def test_convplan_grouped():
     # grouped conv: spans per (r, s) of length Cg
     plan = ConvPlan(1, 27, 27, 96, 256, 5, 5, (1, 1), (2, 2), 2, "cpu")
     assert plan.Cg == 48
-    assert plan.Kdf == 5 * 5 * 48
+    assert plan.Kdf == (5 * 5 * 48 + 63) // 64 * 64
     assert plan.Kgp == 128
     assert plan.Kphys == 256
 
@@ -215,6 +221,6 @@ def test_convplan_flat_roundtrip_grouped():
     plan = ConvPlan(1, 8, 8, 32, 24, 3, 3, (1, 1), (1, 1), 2, "cpu")
     w = torch.randn(24, 16, 3, 3)     # Kg=12 per group, Cg=16
     flat = plan.w_to_flat(w)
-    assert flat.shape == (2, plan.Kdf, plan.Kgp)
+    assert flat.shape == (2, plan.Kgp, plan.Kdf)
     back = plan.flat_to_w(flat)
     assert torch.allclose(w, back)

@@ -3,7 +3,7 @@
 
 GPU path (bf16 compute, fp32 master):
   * Conv2d / Linear keep their master weights as fp32 tensors in the
-    kernels' flat padded layout ([G, Kdf, Kgp] — see ops/hip/conv.hip),
+    kernels' flat padded layout ([G, Kgp, Kdf] — see ops/hip/conv.hip),
     refresh a bf16 working copy plus a repacked dgrad copy after every
     optimizer step, and run the implicit-GEMM HIP kernels through a
     torch.autograd.Function so fp32 grads land in ``weight.grad``.
@@ -30,7 +30,8 @@ from theanompi_amd.ops.functional import cpad_for
 # gather-table / repack-map builders (host side, run once per layer)
 # ---------------------------------------------------------------------------
 
-PIECE = 16  # flat-k granularity in elements (32 B of bf16)
+PIECE = 16   # flat-k granularity in elements (32 B of bf16)
+KPAD = 64    # flat reduction dim padded to one GEMM K-tile
 
 
 def _build_spans(R, S, Cg, Cphys, combine_sc):
@@ -50,7 +51,9 @@ def _build_spans(R, S, Cg, Cphys, combine_sc):
                 for c in range(Cg):
                     elem2flat[(r, s, c)] = base + s * Cg + c
             base += (length + PIECE - 1) // PIECE * PIECE
-    else:
+        base = (base + KPAD - 1) // KPAD * KPAD
+        return spans, base, elem2flat
+    if True:
         for r in range(R):
             for s in range(S):
                 length = Cg
@@ -58,6 +61,7 @@ def _build_spans(R, S, Cg, Cphys, combine_sc):
                 for c in range(Cg):
                     elem2flat[(r, s, c)] = base + c
                 base += (length + PIECE - 1) // PIECE * PIECE
+    base = (base + KPAD - 1) // KPAD * KPAD
     return spans, base, elem2flat
 
 
@@ -111,6 +115,8 @@ class ConvPlan:
         spans, self.Kdf, self.e2f = _build_spans(R, S, self.Cg, C, combine)
         tab = _spans_to_table(spans, self.Kdf, self.Wp, C)
         self.table = torch.from_numpy(tab).to(device)
+        self.kdf_mask = self.build_kdf_mask() if str(device) != "cpu" \
+            else None
 
         # ---- dgrad plans: one per (sh x sw) output class --------------------
         self.dgrad_plans = []
@@ -146,6 +152,7 @@ class ConvPlan:
                         for k in range(self.Kgp):
                             e2f2[(r, s, k)] = base + k
                         base += self.Kgp
+                base = (base + KPAD - 1) // KPAD * KPAD
                 tab2 = _spans_to_table(spans2, base, Qp, self.Kphys)
                 self.dgrad_plans.append({
                     "cls": (i, j), "Hi": Hi, "Wj": Wj,
@@ -159,11 +166,12 @@ class ConvPlan:
     # -- weight layout ------------------------------------------------------
 
     def flat_shape(self):
-        return (self.G, self.Kdf, self.Kgp)
+        # out-channel-major so the GEMM's B tile is k-contiguous per row
+        return (self.G, self.Kgp, self.Kdf)
 
     def w_to_flat(self, w_kcrs, out=None):
-        """logical [K, Cg, R, S] fp32 -> flat [G, Kdf, Kgp]."""
-        G, Kg, Kgp = self.G, self.Kg, self.Kgp
+        """logical [K, Cg, R, S] fp32 -> flat [G, Kgp, Kdf]."""
+        G, Kg = self.G, self.Kg
         flat = out if out is not None else torch.zeros(
             self.flat_shape(), dtype=torch.float32, device=w_kcrs.device)
         flat.zero_()
@@ -172,7 +180,7 @@ class ConvPlan:
                 fd0 = self.e2f[(r, s, 0)]
                 for g in range(G):
                     blk = w_kcrs[g * Kg:(g + 1) * Kg, :, r, s]  # [Kg, Cg]
-                    flat[g, fd0:fd0 + self.Cg, :Kg] = blk.t()
+                    flat[g, :Kg, fd0:fd0 + self.Cg] = blk
         return flat
 
     def flat_to_w(self, flat):
@@ -184,8 +192,17 @@ class ConvPlan:
                 fd0 = self.e2f[(r, s, 0)]
                 for g in range(G):
                     w[g * Kg:(g + 1) * Kg, :, r, s] = \
-                        flat[g, fd0:fd0 + self.Cg, :Kg].t()
+                        flat[g, :Kg, fd0:fd0 + self.Cg]
         return w
+
+    def build_kdf_mask(self):
+        """f32 [Kdf]: 1 at real (r,s,c) slots, 0 at padding slots — used
+        to zero wgrad's contributions at padded flat positions so padded
+        master weights stay exactly zero."""
+        m = torch.zeros(self.Kdf, dtype=torch.float32)
+        for fd in self.e2f.values():
+            m[fd] = 1.0
+        return m.to(self.device)
 
     def build_repack_maps(self):
         """int32 maps dest-linear -> src-linear (-1 => zero) for each
@@ -194,21 +211,22 @@ class ConvPlan:
             return self._repack_maps
         maps = []
         G, Kgp, Kg = self.G, self.Kgp, self.Kg
-        src_strideG = self.Kdf * Kgp
+        src_strideG = Kgp * self.Kdf
         for plan in self.dgrad_plans:
             if plan is None:
                 maps.append(None)
                 continue
             Kdf2 = plan["Kdf2"]
-            m = np.full((G, Kdf2, self.Cg), -1, dtype=np.int32)
+            m = np.full((G, self.Cg, Kdf2), -1, dtype=np.int32)
             car = np.arange(self.Cg, dtype=np.int32)
             for (r, s, k), fd2 in plan["e2f2"].items():
                 if k >= Kg:
                     continue  # padded k: stays zero
                 fd = self.e2f[(r, s, 0)]
-                # dest[g, fd2, c] = w_flat[g, fd + c, k]
+                # dest[g, c, fd2] = w_flat[g, k, fd + c]
                 for g in range(G):
-                    m[g, fd2, :] = (g * src_strideG + (fd + car) * Kgp + k)
+                    m[g, :, fd2] = (g * src_strideG + k * self.Kdf
+                                    + fd + car)
             mt = torch.from_numpy(m).to(self.device)
             maps.append(mt)
         self._repack_maps = maps
@@ -262,6 +280,8 @@ class _HipConvFn(torch.autograd.Function):
                      plan.Wp * plan.Cphys, plan.Cphys,
                      plan.stride[0], plan.stride[1],
                      plan.Cg if plan.G > 1 else 0)
+        if plan.kdf_mask is not None:
+            dw.mul_(plan.kdf_mask)   # keep padded flat slots exactly zero
         dx = None
         if ctx.needs_input_grad[0]:
             dx = layer._dgrad(dy)
@@ -462,7 +482,7 @@ class Linear(_Base):
             self.plan = ConvPlan(batch or 1, 1, 1, inp, n_out, 1, 1,
                                  (1, 1), (0, 0), 1, self.device_t)
             flat = torch.zeros(self.plan.flat_shape(), device=self.device_t)
-            flat[0, :n_in, :n_out] = w0.to(self.device_t)
+            flat[0, :n_out, :n_in] = w0.t().to(self.device_t)
             self.weight = nn.Parameter(flat)
             bp = torch.zeros(self.plan.Kphys, device=self.device_t)
             bp[:n_out] = b0
@@ -495,8 +515,8 @@ class Linear(_Base):
 
     def logical_params(self):
         if self.gpu:
-            return [("W", lambda: self.weight.data[0, :self.n_in,
-                                                   :self.n_out],
+            return [("W", lambda: self.weight.data[0, :self.n_out,
+                                                   :self.n_in].t(),
                      self._set_w),
                     ("b", lambda: self.bias.data[:self.n_out],
                      self._set_b)]
@@ -507,7 +527,8 @@ class Linear(_Base):
 
     def _set_w(self, v):
         self.weight.data.zero_()
-        self.weight.data[0, :self.n_in, :self.n_out] = v.to(self.device_t)
+        self.weight.data[0, :self.n_out, :self.n_in] = \
+            v.t().to(self.device_t)
         self.post_step()
 
     def _set_b(self, v):
