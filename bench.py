@@ -1,0 +1,149 @@
+#!/usr/bin/env python3
+"""Driver benchmark contract (see BASELINE.json): measures the
+reference's headline metric — images/sec (whole job) for AlexNet-128b /
+GoogLeNet-32b BSP data-parallel training — on 1..8 MI355X.
+
+Single GPU:   python bench.py --steps 40 --warmup 10
+Multi GPU:    python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+                  --master-addr 127.0.0.1 bench.py --gpus N ...
+
+One rank per GPU over RCCL; synthetic ImageNet-shaped data (random,
+device-staged through the pinned double-buffer loader), random-init
+weights, bf16 compute with fp32 master weights.  Rank 0 prints ONE JSON
+line with the whole-job aggregate.
+"""
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+MODELS = {
+    "alexnet": ("theanompi_amd.models.alex_net", "AlexNet", 128),
+    "googlenet": ("theanompi_amd.models.googlenet", "GoogLeNet", 32),
+    "resnet50": ("theanompi_amd.models.resnet50", "ResNet50", 256),
+    "vgg16": ("theanompi_amd.models.vgg16", "VGG16", 64),
+    "cifar10": ("theanompi_amd.models.cifar10", "Cifar10_model", 128),
+}
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=40)
+    ap.add_argument("--warmup", type=int, default=10)
+    ap.add_argument("--model", default="alexnet", choices=sorted(MODELS))
+    ap.add_argument("--batch", type=int, default=0,
+                    help="per-GPU batch (0 = model default)")
+    ap.add_argument("--sync-type", default="cdd",
+                    choices=["cdd", "avg", "grad"])
+    ap.add_argument("--strategy", default="rccl_bf16")
+    ap.add_argument("--bucket-mb", type=float, default=64)
+    args = ap.parse_args()
+
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", str(args.gpus)))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    on_gpu = torch.cuda.is_available()
+    device = f"cuda:{local_rank}" if on_gpu else "cpu"
+    if on_gpu:
+        torch.cuda.set_device(local_rank)
+
+    import importlib
+    modname, clsname, def_batch = MODELS[args.model]
+    batch = args.batch or def_batch
+    cls = getattr(importlib.import_module(modname), clsname)
+
+    dist_on = world > 1
+    if dist_on:
+        import torch.distributed as dist
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "29555")
+        dist.init_process_group("nccl" if on_gpu else "gloo",
+                                rank=rank, world_size=world)
+
+    model = cls({"rank": rank, "size": world, "device": device,
+                 "verbose": rank == 0, "batch_size": batch,
+                 "n_iters": 16})
+    model.compile_iter_fns(args.sync_type)
+
+    from theanompi_amd.lib.exchanger import BSP_Exchanger
+    exchanger = BSP_Exchanger(
+        {"sync_type": args.sync_type, "exchange_strategy":
+             (args.strategy if on_gpu else "ar"),
+         "size": world, "rank": rank, "bucket_mb": args.bucket_mb},
+        model.params, model.vels)
+
+    def barrier():
+        if dist_on:
+            import torch.distributed as dist
+            dist.barrier()
+
+    def sync():
+        if on_gpu:
+            torch.cuda.synchronize()
+
+    def step(i):
+        model.train_iter(i % model.data.n_batch_train)
+        if args.sync_type == "grad":
+            exchanger.set_grad_buffers(model.opt.grads())
+        exchanger.exchange()
+        model.apply_update()
+
+    for i in range(args.warmup):
+        step(i)
+    barrier()
+    sync()
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        step(args.warmup + i)
+    barrier()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks (whole-job time)
+    if dist_on:
+        import torch.distributed as dist
+        t = torch.tensor([elapsed],
+                         device=device if on_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t[0])
+
+    if rank == 0:
+        imgs = world * batch * args.steps
+        value = imgs / elapsed
+        out = {
+            "metric": "images/sec (whole node) "
+                      f"{args.model} BSP",
+            "value": round(value, 2),
+            "unit": "images/sec",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(elapsed / args.steps * 1000, 3),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16" if on_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": f"{args.model}-{batch}b",
+                "global_batch": world * batch,
+                "image": getattr(model, "image_size", 0),
+                "parallelism": f"dp{world}",
+                "sync_type": args.sync_type,
+                "strategy": args.strategy if on_gpu else "ar",
+            },
+        }
+        print(json.dumps(out), flush=True)
+
+    if dist_on:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

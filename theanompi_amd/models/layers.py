@@ -377,6 +377,7 @@ class Conv2d(_Base):
                                                  device=self.device_t))
             self.out_shape = (self.plan.P, self.plan.Q, self.plan.Kphys)
             self.out_logical_c = K
+            self.post_step()          # fill w_bf16 + dgrad repack copies
         else:
             self.weight = nn.Parameter(w0)
             self.bias = nn.Parameter(b0)
@@ -494,7 +495,9 @@ class Linear(_Base):
             self._wdg = [None if m is None else
                          torch.zeros(m.shape, dtype=torch.bfloat16,
                                      device=self.device_t) for m in maps]
+            self.in_phys = inp
             self.out_features_phys = self.plan.Kphys
+            self.post_step()          # fill w_bf16 + dgrad repack copies
         else:
             self.weight = nn.Parameter(w0)
             self.bias = nn.Parameter(b0)
@@ -503,7 +506,9 @@ class Linear(_Base):
     def forward(self, x):
         if self.gpu:
             n = x.shape[0]
-            x4 = x.view(n, 1, 1, -1)
+            if x.shape[1] != self.in_phys:
+                x = F.pad(x, (0, self.in_phys - x.shape[1]))
+            x4 = x.reshape(n, 1, 1, self.in_phys)
             y = _HipConvFn.apply(x4, self.weight, self.bias, self)
             return y.view(n, -1)
         if x.shape[-1] != self.n_in:

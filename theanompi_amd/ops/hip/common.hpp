@@ -1,0 +1,65 @@
+// Common definitions for the theanompi_amd CDNA4 (gfx950) kernel library.
+// Hand-written HIP targeting MI355X: wave64, MFMA bf16 matrix cores,
+// 160 KiB LDS per CU.  No CUDA-compat paths, no hipify, no library math.
+#pragma once
+
+#include <hip/hip_runtime.h>
+
+#define DEV __device__ __forceinline__
+
+typedef __attribute__((ext_vector_type(2))) float f32x2;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+typedef __attribute__((ext_vector_type(4))) __bf16 bf16x4;
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+typedef __attribute__((ext_vector_type(4))) short short4v;
+typedef __attribute__((ext_vector_type(8))) short short8v;
+
+// 16-byte and 8-byte global load/store carriers with explicit (low)
+// alignment so pixel bases aligned only to the channel pad (8 B when
+// C_phys == 4) are legal.
+struct __align__(8) pack16 { unsigned int u[4]; };
+struct __align__(4) pack8  { unsigned int u[2]; };
+
+static inline int cdiv(int a, int b) { return (a + b - 1) / b; }
+
+DEV float bf2f(__bf16 v) { return (float)v; }
+DEV __bf16 f2bf(float v) { return (__bf16)v; }
+
+// ---------------------------------------------------------------------------
+// MFMA fragment addressing (v_mfma_f32_16x16x32_bf16)
+//
+// C/D layout (fixed, dtype-independent on gfx950): lane l holds 4 fp32,
+//   col = l & 15, row = (l >> 4) * 4 + reg.
+// A/B input layout: lane l holds 8 bf16 along K at row/col (l & 15).
+// The K offsets of those 8 registers are probed at runtime (bindings.cpp
+// mfma_probe): either 8 contiguous (koff = (l>>4)*8 + j) or two halves
+// of 4 (koff = (l>>4)*4 + j%4 + (j/4)*16).  Kernels take LAYOUT as a
+// template parameter and the host picks the instantiation once.
+// ---------------------------------------------------------------------------
+
+#define FRAG_CONTIG8 0
+#define FRAG_HALVES  1
+
+// Load one A/B fragment (8 bf16 along K) for lane `lane` from an LDS row
+// starting at `row_ptr` (element pointer to k-offset 0 of the 32-wide
+// K-step).  Rows are padded so 16-B alignment holds for layout 0 and 8-B
+// for layout 1.
+template <int LAYOUT>
+DEV bf16x8 frag_load(const __bf16* row_ptr, int lane) {
+  if constexpr (LAYOUT == FRAG_CONTIG8) {
+    return *(const bf16x8*)(row_ptr + (lane >> 4) * 8);
+  } else {
+    bf16x4 lo = *(const bf16x4*)(row_ptr + (lane >> 4) * 4);
+    bf16x4 hi = *(const bf16x4*)(row_ptr + (lane >> 4) * 4 + 16);
+    bf16x8 r;
+    r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
+    r[4] = hi[0]; r[5] = hi[1]; r[6] = hi[2]; r[7] = hi[3];
+    return r;
+  }
+}
+
+#define HIP_KERNEL_ASSERT_OK(expr)                                          \
+  do {                                                                      \
+    hipError_t _e = (expr);                                                 \
+    if (_e != hipSuccess) return _e;                                        \
+  } while (0)

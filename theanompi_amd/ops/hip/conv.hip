@@ -1,0 +1,375 @@
+// Implicit-GEMM convolution kernels for MI355X (gfx950), bf16 NHWC.
+//
+// One gather-GEMM kernel covers conv fwd, conv dgrad (per stride class)
+// and FC fwd/dgrad (R=S=1): replaces cuDNN conv fwd/bgrad + cuBLAS FC
+// GEMMs of the reference.  A TN split-K
+// variant produces wgrad (N3).
+//
+// GEMM view (fwd):  Y[M=N*P*Q, Kgp] = gatherA(X)[M, Kdf] @ W[Kdf, Kgp]
+//   - A rows are output pixels; the flat reduction dim (r,s,c) is read
+//     through a 16-element-piece offset TABLE (host-built; see
+//     models/layers.py ConvPlan).  Inputs are physically halo-padded so
+//     the gather needs no bounds checks.
+//   - W is stored out-channel-major [Kgp][Kdf] so both LDS tiles are
+//     k-contiguous per row and MFMA fragments load as 16-B ds_reads.
+//   - MFMA v_mfma_f32_16x16x32_bf16, 128x128 (or 128x64) block tile,
+//     BK=64, 4 waves, double-buffered LDS with stage-ahead (the
+//     2-phase schedule).
+//
+// Numerics: bf16 inputs, fp32 MFMA accumulation, bf16 outputs (wgrad:
+// fp32 atomic accumulation).
+
+#include "common.hpp"
+
+// ---------------------------------------------------------------------------
+// conv_gather_gemm: fwd / dgrad / linear
+// ---------------------------------------------------------------------------
+
+struct GatherGemmArgs {
+  const __bf16* __restrict__ X;     // gather source (halo-padded)
+  const __bf16* __restrict__ W;     // [G][BNtot][Kdf]
+  const float* __restrict__ bias;   // [G*BNtot] or nullptr
+  const int* __restrict__ table;    // [Kdf/16] element offsets
+  __bf16* __restrict__ Y;
+  int M, P, Q;                      // M = N*P*Q output pixels
+  long sN, sP, sQ, outOff;          // output strides/base (elements)
+  long inNS, inPS, inQS;            // input pixel strides (elements)
+  int groupInOff;                   // input channel offset per group
+  int Kdf;                          // flat reduction size (mult of 64)
+  int BNtot;                        // output channels per group (Kgp)
+  int relu;
+};
+
+template <int BM, int BN, int LAYOUT>
+__global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
+  constexpr int BK = 64;
+  constexpr int LDK = BK + 8;            // row pad: 16 B
+  constexpr int WAVES_M = (BN == 128) ? 2 : 4;
+  constexpr int WAVES_N = BN / 64;       // 2 or 1
+  constexpr int TM = BM / WAVES_M;       // 64 or 32
+  constexpr int FM = TM / 16;            // 4 or 2
+  constexpr int FN = 64 / 16;            // wave tile N = 64 always
+
+  __shared__ __align__(16) __bf16 As[2][BM][LDK];
+  __shared__ __align__(16) __bf16 Bs[2][BN][LDK];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int mTile = blockIdx.x, nTile = blockIdx.y, g = blockIdx.z;
+
+  const __bf16* Wg = a.W + (long)g * a.BNtot * a.Kdf;
+  const long gInOff = (long)g * a.groupInOff;
+  const int bn0 = nTile * BN;
+
+  // --- per-thread staging descriptors (fixed across K) --------------------
+  constexpr int A_PIECES = (BM * 8) / 256;   // 16-B pieces per thread
+  constexpr int B_PIECES = (BN * 8) / 256;
+  long aRowBase[A_PIECES];
+  int aRow[A_PIECES];
+  const int p8 = (tid & 7) * 8;              // element offset of my piece
+#pragma unroll
+  for (int t = 0; t < A_PIECES; ++t) {
+    int row = (tid >> 3) + t * 32;
+    int m = mTile * BM + row;
+    if (m >= a.M) m = a.M - 1;               // clamped rows are discarded
+    int q = m % a.Q;
+    int tmp = m / a.Q;
+    int p = tmp % a.P;
+    int n = tmp / a.P;
+    aRow[t] = row;
+    aRowBase[t] = (long)n * a.inNS + (long)p * a.inPS + (long)q * a.inQS
+                  + gInOff;
+  }
+  const __bf16* bRowPtr[B_PIECES];
+#pragma unroll
+  for (int t = 0; t < B_PIECES; ++t) {
+    int row = (tid >> 3) + t * 32;
+    int n = bn0 + row;
+    if (n >= a.BNtot) n = a.BNtot - 1;       // discarded at C-write
+    bRowPtr[t] = Wg + (long)n * a.Kdf;
+  }
+
+  // --- staging ------------------------------------------------------------
+  auto stage = [&](int buf, int k0) {
+#pragma unroll
+    for (int t = 0; t < A_PIECES; ++t) {
+      int off = a.table[(k0 + p8) >> 4] + (p8 & 15);
+      pack16 v = *(const pack16*)(a.X + aRowBase[t] + off);
+      *(pack16*)&As[buf][aRow[t]][p8] = v;
+    }
+#pragma unroll
+    for (int t = 0; t < B_PIECES; ++t) {
+      pack16 v = *(const pack16*)(bRowPtr[t] + k0 + p8);
+      *(pack16*)&Bs[buf][(tid >> 3) + t * 32][p8] = v;
+    }
+  };
+
+  // --- main loop ------------------------------------------------------------
+  const int wm = (WAVES_N == 2) ? (wave >> 1) : wave;
+  const int wn = (WAVES_N == 2) ? (wave & 1) : 0;
+  const int m0 = wm * TM;
+  const int n0 = wn * 64;
+  const int lrow = lane & 15;
+
+  f32x4 acc[FM][FN];
+#pragma unroll
+  for (int i = 0; i < FM; ++i)
+#pragma unroll
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nt = a.Kdf / BK;
+  stage(0, 0);
+  __syncthreads();
+  for (int t = 0; t < nt; ++t) {
+    if (t + 1 < nt) stage((t + 1) & 1, (t + 1) * BK);
+    const int buf = t & 1;
+#pragma unroll
+    for (int ks = 0; ks < BK; ks += 32) {
+      bf16x8 af[FM], bf[FN];
+#pragma unroll
+      for (int i = 0; i < FM; ++i)
+        af[i] = frag_load<LAYOUT>(&As[buf][m0 + i * 16 + lrow][ks], lane);
+#pragma unroll
+      for (int j = 0; j < FN; ++j)
+        bf[j] = frag_load<LAYOUT>(&Bs[buf][n0 + j * 16 + lrow][ks], lane);
+#pragma unroll
+      for (int i = 0; i < FM; ++i)
+#pragma unroll
+        for (int j = 0; j < FN; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // --- epilogue: bias + relu + strided bf16 store ---------------------------
+  const int cOut0 = g * a.BNtot + bn0;
+#pragma unroll
+  for (int i = 0; i < FM; ++i) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int rowl = m0 + i * 16 + (lane >> 4) * 4 + reg;
+      int m = mTile * BM + rowl;
+      if (m >= a.M) continue;
+      int q = m % a.Q;
+      int tmp = m / a.Q;
+      int p = tmp % a.P;
+      int n = tmp / a.P;
+      long obase = a.outOff + (long)n * a.sN + (long)p * a.sP
+                   + (long)q * a.sQ + cOut0;
+#pragma unroll
+      for (int j = 0; j < FN; ++j) {
+        int nl = n0 + j * 16 + (lane & 15);
+        if (bn0 + nl >= a.BNtot) continue;
+        float v = acc[i][j][reg];
+        if (a.bias) v += a.bias[cOut0 + nl];
+        if (a.relu) v = v > 0.f ? v : 0.f;
+        a.Y[obase + nl] = (__bf16)v;
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// conv_wgrad: TN split-K GEMM, fp32 atomic accumulation
+//   DW[g][kout][fd] += sum_pix DY[pix][g*Kgp+kout] * gatherA(X)[pix][fd]
+// ---------------------------------------------------------------------------
+
+struct WgradArgs {
+  const __bf16* __restrict__ X;     // halo-padded input
+  const __bf16* __restrict__ DY;    // [M][Kphys]
+  const int* __restrict__ table;    // [Kdf/16]
+  float* __restrict__ DW;           // [G][Kgp][Kdf]
+  int M, P, Q;
+  long inNS, inPS, inQS;
+  int groupInOff;
+  int Kdf, Kgp, Kphys;
+  int splitK;
+};
+
+template <int LAYOUT>
+__global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
+  constexpr int BMK = 64;   // kout tile
+  constexpr int BNF = 64;   // fd tile
+  constexpr int BKP = 64;   // pixel chunk
+  constexpr int LDK = BKP + 8;
+
+  __shared__ __align__(16) __bf16 DYt[2][BMK][LDK];  // [kout][pix]
+  __shared__ __align__(16) __bf16 XT[2][BNF][LDK];   // [fd][pix]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int kTile = blockIdx.x, fTile = blockIdx.y;
+  const int g = blockIdx.z / a.splitK;
+  const int slice = blockIdx.z % a.splitK;
+
+  const int kout0 = kTile * BMK;
+  const int fd0 = fTile * BNF;
+  const long gInOff = (long)g * a.groupInOff;
+  const long dyTotal = (long)a.M * a.Kphys;
+
+  const int pixl = tid >> 3;                // 0..31 (piece row within 32)
+  const int p8 = (tid & 7) * 8;
+
+  auto stage = [&](int buf, int pix0) {
+#pragma unroll
+    for (int t = 0; t < 2; ++t) {           // 64 pix rows / 32 per pass
+      int pix = pixl + t * 32;
+      int pa = pix0 + pix;
+      // ---- dy piece (transposed write), zero beyond M ----
+      pack16 v;
+      if (pa < a.M) {
+        long addr = (long)pa * a.Kphys + g * a.Kgp + kout0 + p8;
+        if (addr + 8 > dyTotal) addr = dyTotal - 8;
+        v = *(const pack16*)(a.DY + addr);
+      } else {
+        v.u[0] = v.u[1] = v.u[2] = v.u[3] = 0;
+      }
+      const __bf16* e = (const __bf16*)&v;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) DYt[buf][p8 + i][pix] = e[i];
+      // ---- x gather piece (transposed write) ----
+      int pc = pa < a.M ? pa : a.M - 1;
+      int q = pc % a.Q;
+      int tmp = pc / a.Q;
+      int p = tmp % a.P;
+      int n = tmp / a.P;
+      long base = (long)n * a.inNS + (long)p * a.inPS + (long)q * a.inQS
+                  + gInOff;
+      int off = a.table[(fd0 + p8) >> 4] + (p8 & 15);
+      pack16 w = *(const pack16*)(a.X + base + off);
+      const __bf16* f = (const __bf16*)&w;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) XT[buf][p8 + i][pix] = f[i];
+    }
+  };
+
+  const int wm = wave >> 1, wn = wave & 1;  // 2x2 waves, 32x32 each
+  const int m0 = wm * 32, n0 = wn * 32;
+  const int lrow = lane & 15;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nchunks = (a.M + BKP - 1) / BKP;
+  int ci = slice;
+  if (ci < nchunks) {
+    stage(0, ci * BKP);
+    __syncthreads();
+    int t = 0;
+    for (; ci < nchunks; ci += a.splitK, ++t) {
+      int cnext = ci + a.splitK;
+      if (cnext < nchunks) stage((t + 1) & 1, cnext * BKP);
+      const int buf = t & 1;
+#pragma unroll
+      for (int ks = 0; ks < BKP; ks += 32) {
+        bf16x8 af[2], bf[2];
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+          af[i] = frag_load<LAYOUT>(&DYt[buf][m0 + i * 16 + lrow][ks], lane);
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          bf[j] = frag_load<LAYOUT>(&XT[buf][n0 + j * 16 + lrow][ks], lane);
+#pragma unroll
+        for (int i = 0; i < 2; ++i)
+#pragma unroll
+          for (int j = 0; j < 2; ++j)
+            acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                af[i], bf[j], acc[i][j], 0, 0, 0);
+      }
+      __syncthreads();
+    }
+  }
+
+  // epilogue: atomic accumulate fp32
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int kout = kout0 + m0 + i * 16 + (lane >> 4) * 4 + reg;
+      if (kout >= a.Kgp) continue;
+      long base = ((long)g * a.Kgp + kout) * a.Kdf + fd0;
+#pragma unroll
+      for (int j = 0; j < 2; ++j) {
+        int fd = n0 + j * 16 + (lane & 15);
+        atomicAdd(&a.DW[base + fd], acc[i][j][reg]);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// bias grad: db[c] += sum over rows of DY[row][c]
+// ---------------------------------------------------------------------------
+
+__global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
+                             float* __restrict__ db, long rows, int C,
+                             int slices) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int slice = blockIdx.y;
+  if (c >= C) return;
+  float s = 0.f;
+  for (long r = slice; r < rows; r += slices)
+    s += (float)dy[r * C + c];
+  atomicAdd(&db[c], s);
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+extern "C" {
+
+hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
+                                   int layout, hipStream_t stream) {
+  GatherGemmArgs a = *args;
+  int BN = a.BNtot > 64 ? 128 : 64;
+  dim3 grid(cdiv(a.M, 128), cdiv(a.BNtot, BN), G);
+  dim3 block(256);
+  if (BN == 128) {
+    if (layout == FRAG_CONTIG8)
+      hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_CONTIG8>), grid,
+                         block, 0, stream, a);
+    else
+      hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_HALVES>), grid,
+                         block, 0, stream, a);
+  } else {
+    if (layout == FRAG_CONTIG8)
+      hipLaunchKernelGGL((conv_gather_gemm<128, 64, FRAG_CONTIG8>), grid,
+                         block, 0, stream, a);
+    else
+      hipLaunchKernelGGL((conv_gather_gemm<128, 64, FRAG_HALVES>), grid,
+                         block, 0, stream, a);
+  }
+  return hipGetLastError();
+}
+
+hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
+                             hipStream_t stream) {
+  WgradArgs a = *args;
+  dim3 grid(cdiv(a.Kgp, 64), cdiv(a.Kdf, 64), G * a.splitK);
+  dim3 block(256);
+  if (layout == FRAG_CONTIG8)
+    hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8>), grid, block, 0,
+                       stream, a);
+  else
+    hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES>), grid, block, 0,
+                       stream, a);
+  return hipGetLastError();
+}
+
+hipError_t launch_bias_wgrad(const void* dy, float* db, long rows, int C,
+                             hipStream_t stream) {
+  int slices = 64;
+  dim3 grid(cdiv(C, 256), slices);
+  hipLaunchKernelGGL(bias_wgrad_k, grid, dim3(256), 0, stream,
+                     (const __bf16*)dy, db, rows, C, slices);
+  return hipGetLastError();
+}
+
+}  // extern "C"

@@ -1,0 +1,238 @@
+// Elementwise / data-movement kernels:
+// halo padding, relu fwd/bwd, residual add, dropout (counter-based RNG),
+// repack gather, bf16<->fp32 pack.  All memory-bound: vectorized 8-16 B
+// per lane, grid-stride.
+
+#include "common.hpp"
+
+// ---- halo pad -------------------------------------------------------------
+// x [N,H,W,C] -> y [N, H+t+b, W+l+r, C], borders = fill. C % 4 == 0 so
+// 4-element (8 B) chunks never straddle a pixel.
+
+__global__ void halo_pad_k(const __bf16* __restrict__ x,
+                           __bf16* __restrict__ y, int N, int H, int W,
+                           int C, int t, int b, int l, int r, float fill,
+                           long total4) {
+  const int Hp = H + t + b, Wp = W + l + r;
+  const __bf16 f = (__bf16)fill;
+  for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < total4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
+    int c = (int)(i % C);
+    long pix = i / C;
+    int wp = (int)(pix % Wp);
+    long tmp = pix / Wp;
+    int hp = (int)(tmp % Hp);
+    int n = (int)(tmp / Hp);
+    int h = hp - t, w = wp - l;
+    pack8 v;
+    if (h >= 0 && h < H && w >= 0 && w < W) {
+      v = *(const pack8*)(x + (((long)n * H + h) * W + w) * C + c);
+    } else {
+      __bf16* e = (__bf16*)&v;
+      e[0] = f; e[1] = f; e[2] = f; e[3] = f;
+    }
+    *(pack8*)(y + i) = v;
+  }
+}
+
+// ---- relu / add -----------------------------------------------------------
+
+__global__ void relu_fwd_k(const __bf16* __restrict__ x,
+                           __bf16* __restrict__ y, long n4) {
+  const __bf16 z = (__bf16)0.f;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x4 v = *(const bf16x4*)(x + i * 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) v[j] = v[j] > z ? v[j] : z;
+    *(bf16x4*)(y + i * 4) = v;
+  }
+}
+
+__global__ void relu_bwd_k(const __bf16* __restrict__ dy,
+                           const __bf16* __restrict__ y,
+                           __bf16* __restrict__ dx, long n4) {
+  const __bf16 z = (__bf16)0.f;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x4 g = *(const bf16x4*)(dy + i * 4);
+    bf16x4 v = *(const bf16x4*)(y + i * 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) g[j] = v[j] > z ? g[j] : z;
+    *(bf16x4*)(dx + i * 4) = g;
+  }
+}
+
+__global__ void add_relu_k(const __bf16* __restrict__ p,
+                           const __bf16* __restrict__ q,
+                           __bf16* __restrict__ y, long n4, int relu) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x4 u = *(const bf16x4*)(p + i * 4);
+    bf16x4 v = *(const bf16x4*)(q + i * 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float s = (float)u[j] + (float)v[j];
+      if (relu && s < 0.f) s = 0.f;
+      u[j] = (__bf16)s;
+    }
+    *(bf16x4*)(y + i * 4) = u;
+  }
+}
+
+// ---- dropout --------------------------------------------------------------
+// Counter-based RNG (xorshift-mixed, Philox-style statelessness): mask
+// regenerable from (seed, index) but we store it for the backward pass.
+
+DEV unsigned int mix32(unsigned long long x) {
+  x ^= x >> 33; x *= 0xff51afd7ed558ccdULL;
+  x ^= x >> 33; x *= 0xc4ceb9fe1a85ec53ULL;
+  x ^= x >> 33;
+  return (unsigned int)x;
+}
+
+__global__ void dropout_fwd_k(const __bf16* __restrict__ x,
+                              __bf16* __restrict__ y,
+                              unsigned char* __restrict__ mask, long n,
+                              float p, unsigned long long seed) {
+  const float scale = 1.f / (1.f - p);
+  const unsigned int thr = (unsigned int)(p * 4294967296.f);
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    unsigned int u = mix32(seed * 0x9E3779B97F4A7C15ULL + (unsigned long long)i);
+    bool keep = u >= thr;
+    mask[i] = keep;
+    y[i] = keep ? (__bf16)((float)x[i] * scale) : (__bf16)0.f;
+  }
+}
+
+__global__ void dropout_bwd_k(const __bf16* __restrict__ dy,
+                              const unsigned char* __restrict__ mask,
+                              __bf16* __restrict__ dx, long n, float p) {
+  const float scale = 1.f / (1.f - p);
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    dx[i] = mask[i] ? (__bf16)((float)dy[i] * scale) : (__bf16)0.f;
+}
+
+// ---- repack gather (dgrad weight copies, transposes) ----------------------
+
+__global__ void gather_bf16_k(const __bf16* __restrict__ src,
+                              const int* __restrict__ map,
+                              __bf16* __restrict__ dst, long n) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    int m = map[i];
+    dst[i] = m < 0 ? (__bf16)0.f : src[m];
+  }
+}
+
+// ---- pack / unpack (N13: bf16 wire format) --------------------------------
+
+__global__ void pack_bf16_k(const float* __restrict__ src,
+                            __bf16* __restrict__ dst, long n4) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    f32x4 v = *(const f32x4*)(src + i * 4);
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = (__bf16)v[j];
+    *(bf16x4*)(dst + i * 4) = o;
+  }
+}
+
+__global__ void unpack_bf16_k(const __bf16* __restrict__ src,
+                              float* __restrict__ dst, long n4) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    bf16x4 v = *(const bf16x4*)(src + i * 4);
+    f32x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = (float)v[j];
+    *(f32x4*)(dst + i * 4) = o;
+  }
+}
+
+// ---------------------------------------------------------------------------
+
+static inline int ew_grid(long work, int block = 256) {
+  long g = (work + block - 1) / block;
+  if (g > 2048) g = 2048;                 // grid-stride the rest
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+extern "C" {
+
+hipError_t launch_halo_pad(const void* x, void* y, int N, int H, int W,
+                           int C, int t, int b, int l, int r, float fill,
+                           hipStream_t stream) {
+  long total4 = (long)N * (H + t + b) * (W + l + r) * C / 4;
+  hipLaunchKernelGGL(halo_pad_k, dim3(ew_grid(total4)), dim3(256), 0,
+                     stream, (const __bf16*)x, (__bf16*)y, N, H, W, C, t,
+                     b, l, r, fill, total4);
+  return hipGetLastError();
+}
+
+hipError_t launch_relu_fwd(const void* x, void* y, long n,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL(relu_fwd_k, dim3(ew_grid(n / 4)), dim3(256), 0,
+                     stream, (const __bf16*)x, (__bf16*)y, n / 4);
+  return hipGetLastError();
+}
+
+hipError_t launch_relu_bwd(const void* dy, const void* y, void* dx, long n,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL(relu_bwd_k, dim3(ew_grid(n / 4)), dim3(256), 0,
+                     stream, (const __bf16*)dy, (const __bf16*)y,
+                     (__bf16*)dx, n / 4);
+  return hipGetLastError();
+}
+
+hipError_t launch_add_relu(const void* p, const void* q, void* y, long n,
+                           int relu, hipStream_t stream) {
+  hipLaunchKernelGGL(add_relu_k, dim3(ew_grid(n / 4)), dim3(256), 0,
+                     stream, (const __bf16*)p, (const __bf16*)q,
+                     (__bf16*)y, n / 4, relu);
+  return hipGetLastError();
+}
+
+hipError_t launch_dropout_fwd(const void* x, void* y, unsigned char* mask,
+                              long n, float p, unsigned long long seed,
+                              hipStream_t stream) {
+  hipLaunchKernelGGL(dropout_fwd_k, dim3(ew_grid(n)), dim3(256), 0, stream,
+                     (const __bf16*)x, (__bf16*)y, mask, n, p, seed);
+  return hipGetLastError();
+}
+
+hipError_t launch_dropout_bwd(const void* dy, const unsigned char* mask,
+                              void* dx, long n, float p,
+                              hipStream_t stream) {
+  hipLaunchKernelGGL(dropout_bwd_k, dim3(ew_grid(n)), dim3(256), 0, stream,
+                     (const __bf16*)dy, mask, (__bf16*)dx, n, p);
+  return hipGetLastError();
+}
+
+hipError_t launch_gather_bf16(const void* src, const int* map, void* dst,
+                              long n, hipStream_t stream) {
+  hipLaunchKernelGGL(gather_bf16_k, dim3(ew_grid(n)), dim3(256), 0, stream,
+                     (const __bf16*)src, map, (__bf16*)dst, n);
+  return hipGetLastError();
+}
+
+hipError_t launch_pack_bf16(const float* src, void* dst, long n,
+                            hipStream_t stream) {
+  hipLaunchKernelGGL(pack_bf16_k, dim3(ew_grid(n / 4)), dim3(256), 0,
+                     stream, src, (__bf16*)dst, n / 4);
+  return hipGetLastError();
+}
+
+hipError_t launch_unpack_bf16(const void* src, float* dst, long n,
+                              hipStream_t stream) {
+  hipLaunchKernelGGL(unpack_bf16_k, dim3(ew_grid(n / 4)), dim3(256), 0,
+                     stream, (const __bf16*)src, dst, n / 4);
+  return hipGetLastError();
+}
+
+}  // extern "C"

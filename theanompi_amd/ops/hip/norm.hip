@@ -1,0 +1,259 @@
+// Normalisation kernels, NHWC bf16, fp32 statistics (SURVEY.md §2.6
+// N6 LRN cross-map, N7 BatchNorm).
+
+#include "common.hpp"
+
+// ---------------------------------------------------------------------------
+// LRN (cross-map):  scale = k + alpha/size * sum_{|j-c|<=size/2} x_j^2
+//                   y = x * scale^(-beta)
+// ---------------------------------------------------------------------------
+
+__global__ void lrn_fwd_k(const __bf16* __restrict__ x,
+                          __bf16* __restrict__ y,
+                          float* __restrict__ scale, long rows, int C,
+                          int size, float alpha, float beta, float k) {
+  const int half = size / 2;
+  const float a = alpha / size;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long row = i / C;
+    const __bf16* xr = x + row * C;
+    float s = 0.f;
+    int lo = c - half < 0 ? 0 : c - half;
+    int hi = c + half >= C ? C - 1 : c + half;
+    for (int j = lo; j <= hi; ++j) {
+      float v = (float)xr[j];
+      s += v * v;
+    }
+    float sc = k + a * s;
+    scale[i] = sc;
+    y[i] = (__bf16)((float)xr[c] * __powf(sc, -beta));
+  }
+}
+
+// dx_c = dy_c*scale_c^-b - (2*a*b/n) * x_c * sum_{j in win(c)} dy_j*y_j/scale_j
+__global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
+                          const __bf16* __restrict__ x,
+                          const __bf16* __restrict__ y,
+                          const float* __restrict__ scale,
+                          __bf16* __restrict__ dx, long rows, int C,
+                          int size, float alpha, float beta, float k) {
+  const int half = size / 2;
+  const float coef = 2.f * alpha * beta / size;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long row = i / C;
+    long b0 = row * C;
+    float s = 0.f;
+    int lo = c - half < 0 ? 0 : c - half;
+    int hi = c + half >= C ? C - 1 : c + half;
+    for (int j = lo; j <= hi; ++j)
+      s += (float)dy[b0 + j] * (float)y[b0 + j] / scale[b0 + j];
+    float d = (float)dy[i] * __powf(scale[i], -beta)
+              - coef * (float)x[i] * s;
+    dx[i] = (__bf16)d;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// BatchNorm (training): per-channel mean/var over N*H*W rows.
+// Pass 1: partial sums (fp32 atomics into [C] buffers).
+// Pass 2 (host-side tiny kernel): finalize mean/invstd + running stats.
+// Pass 3: normalize (+ optional fused ReLU).
+// ---------------------------------------------------------------------------
+
+__global__ void bn_sums_k(const __bf16* __restrict__ x, float* __restrict__ s1,
+                          float* __restrict__ s2, long rows, int C,
+                          int slices) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int slice = blockIdx.y;
+  if (c >= C) return;
+  float a = 0.f, b = 0.f;
+  for (long r = slice; r < rows; r += slices) {
+    float v = (float)x[r * C + c];
+    a += v;
+    b += v * v;
+  }
+  atomicAdd(&s1[c], a);
+  atomicAdd(&s2[c], b);
+}
+
+__global__ void bn_finalize_k(const float* __restrict__ s1,
+                              const float* __restrict__ s2,
+                              float* __restrict__ mean,
+                              float* __restrict__ invstd,
+                              float* __restrict__ rmean,
+                              float* __restrict__ rvar, long rows, int C,
+                              float momentum, float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float m = s1[c] / rows;
+  float v = s2[c] / rows - m * m;
+  if (v < 0.f) v = 0.f;
+  mean[c] = m;
+  invstd[c] = rsqrtf(v + eps);
+  // running stats (unbiased var like torch)
+  float unb = rows > 1 ? v * rows / (rows - 1) : v;
+  rmean[c] = (1.f - momentum) * rmean[c] + momentum * m;
+  rvar[c] = (1.f - momentum) * rvar[c] + momentum * unb;
+}
+
+__global__ void bn_apply_k(const __bf16* __restrict__ x,
+                           __bf16* __restrict__ y,
+                           const float* __restrict__ mean,
+                           const float* __restrict__ invstd,
+                           const float* __restrict__ gamma,
+                           const float* __restrict__ beta, long rows,
+                           int C, int relu) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float v = ((float)x[i] - mean[c]) * invstd[c] * gamma[c] + beta[c];
+    if (relu && v < 0.f) v = 0.f;
+    y[i] = (__bf16)v;
+  }
+}
+
+// backward pass 1: per-channel  sum(dy)  and  sum(dy * xhat)
+// (dy masked by fused relu via y>0 when relu=1)
+__global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
+                              const __bf16* __restrict__ x,
+                              const __bf16* __restrict__ y,
+                              const float* __restrict__ mean,
+                              const float* __restrict__ invstd,
+                              float* __restrict__ sdy,
+                              float* __restrict__ sdyx, long rows, int C,
+                              int slices, int relu) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  int slice = blockIdx.y;
+  if (c >= C) return;
+  float a = 0.f, b = 0.f;
+  const float m = mean[c], is = invstd[c];
+  for (long r = slice; r < rows; r += slices) {
+    long i = r * C + c;
+    float g = (float)dy[i];
+    if (relu && (float)y[i] <= 0.f) g = 0.f;
+    float xh = ((float)x[i] - m) * is;
+    a += g;
+    b += g * xh;
+  }
+  atomicAdd(&sdy[c], a);
+  atomicAdd(&sdyx[c], b);
+}
+
+// backward pass 2: dx = invstd*gamma*(dy - sdy/M - xhat*sdyx/M)
+// and dgamma = sdyx, dbeta = sdy (copied by one thread per channel).
+__global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
+                               const __bf16* __restrict__ x,
+                               const __bf16* __restrict__ y,
+                               const float* __restrict__ mean,
+                               const float* __restrict__ invstd,
+                               const float* __restrict__ gamma,
+                               const float* __restrict__ sdy,
+                               const float* __restrict__ sdyx,
+                               __bf16* __restrict__ dx, long rows, int C,
+                               int relu) {
+  const float invM = 1.f / rows;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    float g = (float)dy[i];
+    if (relu && (float)y[i] <= 0.f) g = 0.f;
+    float xh = ((float)x[i] - mean[c]) * invstd[c];
+    float d = invstd[c] * gamma[c]
+              * (g - sdy[c] * invM - xh * sdyx[c] * invM);
+    dx[i] = (__bf16)d;
+  }
+}
+
+static inline int ngrid(long work) {
+  long g = (work + 255) / 256;
+  if (g > 4096) g = 4096;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+__global__ void bn_eval_stats_k(const float* __restrict__ rmean,
+                                const float* __restrict__ rvar,
+                                float* __restrict__ mean,
+                                float* __restrict__ invstd, int C,
+                                float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  mean[c] = rmean[c];
+  invstd[c] = rsqrtf(rvar[c] + eps);
+}
+
+extern "C" {
+
+hipError_t launch_lrn_fwd(const void* x, void* y, float* scale, long rows,
+                          int C, int size, float alpha, float beta,
+                          float k, hipStream_t stream) {
+  hipLaunchKernelGGL(lrn_fwd_k, dim3(ngrid(rows * C)), dim3(256), 0,
+                     stream, (const __bf16*)x, (__bf16*)y, scale, rows, C,
+                     size, alpha, beta, k);
+  return hipGetLastError();
+}
+
+hipError_t launch_lrn_bwd(const void* dy, const void* x, const void* y,
+                          const float* scale, void* dx, long rows, int C,
+                          int size, float alpha, float beta, float k,
+                          hipStream_t stream) {
+  hipLaunchKernelGGL(lrn_bwd_k, dim3(ngrid(rows * C)), dim3(256), 0,
+                     stream, (const __bf16*)dy, (const __bf16*)x,
+                     (const __bf16*)y, scale, (__bf16*)dx, rows, C, size,
+                     alpha, beta, k);
+  return hipGetLastError();
+}
+
+hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
+                               float* s2, float* mean, float* invstd,
+                               float* rmean, float* rvar,
+                               const float* gamma, const float* beta,
+                               long rows, int C, float momentum,
+                               float eps, int relu, hipStream_t stream) {
+  int slices = 64;
+  hipLaunchKernelGGL(bn_sums_k, dim3(cdiv(C, 256), slices), dim3(256), 0,
+                     stream, (const __bf16*)x, s1, s2, rows, C, slices);
+  hipLaunchKernelGGL(bn_finalize_k, dim3(cdiv(C, 256)), dim3(256), 0,
+                     stream, s1, s2, mean, invstd, rmean, rvar, rows, C,
+                     momentum, eps);
+  hipLaunchKernelGGL(bn_apply_k, dim3(ngrid(rows * C)), dim3(256), 0,
+                     stream, (const __bf16*)x, (__bf16*)y, mean, invstd,
+                     gamma, beta, rows, C, relu);
+  return hipGetLastError();
+}
+
+hipError_t launch_bn_fwd_eval(const void* x, void* y, float* mean,
+                               float* invstd, const float* rmean,
+                               const float* rvar, const float* gamma,
+                               const float* beta, long rows, int C,
+                               float eps, int relu, hipStream_t stream) {
+  hipLaunchKernelGGL(bn_eval_stats_k, dim3(cdiv(C, 256)), dim3(256), 0,
+                     stream, rmean, rvar, mean, invstd, C, eps);
+  hipLaunchKernelGGL(bn_apply_k, dim3(ngrid(rows * C)), dim3(256), 0,
+                     stream, (const __bf16*)x, (__bf16*)y, mean, invstd,
+                     gamma, beta, rows, C, relu);
+  return hipGetLastError();
+}
+
+hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
+                         const float* mean, const float* invstd,
+                         const float* gamma, float* sdy, float* sdyx,
+                         void* dx, long rows, int C, int relu,
+                         hipStream_t stream) {
+  int slices = 64;
+  hipLaunchKernelGGL(bn_bwd_sums_k, dim3(cdiv(C, 256), slices), dim3(256),
+                     0, stream, (const __bf16*)dy, (const __bf16*)x,
+                     (const __bf16*)y, mean, invstd, sdy, sdyx, rows, C,
+                     slices, relu);
+  hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C)), dim3(256), 0,
+                     stream, (const __bf16*)dy, (const __bf16*)x,
+                     (const __bf16*)y, mean, invstd, gamma, sdy, sdyx,
+                     (__bf16*)dx, rows, C, relu);
+  return hipGetLastError();
+}
+
+}  // extern "C"

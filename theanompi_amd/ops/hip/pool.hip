@@ -1,0 +1,187 @@
+// Pooling kernels, NHWC bf16.  Max pool saves a
+// per-element window argmax (u8) so the backward is a pure gather (no
+// atomics); avg pool backward is analytic.  Inputs arrive halo-padded
+// (max: -inf fill, avg: 0 fill + count includes padding).
+
+#include "common.hpp"
+
+// thread handles 4 channels of one output pixel.
+__global__ void maxpool_fwd_k(const __bf16* __restrict__ xp,
+                              __bf16* __restrict__ y,
+                              unsigned char* __restrict__ am, int N,
+                              int Hp, int Wp, int C, int P, int Q, int kh,
+                              int kw, int sh, int sw, long total4) {
+  for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < total4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
+    int c = (int)(i % C);
+    long pix = i / C;
+    int q = (int)(pix % Q);
+    long tmp = pix / Q;
+    int p = (int)(tmp % P);
+    int n = (int)(tmp / P);
+    const __bf16* base = xp + (((long)n * Hp + p * sh) * Wp + q * sw) * C + c;
+    float best[4] = {-3.0e38f, -3.0e38f, -3.0e38f, -3.0e38f};
+    int arg[4] = {0, 0, 0, 0};
+    for (int r = 0; r < kh; ++r)
+      for (int s = 0; s < kw; ++s) {
+        bf16x4 v = *(const bf16x4*)(base + ((long)r * Wp + s) * C);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float f = (float)v[j];
+          if (f > best[j]) { best[j] = f; arg[j] = r * kw + s; }
+        }
+      }
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      o[j] = (__bf16)best[j];
+      am[i + j] = (unsigned char)arg[j];
+    }
+    *(bf16x4*)(y + i) = o;
+  }
+}
+
+// thread handles one channel of one INPUT pixel; sums dy over the few
+// windows that contain it and selected it.
+__global__ void maxpool_bwd_k(const __bf16* __restrict__ dy,
+                              const unsigned char* __restrict__ am,
+                              __bf16* __restrict__ dx, int N, int H,
+                              int W, int C, int P, int Q, int kh, int kw,
+                              int sh, int sw, int ph, int pw, long total) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long pix = i / C;
+    int w = (int)(pix % W);
+    long tmp = pix / W;
+    int h = (int)(tmp % H);
+    int n = (int)(tmp / H);
+    int hp = h + ph, wp = w + pw;      // position in padded coords
+    float acc = 0.f;
+    int pLo = (hp - kh + sh) / sh; if (pLo < 0) pLo = 0;
+    int pHi = hp / sh; if (pHi > P - 1) pHi = P - 1;
+    int qLo = (wp - kw + sw) / sw; if (qLo < 0) qLo = 0;
+    int qHi = wp / sw; if (qHi > Q - 1) qHi = Q - 1;
+    for (int p = pLo; p <= pHi; ++p)
+      for (int q = qLo; q <= qHi; ++q) {
+        int r = hp - p * sh, s = wp - q * sw;
+        if (r < 0 || r >= kh || s < 0 || s >= kw) continue;
+        long oi = (((long)n * P + p) * Q + q) * C + c;
+        if (am[oi] == (unsigned char)(r * kw + s)) acc += (float)dy[oi];
+      }
+    dx[i] = (__bf16)acc;
+  }
+}
+
+__global__ void avgpool_fwd_k(const __bf16* __restrict__ xp,
+                              __bf16* __restrict__ y, int N, int Hp,
+                              int Wp, int C, int P, int Q, int kh, int kw,
+                              int sh, int sw, long total4) {
+  const float inv = 1.f / (kh * kw);
+  for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < total4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
+    int c = (int)(i % C);
+    long pix = i / C;
+    int q = (int)(pix % Q);
+    long tmp = pix / Q;
+    int p = (int)(tmp % P);
+    int n = (int)(tmp / P);
+    const __bf16* base = xp + (((long)n * Hp + p * sh) * Wp + q * sw) * C + c;
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+    for (int r = 0; r < kh; ++r)
+      for (int s = 0; s < kw; ++s) {
+        bf16x4 v = *(const bf16x4*)(base + ((long)r * Wp + s) * C);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) acc[j] += (float)v[j];
+      }
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = (__bf16)(acc[j] * inv);
+    *(bf16x4*)(y + i) = o;
+  }
+}
+
+__global__ void avgpool_bwd_k(const __bf16* __restrict__ dy,
+                              __bf16* __restrict__ dx, int N, int H,
+                              int W, int C, int P, int Q, int kh, int kw,
+                              int sh, int sw, int ph, int pw, long total) {
+  const float inv = 1.f / (kh * kw);
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = (int)(i % C);
+    long pix = i / C;
+    int w = (int)(pix % W);
+    long tmp = pix / W;
+    int h = (int)(tmp % H);
+    int n = (int)(tmp / H);
+    int hp = h + ph, wp = w + pw;
+    float acc = 0.f;
+    int pLo = (hp - kh + sh) / sh; if (pLo < 0) pLo = 0;
+    int pHi = hp / sh; if (pHi > P - 1) pHi = P - 1;
+    int qLo = (wp - kw + sw) / sw; if (qLo < 0) qLo = 0;
+    int qHi = wp / sw; if (qHi > Q - 1) qHi = Q - 1;
+    for (int p = pLo; p <= pHi; ++p)
+      for (int q = qLo; q <= qHi; ++q) {
+        int r = hp - p * sh, s = wp - q * sw;
+        if (r < 0 || r >= kh || s < 0 || s >= kw) continue;
+        acc += (float)dy[(((long)n * P + p) * Q + q) * C + c];
+      }
+    dx[i] = (__bf16)(acc * inv);
+  }
+}
+
+static inline int pool_grid(long work) {
+  long g = (work + 255) / 256;
+  if (g > 4096) g = 4096;
+  if (g < 1) g = 1;
+  return (int)g;
+}
+
+extern "C" {
+
+hipError_t launch_maxpool_fwd(const void* xp, void* y, unsigned char* am,
+                              int N, int Hp, int Wp, int C, int P, int Q,
+                              int kh, int kw, int sh, int sw,
+                              hipStream_t stream) {
+  long total4 = (long)N * P * Q * C / 4;
+  hipLaunchKernelGGL(maxpool_fwd_k, dim3(pool_grid(total4)), dim3(256), 0,
+                     stream, (const __bf16*)xp, (__bf16*)y, am, N, Hp, Wp,
+                     C, P, Q, kh, kw, sh, sw, total4);
+  return hipGetLastError();
+}
+
+hipError_t launch_maxpool_bwd(const void* dy, const unsigned char* am,
+                              void* dx, int N, int H, int W, int C, int P,
+                              int Q, int kh, int kw, int sh, int sw,
+                              int ph, int pw, hipStream_t stream) {
+  long total = (long)N * H * W * C;
+  hipLaunchKernelGGL(maxpool_bwd_k, dim3(pool_grid(total)), dim3(256), 0,
+                     stream, (const __bf16*)dy, am, (__bf16*)dx, N, H, W,
+                     C, P, Q, kh, kw, sh, sw, ph, pw, total);
+  return hipGetLastError();
+}
+
+hipError_t launch_avgpool_fwd(const void* xp, void* y, int N, int Hp,
+                              int Wp, int C, int P, int Q, int kh, int kw,
+                              int sh, int sw, hipStream_t stream) {
+  long total4 = (long)N * P * Q * C / 4;
+  hipLaunchKernelGGL(avgpool_fwd_k, dim3(pool_grid(total4)), dim3(256), 0,
+                     stream, (const __bf16*)xp, (__bf16*)y, N, Hp, Wp, C,
+                     P, Q, kh, kw, sh, sw, total4);
+  return hipGetLastError();
+}
+
+hipError_t launch_avgpool_bwd(const void* dy, void* dx, int N, int H,
+                              int W, int C, int P, int Q, int kh, int kw,
+                              int sh, int sw, int ph, int pw,
+                              hipStream_t stream) {
+  long total = (long)N * H * W * C;
+  hipLaunchKernelGGL(avgpool_bwd_k, dim3(pool_grid(total)), dim3(256), 0,
+                     stream, (const __bf16*)dy, (__bf16*)dx, N, H, W, C,
+                     P, Q, kh, kw, sh, sw, ph, pw, total);
+  return hipGetLastError();
+}
+
+}  // extern "C"

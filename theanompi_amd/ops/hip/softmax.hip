@@ -1,0 +1,217 @@
+// Fused softmax + cross-entropy and the fused
+// SGD+momentum update (N12), plus the MFMA fragment-layout probe.
+
+#include "common.hpp"
+
+// ---------------------------------------------------------------------------
+// softmax-xent fwd: one 256-thread block per row.  logits bf16 [B][Cp],
+// labels int64, columns >= nCls masked to -inf.  Outputs: probs bf16,
+// per-row loss f32.
+// ---------------------------------------------------------------------------
+
+__global__ void softmax_xent_fwd_k(const __bf16* __restrict__ logits,
+                                   const long* __restrict__ labels,
+                                   __bf16* __restrict__ probs,
+                                   float* __restrict__ loss, int Cp,
+                                   int nCls) {
+  const int b = blockIdx.x;
+  const int tid = threadIdx.x;
+  const __bf16* row = logits + (long)b * Cp;
+  __shared__ float red[8];
+
+  // 1) row max
+  float mx = -3.0e38f;
+  for (int c = tid; c < nCls; c += blockDim.x) {
+    float v = (float)row[c];
+    mx = v > mx ? v : mx;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    float o = __shfl_down(mx, off, 64);
+    mx = o > mx ? o : mx;
+  }
+  if ((tid & 63) == 0) red[tid >> 6] = mx;
+  __syncthreads();
+  if (tid < 8) {
+    float v = (tid < (blockDim.x >> 6)) ? red[tid] : -3.0e38f;
+    for (int off = 4; off > 0; off >>= 1) {
+      float o = __shfl_down(v, off, 64);
+      v = o > v ? o : v;
+    }
+    if (tid == 0) red[0] = v;
+  }
+  __syncthreads();
+  mx = red[0];
+  __syncthreads();
+
+  // 2) exp-sum
+  float sum = 0.f;
+  for (int c = tid; c < nCls; c += blockDim.x)
+    sum += __expf((float)row[c] - mx);
+  for (int off = 32; off > 0; off >>= 1) sum += __shfl_down(sum, off, 64);
+  if ((tid & 63) == 0) red[tid >> 6] = sum;
+  __syncthreads();
+  if (tid < 8) {
+    float v = (tid < (blockDim.x >> 6)) ? red[tid] : 0.f;
+    for (int off = 4; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+    if (tid == 0) red[0] = v;
+  }
+  __syncthreads();
+  sum = red[0];
+  const float inv = 1.f / sum;
+
+  // 3) probs + loss
+  __bf16* prow = probs + (long)b * Cp;
+  for (int c = tid; c < Cp; c += blockDim.x)
+    prow[c] = c < nCls
+                  ? (__bf16)(__expf((float)row[c] - mx) * inv)
+                  : (__bf16)0.f;
+  if (tid == 0) {
+    int lab = (int)labels[b];
+    loss[b] = __logf(sum) - ((float)row[lab] - mx);
+  }
+}
+
+__global__ void softmax_xent_bwd_k(const __bf16* __restrict__ probs,
+                                   const long* __restrict__ labels,
+                                   __bf16* __restrict__ dlogits, int Cp,
+                                   int nCls, float scale) {
+  const int b = blockIdx.x;
+  const int lab = (int)labels[b];
+  const __bf16* prow = probs + (long)b * Cp;
+  __bf16* drow = dlogits + (long)b * Cp;
+  for (int c = threadIdx.x; c < Cp; c += blockDim.x) {
+    float v = c < nCls ? (float)prow[c] : 0.f;
+    if (c == lab) v -= 1.f;
+    drow[c] = (__bf16)(v * scale);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// fused SGD + momentum (flat fp32 master + fp32 grad + fp32 vel; optional
+// bf16 mirror write).  phases: 0 = full, 1 = cdd phase1, 2 = cdd phase2.
+// ---------------------------------------------------------------------------
+
+__global__ void sgd_step_k(float* __restrict__ w,
+                           const float* __restrict__ g,
+                           float* __restrict__ v,
+                           __bf16* __restrict__ out, long n4, float lr,
+                           float mu, float wd, int phase, long n) {
+  for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < n4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
+    f32x4 wv = *(const f32x4*)(w + i);
+    f32x4 gv = *(const f32x4*)(g + i);
+    f32x4 vv = *(const f32x4*)(v + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (phase == 2) {
+        wv[j] += gv[j];                 // gv = exchanged averaged vels
+      } else {
+        float gr = gv[j] + wd * wv[j];
+        vv[j] = mu * vv[j] - lr * gr;
+        if (phase == 0) wv[j] += vv[j];
+      }
+    }
+    if (phase != 2) *(f32x4*)(v + i) = vv;
+    if (phase != 1) {
+      *(f32x4*)(w + i) = wv;
+      if (out) {
+        bf16x4 o;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) o[j] = (__bf16)wv[j];
+        *(bf16x4*)(out + i) = o;
+      }
+    }
+  }
+}
+
+// tail (n % 4) handled scalar
+__global__ void sgd_step_tail_k(float* w, const float* g, float* v,
+                                __bf16* out, long start, long n, float lr,
+                                float mu, float wd, int phase) {
+  long i = start + threadIdx.x;
+  if (i >= n) return;
+  float wv = w[i], gv = g[i], vv = v[i];
+  if (phase == 2) {
+    wv += gv;
+  } else {
+    vv = mu * vv - lr * (gv + wd * wv);
+    if (phase == 0) wv += vv;
+  }
+  if (phase != 2) v[i] = vv;
+  if (phase != 1) {
+    w[i] = wv;
+    if (out) out[i] = (__bf16)wv;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// MFMA fragment-layout probe: one wave computes D = A(16x32) @ B(32x16)
+// with the given LAYOUT; host compares against a CPU reference to pick
+// the layout used by every GEMM kernel (see common.hpp).
+// ---------------------------------------------------------------------------
+
+template <int LAYOUT>
+__global__ void mfma_probe_k(const __bf16* __restrict__ A,  // [16][32]
+                             const __bf16* __restrict__ B,  // [16][32] (n-major!)
+                             float* __restrict__ D) {       // [16][16]
+  int lane = threadIdx.x;
+  if (lane >= 64) return;
+  bf16x8 a = frag_load<LAYOUT>(A + (lane & 15) * 32, lane);
+  bf16x8 b = frag_load<LAYOUT>(B + (lane & 15) * 32, lane);
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r)
+    D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
+}
+
+extern "C" {
+
+hipError_t launch_softmax_xent_fwd(const void* logits, const long* labels,
+                                   void* probs, float* loss, int B,
+                                   int Cp, int nCls, hipStream_t stream) {
+  hipLaunchKernelGGL(softmax_xent_fwd_k, dim3(B), dim3(256), 0, stream,
+                     (const __bf16*)logits, labels, (__bf16*)probs, loss,
+                     Cp, nCls);
+  return hipGetLastError();
+}
+
+hipError_t launch_softmax_xent_bwd(const void* probs, const long* labels,
+                                   void* dlogits, int B, int Cp, int nCls,
+                                   float scale, hipStream_t stream) {
+  hipLaunchKernelGGL(softmax_xent_bwd_k, dim3(B), dim3(256), 0, stream,
+                     (const __bf16*)probs, labels, (__bf16*)dlogits, Cp,
+                     nCls, scale);
+  return hipGetLastError();
+}
+
+hipError_t launch_sgd_step(float* w, const float* g, float* v, void* out,
+                           long n, float lr, float mu, float wd,
+                           int phase, hipStream_t stream) {
+  long n4 = n / 4;
+  if (n4 > 0) {
+    long grid = (n4 + 255) / 256;
+    if (grid > 2048) grid = 2048;
+    hipLaunchKernelGGL(sgd_step_k, dim3((int)grid), dim3(256), 0, stream,
+                       w, g, v, (__bf16*)out, n4, lr, mu, wd, phase, n);
+  }
+  if (n % 4) {
+    hipLaunchKernelGGL(sgd_step_tail_k, dim3(1), dim3(4), 0, stream, w, g,
+                       v, (__bf16*)out, n4 * 4, n, lr, mu, wd, phase);
+  }
+  return hipGetLastError();
+}
+
+hipError_t launch_mfma_probe(const void* A, const void* B, float* D,
+                             int layout, hipStream_t stream) {
+  if (layout == FRAG_CONTIG8)
+    hipLaunchKernelGGL(mfma_probe_k<FRAG_CONTIG8>, dim3(1), dim3(64), 0,
+                       stream, (const __bf16*)A, (const __bf16*)B, D);
+  else
+    hipLaunchKernelGGL(mfma_probe_k<FRAG_HALVES>, dim3(1), dim3(64), 0,
+                       stream, (const __bf16*)A, (const __bf16*)B, D);
+  return hipGetLastError();
+}
+
+}  // extern "C"
