@@ -20,8 +20,8 @@ def q(x):
 
 
 def rel_err(got, ref):
-    got = got.float()
-    ref = ref.float()
+    got = got.detach().float().cpu()
+    ref = ref.detach().float().cpu()
     denom = ref.abs().max().clamp_min(1e-6)
     return ((got - ref).abs().max() / denom).item()
 
@@ -170,7 +170,9 @@ def test_conv_1x1():
 
 
 def test_conv_stem_11x11_s4():
-    conv_case(2, 83, 83, 3, 96, 11, 4, 0)
+    # stem layers run with C_phys = 4 (RGB + one zero pad channel fed by
+    # the loader); the kernel is exercised on all 4 channels here.
+    conv_case(2, 83, 83, 4, 96, 11, 4, 0)
 
 
 def test_conv_5x5_grouped():
