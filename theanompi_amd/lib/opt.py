@@ -95,19 +95,57 @@ class SGDMomentum:
 
     # -- public API ------------------------------------------
 
+    def _batched_gpu(self, phase, grad_of):
+        """One kernel launch for every GPU rec (N12 batched)."""
+        C = _ops.ext()
+        ws, gs, vs, outs, lrs, wds = [], [], [], [], [], []
+        self._empty_bf16 = getattr(self, "_empty_bf16", None)
+        for r in self.recs:
+            if not r.master.is_cuda:
+                g = grad_of(r)
+                if g is not None:
+                    self._torch_step(r, g, phase)
+                continue
+            g = grad_of(r)
+            if g is None:
+                continue
+            if self._empty_bf16 is None:
+                self._empty_bf16 = torch.empty(
+                    0, dtype=torch.bfloat16, device=r.master.device)
+            ws.append(r.master.data.view(-1))
+            gs.append(g.view(-1))
+            vs.append(r.vel.view(-1))
+            outs.append(r.bf16_out.view(-1)
+                        if (r.bf16_out is not None and phase != 1)
+                        else self._empty_bf16)
+            lrs.append(self.lr * r.lr_mult)
+            wds.append(self.weight_decay * r.wd_mult)
+        if ws:
+            C.sgd_step_multi(ws, gs, vs, outs, lrs, wds,
+                             self.momentum, phase)
+
     def step(self):
         """Full fused step: v = mu*v - lr*(g + wd*w); w += v."""
-        for r in self.recs:
-            if r.master.grad is not None:
-                self._dispatch(r, r.master.grad, 0)
+        if self._gpu:
+            self._batched_gpu(0, lambda r: r.master.grad)
+        else:
+            for r in self.recs:
+                if r.master.grad is not None:
+                    self._dispatch(r, r.master.grad, 0)
 
     def phase1(self):
         """cdd pre-exchange: compute velocities only."""
-        for r in self.recs:
-            if r.master.grad is not None:
-                self._dispatch(r, r.master.grad, 1)
+        if self._gpu:
+            self._batched_gpu(1, lambda r: r.master.grad)
+        else:
+            for r in self.recs:
+                if r.master.grad is not None:
+                    self._dispatch(r, r.master.grad, 1)
 
     def phase2(self):
         """cdd post-exchange: apply the (exchanged, averaged) velocities."""
-        for r in self.recs:
-            self._dispatch(r, r.vel, 2)
+        if self._gpu:
+            self._batched_gpu(2, lambda r: r.vel)
+        else:
+            for r in self.recs:
+                self._dispatch(r, r.vel, 2)

@@ -35,7 +35,7 @@ class AlexNet(BaseModel):
         m.append(L.Conv2d((227, 227, 4 if self.gpu else 3), 96, 11,
                           stride=4, pad=0, relu=True,
                           init=("normal", 0.01), bias_init=0.0,
-                          device=dev, batch=B))
+                          device=dev, batch=B, needs_dgrad=False))
         if use_lrn:
             m.append(L.LRN(5, 1e-4, 0.75, 1.0))
         m.append(L.MaxPool2d(3, 2))                      # -> 27x27x96

@@ -83,16 +83,30 @@ class BaseModel:
         self.vels = [r.vel for r in self.recs]
 
     def _refresh_layers(self):
-        """Repack dgrad weight copies after a weight change (GPU)."""
+        """Repack dgrad weight copies after a weight change (GPU).
+        All layers' repacks run as ONE batched gather launch (the
+        per-layer launches were 5.8% of the AlexNet step)."""
         from theanompi_amd import ops as _ops
         if not self._post_step_layers:
             return
         C = _ops.ext()
-        for mod in self._post_step_layers:
-            for m, wdg in zip(mod.plan.build_repack_maps(), mod._wdg):
-                if m is not None:
-                    C.gather_bf16(mod.w_bf16.view(-1), m.view(-1),
-                                  wdg.view(-1))
+        if not hasattr(self, "_repack_desc"):
+            srcs, maps, dsts = [], [], []
+            for mod in self._post_step_layers:
+                for m, wdg in zip(mod.plan.build_repack_maps(), mod._wdg):
+                    if m is not None:
+                        srcs.append(mod.w_bf16.view(-1))
+                        maps.append(m.view(-1))
+                        dsts.append(wdg.view(-1))
+            if srcs:
+                self._repack_desc = C.gather_multi_make(srcs, maps, dsts)
+                self._repack_nseg = len(srcs)
+                self._repack_maxn = max(d.numel() for d in dsts)
+            else:
+                self._repack_desc = None
+        if self._repack_desc is not None:
+            C.gather_multi_run(self._repack_desc, self._repack_nseg,
+                               self._repack_maxn)
 
     def refresh_all(self):
         """Resync every derived weight copy after an out-of-band change to

@@ -33,7 +33,7 @@ class Cifar10_model(BaseModel):
         m = nn.Sequential()
         m.append(L.Conv2d((32, 32, 4 if self.gpu else 3), 32, 5, stride=1,
                           pad=2, relu=True, init=("normal", 0.05),
-                          device=dev, batch=B))
+                          device=dev, batch=B, needs_dgrad=False))
         m.append(L.MaxPool2d(3, 2, 1))                   # 16x16x32
         m.append(L.Conv2d((16, 16, 32), 32, 5, stride=1, pad=2, relu=True,
                           init=("normal", 0.05), device=dev, batch=B))

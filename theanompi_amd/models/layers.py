@@ -330,8 +330,9 @@ class Conv2d(_Base):
 
     def __init__(self, in_shape, K, R, S=None, stride=1, pad=0, groups=1,
                  relu=False, init=("he",), bias_init=0.0, device="cpu",
-                 batch=None):
+                 batch=None, needs_dgrad=True):
         super().__init__()
+        self.needs_dgrad = needs_dgrad
         S = S if S is not None else R
         stride = (stride, stride) if isinstance(stride, int) else stride
         pad = (pad, pad) if isinstance(pad, int) else pad
@@ -366,6 +367,9 @@ class Conv2d(_Base):
             self.register_buffer("w_bf16",
                                  flat.to(torch.bfloat16).contiguous())
             self.b_f32 = self.bias.data  # alias used by kernels
+            if not needs_dgrad:      # first layer: dx never computed
+                self.plan.dgrad_plans = []
+                self.plan._repack_maps = []
             maps = self.plan.build_repack_maps()
             self._wdg = []
             for m in maps:

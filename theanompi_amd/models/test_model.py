@@ -32,7 +32,8 @@ class TestModel(BaseModel):
         B = self.batch_size
         m = nn.Sequential()
         m.append(L.Conv2d((16, 16, 4 if self.gpu else 3), 16, 3, stride=1,
-                          pad=1, relu=True, device=dev, batch=B))
+                          pad=1, relu=True, device=dev, batch=B,
+                          needs_dgrad=False))
         m.append(L.MaxPool2d(2, 2))                      # 8x8x16
         m.append(L.Conv2d((8, 8, 16), 32, 3, stride=1, pad=1, relu=True,
                           device=dev, batch=B))

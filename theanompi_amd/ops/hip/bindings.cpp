@@ -96,7 +96,27 @@ hipError_t launch_sgd_step(float*, const float*, float*, void*, long, float,
                            float, float, int, hipStream_t);
 hipError_t launch_mfma_probe(const void*, const void*, float*, int,
                              hipStream_t);
+hipError_t launch_sgd_multi(const void*, int, long, float, int,
+                            hipStream_t);
+hipError_t launch_gather_multi(const void*, int, long, hipStream_t);
 }
+
+struct SgdSegH {
+  float* w;
+  const float* g;
+  float* v;
+  void* out;
+  long n4;
+  long ntail;
+  float lr, wd;
+};
+
+struct GatherSegH {
+  const void* src;
+  const int* map;
+  void* dst;
+  long n;
+};
 
 // ---- helpers ---------------------------------------------------------------
 
@@ -557,6 +577,65 @@ static void sgd_step(torch::Tensor w, torch::Tensor g, torch::Tensor v,
 
 static long mfma_layout() { return detect_layout(); }
 
+// one launch for the whole parameter set (per optimizer phase)
+static void sgd_step_multi(std::vector<torch::Tensor> ws,
+                           std::vector<torch::Tensor> gs,
+                           std::vector<torch::Tensor> vs,
+                           std::vector<torch::Tensor> outs,
+                           std::vector<double> lrs,
+                           std::vector<double> wds, double mu,
+                           long phase) {
+  int n = (int)ws.size();
+  TORCH_CHECK(n > 0);
+  std::vector<SgdSegH> segs(n);
+  long max_n4 = 1;
+  for (int i = 0; i < n; ++i) {
+    check_f32(ws[i], "w");
+    long nn = ws[i].numel();
+    segs[i] = {ws[i].data_ptr<float>(), gs[i].data_ptr<float>(),
+               vs[i].data_ptr<float>(),
+               outs[i].numel() ? outs[i].data_ptr() : nullptr,
+               nn / 4, nn, (float)lrs[i], (float)wds[i]};
+    if (nn / 4 > max_n4) max_n4 = nn / 4;
+  }
+  auto buf = torch::empty({(long)(n * sizeof(SgdSegH))},
+                          torch::TensorOptions()
+                              .dtype(torch::kUInt8)
+                              .device(ws[0].device()));
+  hipMemcpyAsync(buf.data_ptr(), segs.data(), n * sizeof(SgdSegH),
+                 hipMemcpyHostToDevice, cur_stream());
+  CHECK_HIP(launch_sgd_multi(buf.data_ptr(), n, max_n4, (float)mu,
+                             (int)phase, cur_stream()));
+  // segs is host memory; hipMemcpyAsync from pageable memory completes
+  // synchronously w.r.t. the host, so letting it go out of scope is safe.
+}
+
+// static repack set: build the descriptor buffer once...
+static torch::Tensor gather_multi_make(std::vector<torch::Tensor> srcs,
+                                       std::vector<torch::Tensor> maps,
+                                       std::vector<torch::Tensor> dsts) {
+  int n = (int)srcs.size();
+  std::vector<GatherSegH> segs(n);
+  for (int i = 0; i < n; ++i) {
+    segs[i] = {srcs[i].data_ptr(), maps[i].data_ptr<int>(),
+               dsts[i].data_ptr(), dsts[i].numel()};
+  }
+  auto buf = torch::empty({(long)(n * sizeof(GatherSegH))},
+                          torch::TensorOptions()
+                              .dtype(torch::kUInt8)
+                              .device(srcs[0].device()));
+  hipMemcpyAsync(buf.data_ptr(), segs.data(), n * sizeof(GatherSegH),
+                 hipMemcpyHostToDevice, cur_stream());
+  hipStreamSynchronize(cur_stream());
+  return buf;
+}
+
+// ...then replay with one kernel launch per refresh
+static void gather_multi_run(torch::Tensor desc, long nseg, long max_n) {
+  CHECK_HIP(launch_gather_multi(desc.data_ptr(), (int)nseg, max_n,
+                                cur_stream()));
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("halo_pad", &halo_pad_sym);
   m.def("halo_pad_lrtb", &halo_pad);
@@ -584,4 +663,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("sgd_step", &sgd_step);
   m.def("mfma_layout", &mfma_layout);
+  m.def("sgd_step_multi", &sgd_step_multi);
+  m.def("gather_multi_make", &gather_multi_make);
+  m.def("gather_multi_run", &gather_multi_run);
 }

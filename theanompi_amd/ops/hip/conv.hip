@@ -305,18 +305,28 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
 
 // ---------------------------------------------------------------------------
 // bias grad: db[c] += sum over rows of DY[row][c]
+// Vectorized: each thread owns 8 consecutive channels (16-B loads) of a
+// row slice; consecutive threads cover consecutive channel groups so a
+// wave reads contiguous 1-KiB spans of each row.
 // ---------------------------------------------------------------------------
 
 __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
                              float* __restrict__ db, long rows, int C,
                              int slices) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  int slice = blockIdx.y;
-  if (c >= C) return;
-  float s = 0.f;
-  for (long r = slice; r < rows; r += slices)
-    s += (float)dy[r * C + c];
-  atomicAdd(&db[c], s);
+  const int c8n = C / 8;
+  long t = blockIdx.x * (long)blockDim.x + threadIdx.x;
+  if (t >= (long)c8n * slices) return;
+  int c8 = (int)(t % c8n);
+  int slice = (int)(t / c8n);
+  float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  const __bf16* base = dy + c8 * 8;
+  for (long r = slice; r < rows; r += slices) {
+    bf16x8 v = *(const bf16x8*)(base + r * C);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) s[j] += (float)v[j];
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) atomicAdd(&db[c8 * 8 + j], s[j]);
 }
 
 // ---------------------------------------------------------------------------

@@ -12,23 +12,47 @@ __global__ void lrn_fwd_k(const __bf16* __restrict__ x,
                           __bf16* __restrict__ y,
                           float* __restrict__ scale, long rows, int C,
                           int size, float alpha, float beta, float k) {
+  // one thread per (row, 4-channel group); the +-4 window around the
+  // group is covered by three aligned 8-B loads (size <= 9).
   const int half = size / 2;
   const float a = alpha / size;
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
-       i += (long)gridDim.x * blockDim.x) {
-    int c = (int)(i % C);
-    long row = i / C;
+  const int C4 = C / 4;
+  const long total = rows * C4;
+  for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < total;
+       i4 += (long)gridDim.x * blockDim.x) {
+    int c4 = (int)(i4 % C4);
+    long row = i4 / C4;
     const __bf16* xr = x + row * C;
-    float s = 0.f;
-    int lo = c - half < 0 ? 0 : c - half;
-    int hi = c + half >= C ? C - 1 : c + half;
-    for (int j = lo; j <= hi; ++j) {
-      float v = (float)xr[j];
-      s += v * v;
+    float v[12];
+#pragma unroll
+    for (int b = 0; b < 3; ++b) {
+      int cb = c4 + b - 1;
+      if (cb >= 0 && cb < C4) {
+        bf16x4 t = *(const bf16x4*)(xr + cb * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) v[b * 4 + j] = (float)t[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) v[b * 4 + j] = 0.f;
+      }
     }
-    float sc = k + a * s;
-    scale[i] = sc;
-    y[i] = (__bf16)((float)xr[c] * __powf(sc, -beta));
+    long o = row * C + c4 * 4;
+    bf16x4 out;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int c = c4 * 4 + j;
+      int lo = c - half < 0 ? -c4 * 4 : j - half;        // window rel j
+      int hi = c + half >= C ? C - c4 * 4 - 1 : j + half;
+      float s = 0.f;
+      for (int w = lo; w <= hi; ++w) {
+        float f = v[4 + w];
+        s += f * f;
+      }
+      float sc = k + a * s;
+      scale[o + j] = sc;
+      out[j] = (__bf16)(v[4 + j] * __powf(sc, -beta));
+    }
+    *(bf16x4*)(y + o) = out;
   }
 }
 
@@ -41,19 +65,51 @@ __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
                           int size, float alpha, float beta, float k) {
   const int half = size / 2;
   const float coef = 2.f * alpha * beta / size;
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
-       i += (long)gridDim.x * blockDim.x) {
-    int c = (int)(i % C);
-    long row = i / C;
+  const int C4 = C / 4;
+  const long total = rows * C4;
+  for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < total;
+       i4 += (long)gridDim.x * blockDim.x) {
+    int c4 = (int)(i4 % C4);
+    long row = i4 / C4;
     long b0 = row * C;
-    float s = 0.f;
-    int lo = c - half < 0 ? 0 : c - half;
-    int hi = c + half >= C ? C - 1 : c + half;
-    for (int j = lo; j <= hi; ++j)
-      s += (float)dy[b0 + j] * (float)y[b0 + j] / scale[b0 + j];
-    float d = (float)dy[i] * __powf(scale[i], -beta)
-              - coef * (float)x[i] * s;
-    dx[i] = (__bf16)d;
+    // t[w] = dy*y/scale over the 12-channel neighbourhood
+    float t[12], xv[12], dyv[4];
+#pragma unroll
+    for (int b = 0; b < 3; ++b) {
+      int cb = c4 + b - 1;
+      if (cb >= 0 && cb < C4) {
+        bf16x4 d = *(const bf16x4*)(dy + b0 + cb * 4);
+        bf16x4 yy = *(const bf16x4*)(y + b0 + cb * 4);
+        bf16x4 xx = *(const bf16x4*)(x + b0 + cb * 4);
+        f32x4 sc = *(const f32x4*)(scale + b0 + cb * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          t[b * 4 + j] = (float)d[j] * (float)yy[j] / sc[j];
+          xv[b * 4 + j] = (float)xx[j];
+          if (b == 1) dyv[j] = (float)d[j];
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          t[b * 4 + j] = 0.f;
+          xv[b * 4 + j] = 0.f;
+        }
+      }
+    }
+    const f32x4 scc = *(const f32x4*)(scale + b0 + c4 * 4);
+    long o = b0 + c4 * 4;
+    bf16x4 out;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int c = c4 * 4 + j;
+      int lo = c - half < 0 ? -c4 * 4 : j - half;
+      int hi = c + half >= C ? C - c4 * 4 - 1 : j + half;
+      float s = 0.f;
+      for (int w = lo; w <= hi; ++w) s += t[4 + w];
+      float d = dyv[j] * __powf(scc[j], -beta) - coef * xv[j] * s;
+      out[j] = (__bf16)d;
+    }
+    *(bf16x4*)(dx + o) = out;
   }
 }
 
@@ -191,7 +247,7 @@ extern "C" {
 hipError_t launch_lrn_fwd(const void* x, void* y, float* scale, long rows,
                           int C, int size, float alpha, float beta,
                           float k, hipStream_t stream) {
-  hipLaunchKernelGGL(lrn_fwd_k, dim3(ngrid(rows * C)), dim3(256), 0,
+  hipLaunchKernelGGL(lrn_fwd_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
                      stream, (const __bf16*)x, (__bf16*)y, scale, rows, C,
                      size, alpha, beta, k);
   return hipGetLastError();
@@ -201,7 +257,7 @@ hipError_t launch_lrn_bwd(const void* dy, const void* x, const void* y,
                           const float* scale, void* dx, long rows, int C,
                           int size, float alpha, float beta, float k,
                           hipStream_t stream) {
-  hipLaunchKernelGGL(lrn_bwd_k, dim3(ngrid(rows * C)), dim3(256), 0,
+  hipLaunchKernelGGL(lrn_bwd_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
                      stream, (const __bf16*)dy, (const __bf16*)x,
                      (const __bf16*)y, scale, (__bf16*)dx, rows, C, size,
                      alpha, beta, k);

@@ -42,15 +42,17 @@ __global__ void maxpool_fwd_k(const __bf16* __restrict__ xp,
   }
 }
 
-// thread handles one channel of one INPUT pixel; sums dy over the few
+// thread handles 4 channels of one INPUT pixel; sums dy over the few
 // windows that contain it and selected it.
 __global__ void maxpool_bwd_k(const __bf16* __restrict__ dy,
                               const unsigned char* __restrict__ am,
                               __bf16* __restrict__ dx, int N, int H,
                               int W, int C, int P, int Q, int kh, int kw,
-                              int sh, int sw, int ph, int pw, long total) {
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
-       i += (long)gridDim.x * blockDim.x) {
+                              int sh, int sw, int ph, int pw,
+                              long total4) {
+  for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < total4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
     int c = (int)(i % C);
     long pix = i / C;
     int w = (int)(pix % W);
@@ -58,19 +60,27 @@ __global__ void maxpool_bwd_k(const __bf16* __restrict__ dy,
     int h = (int)(tmp % H);
     int n = (int)(tmp / H);
     int hp = h + ph, wp = w + pw;      // position in padded coords
-    float acc = 0.f;
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
     int pLo = (hp - kh + sh) / sh; if (pLo < 0) pLo = 0;
     int pHi = hp / sh; if (pHi > P - 1) pHi = P - 1;
     int qLo = (wp - kw + sw) / sw; if (qLo < 0) qLo = 0;
     int qHi = wp / sw; if (qHi > Q - 1) qHi = Q - 1;
     for (int p = pLo; p <= pHi; ++p)
       for (int q = qLo; q <= qHi; ++q) {
-        int r = hp - p * sh, s = wp - q * sw;
-        if (r < 0 || r >= kh || s < 0 || s >= kw) continue;
+        int r = hp - p * sh, s2 = wp - q * sw;
+        if (r < 0 || r >= kh || s2 < 0 || s2 >= kw) continue;
         long oi = (((long)n * P + p) * Q + q) * C + c;
-        if (am[oi] == (unsigned char)(r * kw + s)) acc += (float)dy[oi];
+        unsigned char code = (unsigned char)(r * kw + s2);
+        bf16x4 g = *(const bf16x4*)(dy + oi);
+        const unsigned int m4 = *(const unsigned int*)(am + oi);
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (((m4 >> (8 * j)) & 0xff) == code) acc[j] += (float)g[j];
       }
-    dx[i] = (__bf16)acc;
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) o[j] = (__bf16)acc[j];
+    *(bf16x4*)(dx + i) = o;
   }
 }
 
@@ -156,10 +166,10 @@ hipError_t launch_maxpool_bwd(const void* dy, const unsigned char* am,
                               void* dx, int N, int H, int W, int C, int P,
                               int Q, int kh, int kw, int sh, int sw,
                               int ph, int pw, hipStream_t stream) {
-  long total = (long)N * H * W * C;
-  hipLaunchKernelGGL(maxpool_bwd_k, dim3(pool_grid(total)), dim3(256), 0,
+  long total4 = (long)N * H * W * C / 4;
+  hipLaunchKernelGGL(maxpool_bwd_k, dim3(pool_grid(total4)), dim3(256), 0,
                      stream, (const __bf16*)dy, am, (__bf16*)dx, N, H, W,
-                     C, P, Q, kh, kw, sh, sw, ph, pw, total);
+                     C, P, Q, kh, kw, sh, sw, ph, pw, total4);
   return hipGetLastError();
 }
 

@@ -215,3 +215,109 @@ hipError_t launch_mfma_probe(const void* A, const void* B, float* D,
 }
 
 }  // extern "C"
+
+// ---------------------------------------------------------------------------
+// batched variants: one launch covers every parameter tensor (the
+// per-tensor launches were 11% of the AlexNet step — see
+// profiles/alexnet_r01_baseline.md).
+// ---------------------------------------------------------------------------
+
+struct SgdSeg {
+  float* w;
+  const float* g;
+  float* v;
+  __bf16* out;
+  long n4;        // vector chunks
+  long ntail;     // total n (scalar tail handled by last thread)
+  float lr, wd;
+};
+
+__global__ void sgd_multi_k(const SgdSeg* __restrict__ segs, float mu,
+                            int phase) {
+  const SgdSeg s = segs[blockIdx.y];
+  for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < s.n4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
+    f32x4 wv = *(const f32x4*)(s.w + i);
+    f32x4 gv = *(const f32x4*)(s.g + i);
+    f32x4 vv = *(const f32x4*)(s.v + i);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      if (phase == 2) {
+        wv[j] += gv[j];
+      } else {
+        float gr = gv[j] + s.wd * wv[j];
+        vv[j] = mu * vv[j] - s.lr * gr;
+        if (phase == 0) wv[j] += vv[j];
+      }
+    }
+    if (phase != 2) *(f32x4*)(s.v + i) = vv;
+    if (phase != 1) {
+      *(f32x4*)(s.w + i) = wv;
+      if (s.out) {
+        bf16x4 o;
+#pragma unroll
+        for (int j = 0; j < 4; ++j) o[j] = (__bf16)wv[j];
+        *(bf16x4*)(s.out + i) = o;
+      }
+    }
+  }
+  // scalar tail
+  if (blockIdx.x == 0 && threadIdx.x < (s.ntail - s.n4 * 4)) {
+    long i = s.n4 * 4 + threadIdx.x;
+    float wv = s.w[i], gv = s.g[i], vv = s.v[i];
+    if (phase == 2) {
+      wv += gv;
+    } else {
+      vv = mu * vv - s.lr * (gv + s.wd * wv);
+      if (phase == 0) wv += vv;
+    }
+    if (phase != 2) s.v[i] = vv;
+    if (phase != 1) {
+      s.w[i] = wv;
+      if (s.out) s.out[i] = (__bf16)wv;
+    }
+  }
+}
+
+struct GatherSeg {
+  const __bf16* src;
+  const int* map;
+  __bf16* dst;
+  long n;
+};
+
+__global__ void gather_multi_k(const GatherSeg* __restrict__ segs) {
+  const GatherSeg s = segs[blockIdx.y];
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < s.n;
+       i += (long)gridDim.x * blockDim.x) {
+    int m = s.map[i];
+    s.dst[i] = m < 0 ? (__bf16)0.f : s.src[m];
+  }
+}
+
+extern "C" {
+
+hipError_t launch_sgd_multi(const void* segs_dev, int nseg, long max_n4,
+                            float mu, int phase, hipStream_t stream) {
+  long gx = (max_n4 + 255) / 256;
+  if (gx > 512) gx = 512;
+  if (gx < 1) gx = 1;
+  dim3 grid((unsigned)gx, nseg);
+  hipLaunchKernelGGL(sgd_multi_k, grid, dim3(256), 0, stream,
+                     (const SgdSeg*)segs_dev, mu, phase);
+  return hipGetLastError();
+}
+
+hipError_t launch_gather_multi(const void* segs_dev, int nseg, long max_n,
+                               hipStream_t stream) {
+  long gx = (max_n + 255) / 256;
+  if (gx > 512) gx = 512;
+  if (gx < 1) gx = 1;
+  dim3 grid((unsigned)gx, nseg);
+  hipLaunchKernelGGL(gather_multi_k, grid, dim3(256), 0, stream,
+                     (const GatherSeg*)segs_dev);
+  return hipGetLastError();
+}
+
+}  // extern "C"
