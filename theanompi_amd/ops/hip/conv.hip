@@ -311,22 +311,38 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
 // ---------------------------------------------------------------------------
 
 __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
-                             float* __restrict__ db, long rows, int C,
-                             int slices) {
+                             float* __restrict__ db, long rows, int C) {
+  // block = 32 channel-groups x 8 row-lanes; LDS-reduce the row lanes so
+  // each block issues at most 256 atomics (the naive version serialized
+  // on ~1M atomics over a few hundred addresses).
   const int c8n = C / 8;
-  long t = blockIdx.x * (long)blockDim.x + threadIdx.x;
-  if (t >= (long)c8n * slices) return;
-  int c8 = (int)(t % c8n);
-  int slice = (int)(t / c8n);
+  const int cl = threadIdx.x & 31;          // channel-group lane
+  const int rl = threadIdx.x >> 5;          // row lane (0..7)
+  const int c8 = blockIdx.x * 32 + cl;
+  const long sl = (long)blockIdx.y * 8 + rl;
+  const long nsl = (long)gridDim.y * 8;
   float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  const __bf16* base = dy + c8 * 8;
-  for (long r = slice; r < rows; r += slices) {
-    bf16x8 v = *(const bf16x8*)(base + r * C);
+  if (c8 < c8n) {
+    const __bf16* base = dy + c8 * 8;
+    for (long r = sl; r < rows; r += nsl) {
+      bf16x8 v = *(const bf16x8*)(base + r * C);
 #pragma unroll
-    for (int j = 0; j < 8; ++j) s[j] += (float)v[j];
+      for (int j = 0; j < 8; ++j) s[j] += (float)v[j];
+    }
   }
+  __shared__ float red[8][32][8];
 #pragma unroll
-  for (int j = 0; j < 8; ++j) atomicAdd(&db[c8 * 8 + j], s[j]);
+  for (int j = 0; j < 8; ++j) red[rl][cl][j] = s[j];
+  __syncthreads();
+  if (rl == 0 && c8 < c8n) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float a = 0.f;
+#pragma unroll
+      for (int t = 0; t < 8; ++t) a += red[t][cl][j];
+      atomicAdd(&db[c8 * 8 + j], a);
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -375,10 +391,15 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
 
 hipError_t launch_bias_wgrad(const void* dy, float* db, long rows, int C,
                              hipStream_t stream) {
-  int slices = 64;
-  dim3 grid(cdiv(C, 256), slices);
+  int c8n = C / 8;
+  int gx = cdiv(c8n, 32);
+  long gy = (rows + 7) / 8;
+  long target = 1024 / gx;                 // ~1024 blocks total
+  if (gy > target) gy = target;
+  if (gy < 1) gy = 1;
+  dim3 grid(gx, (unsigned)gy);
   hipLaunchKernelGGL(bias_wgrad_k, grid, dim3(256), 0, stream,
-                     (const __bf16*)dy, db, rows, C, slices);
+                     (const __bf16*)dy, db, rows, C);
   return hipGetLastError();
 }
 

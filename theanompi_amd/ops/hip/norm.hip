@@ -49,8 +49,13 @@ __global__ void lrn_fwd_k(const __bf16* __restrict__ x,
         s += f * f;
       }
       float sc = k + a * s;
-      scale[o + j] = sc;
-      out[j] = (__bf16)(v[4 + j] * __powf(sc, -beta));
+      float inv = 1.f / sc;                // saved: bwd needs 1/scale
+      scale[o + j] = inv;
+      // inv^beta; beta = 0.75 fast path: sqrt(inv)*sqrt(sqrt(inv))
+      float pw = (beta == 0.75f)
+                     ? __fsqrt_rn(inv) * __fsqrt_rn(__fsqrt_rn(inv))
+                     : __powf(inv, beta);
+      out[j] = (__bf16)(v[4 + j] * pw);
     }
     *(bf16x4*)(y + o) = out;
   }
@@ -81,10 +86,10 @@ __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
         bf16x4 d = *(const bf16x4*)(dy + b0 + cb * 4);
         bf16x4 yy = *(const bf16x4*)(y + b0 + cb * 4);
         bf16x4 xx = *(const bf16x4*)(x + b0 + cb * 4);
-        f32x4 sc = *(const f32x4*)(scale + b0 + cb * 4);
+        f32x4 sc = *(const f32x4*)(scale + b0 + cb * 4);  // = 1/scale
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-          t[b * 4 + j] = (float)d[j] * (float)yy[j] / sc[j];
+          t[b * 4 + j] = (float)d[j] * (float)yy[j] * sc[j];
           xv[b * 4 + j] = (float)xx[j];
           if (b == 1) dyv[j] = (float)d[j];
         }
@@ -106,7 +111,11 @@ __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
       int hi = c + half >= C ? C - c4 * 4 - 1 : j + half;
       float s = 0.f;
       for (int w = lo; w <= hi; ++w) s += t[4 + w];
-      float d = dyv[j] * __powf(scc[j], -beta) - coef * xv[j] * s;
+      float inv = scc[j];                  // 1/scale
+      float pw = (beta == 0.75f)
+                     ? __fsqrt_rn(inv) * __fsqrt_rn(__fsqrt_rn(inv))
+                     : __powf(inv, beta);
+      float d = dyv[j] * pw - coef * xv[j] * s;
       out[j] = (__bf16)d;
     }
     *(bf16x4*)(dx + o) = out;
