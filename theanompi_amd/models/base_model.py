@@ -92,9 +92,14 @@ class BaseModel:
         C = _ops.ext()
         if not hasattr(self, "_repack_desc"):
             srcs, maps, dsts = [], [], []
+            self._transpose_layers = []
             for mod in self._post_step_layers:
                 for m, wdg in zip(mod.plan.build_repack_maps(), mod._wdg):
-                    if m is not None:
+                    if m is None:
+                        continue
+                    if isinstance(m, str):
+                        self._transpose_layers.append((mod, wdg))
+                    else:
                         srcs.append(mod.w_bf16.view(-1))
                         maps.append(m.view(-1))
                         dsts.append(wdg.view(-1))
@@ -107,6 +112,11 @@ class BaseModel:
         if self._repack_desc is not None:
             C.gather_multi_run(self._repack_desc, self._repack_nseg,
                                self._repack_maxn)
+        for mod, wdg in self._transpose_layers:
+            plan = mod.plan
+            C.transpose_bf16(mod.w_bf16.view(-1), wdg.view(-1),
+                             plan.Kgp, plan.Cg, plan.Kdf,
+                             wdg.shape[-1])
 
     def refresh_all(self):
         """Resync every derived weight copy after an out-of-band change to

@@ -209,6 +209,15 @@ class ConvPlan:
         dgrad class; used by _C.gather_bf16 after every optimizer step."""
         if self._repack_maps is not None:
             return self._repack_maps
+        # FC / 1x1 s1 conv: the repack is a plain transpose of the flat
+        # weight — use the LDS transpose kernel instead of an int32 map
+        # (the map gather was 5% of the AlexNet step and its numpy
+        # construction dominates model build time for big FCs).
+        live = [p for p in self.dgrad_plans if p is not None]
+        if (self.R == 1 and self.S == 1 and self.G == 1
+                and len(live) == 1 and len(self.dgrad_plans) == 1):
+            self._repack_maps = ["transpose"]
+            return self._repack_maps
         maps = []
         G, Kgp, Kg = self.G, self.Kgp, self.Kg
         src_strideG = Kgp * self.Kdf
@@ -371,14 +380,7 @@ class Conv2d(_Base):
                 self.plan.dgrad_plans = []
                 self.plan._repack_maps = []
             maps = self.plan.build_repack_maps()
-            self._wdg = []
-            for m in maps:
-                if m is None:
-                    self._wdg.append(None)
-                else:
-                    self._wdg.append(torch.zeros(m.shape,
-                                                 dtype=torch.bfloat16,
-                                                 device=self.device_t))
+            self._wdg = self._alloc_wdg(maps)
             self.out_shape = (self.plan.P, self.plan.Q, self.plan.Kphys)
             self.out_logical_c = K
             self.post_step()          # fill w_bf16 + dgrad repack copies
@@ -422,16 +424,42 @@ class Conv2d(_Base):
                          plan.Cg)
         return dx
 
+    def _alloc_wdg(self, maps):
+        out = []
+        for i, m in enumerate(maps):
+            if m is None:
+                out.append(None)
+            elif isinstance(m, str):          # "transpose"
+                plan2 = self.plan.dgrad_plans[i]
+                out.append(torch.zeros(
+                    (self.plan.G, self.plan.Cg, plan2["Kdf2"]),
+                    dtype=torch.bfloat16, device=self.device_t))
+            else:
+                out.append(torch.zeros(m.shape, dtype=torch.bfloat16,
+                                       device=self.device_t))
+        return out
+
+    def repack_dgrad(self, C):
+        """Refresh the dgrad weight copies from w_bf16."""
+        for m, wdg in zip(self.plan.build_repack_maps(), self._wdg):
+            if m is None:
+                continue
+            if isinstance(m, str):            # transpose path
+                plan = self.plan
+                C.transpose_bf16(self.w_bf16.view(-1), wdg.view(-1),
+                                 plan.Kgp, plan.Cg, plan.Kdf,
+                                 wdg.shape[-1])
+            else:
+                C.gather_bf16(self.w_bf16.view(-1), m.view(-1),
+                              wdg.view(-1))
+
     def post_step(self):
         if not self.gpu:
             return
         C = _ops.ext()
         self.w_bf16.copy_(self.weight.data)   # fp32 -> bf16 cast copy
         self.b_f32 = self.bias.data
-        for m, wdg in zip(self.plan.build_repack_maps(), self._wdg):
-            if m is not None:
-                C.gather_bf16(self.w_bf16.view(-1), m.view(-1),
-                              wdg.view(-1))
+        self.repack_dgrad(C)
 
     def logical_params(self):
         if self.gpu:
@@ -496,9 +524,7 @@ class Linear(_Base):
                                  flat.to(torch.bfloat16).contiguous())
             self.b_f32 = self.bias.data
             maps = self.plan.build_repack_maps()
-            self._wdg = [None if m is None else
-                         torch.zeros(m.shape, dtype=torch.bfloat16,
-                                     device=self.device_t) for m in maps]
+            self._wdg = self._alloc_wdg(maps)
             self.in_phys = inp
             self.out_features_phys = self.plan.Kphys
             self.post_step()          # fill w_bf16 + dgrad repack copies

@@ -59,6 +59,8 @@ hipError_t launch_dropout_bwd(const void*, const unsigned char*, void*, long,
                               float, hipStream_t);
 hipError_t launch_gather_bf16(const void*, const int*, void*, long,
                               hipStream_t);
+hipError_t launch_transpose_bf16(const void*, void*, int, int, long, long,
+                                 hipStream_t);
 hipError_t launch_pack_bf16(const float*, void*, long, hipStream_t);
 hipError_t launch_unpack_bf16(const void*, float*, long, hipStream_t);
 hipError_t launch_maxpool_fwd(const void*, void*, unsigned char*, int, int,
@@ -400,6 +402,16 @@ static void gather_bf16(torch::Tensor src, torch::Tensor map,
                                dst.data_ptr(), dst.numel(), cur_stream()));
 }
 
+static void transpose_bf16(torch::Tensor src, torch::Tensor dst,
+                           long rows, long cols, long sstride,
+                           long dstride) {
+  check_bf16(src, "src");
+  check_bf16(dst, "dst");
+  CHECK_HIP(launch_transpose_bf16(src.data_ptr(), dst.data_ptr(),
+                                  (int)rows, (int)cols, sstride, dstride,
+                                  cur_stream()));
+}
+
 static void pack_bf16(torch::Tensor src, torch::Tensor dst) {
   check_f32(src, "src");
   check_bf16(dst, "dst");
@@ -649,6 +661,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("gather_bf16", &gather_bf16);
+  m.def("transpose_bf16", &transpose_bf16);
   m.def("pack_bf16", &pack_bf16);
   m.def("unpack_bf16", &unpack_bf16);
   m.def("maxpool_fwd", &maxpool_fwd);

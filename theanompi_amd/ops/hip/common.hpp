@@ -58,6 +58,38 @@ DEV bf16x8 frag_load(const __bf16* row_ptr, int lane) {
   }
 }
 
+// Fragment load with the content XOR-swizzle used by the glds-staged
+// GEMM tiles: LDS slot (row, p8) holds the source 8-element piece
+// p8 ^ (row & 7), so reads XOR the same pattern back (guide T2/rule 21:
+// swizzle on the SOURCE address + the read, LDS destination linear).
+template <int LAYOUT>
+DEV bf16x8 frag_load_swz(const __bf16* row_start, int row, int ks,
+                         int lane) {
+  const int x8 = (row & 7) * 8;
+  if constexpr (LAYOUT == FRAG_CONTIG8) {
+    int koff = ks + (lane >> 4) * 8;
+    return *(const bf16x8*)(row_start + (koff ^ x8));
+  } else {
+    int k1 = (ks + (lane >> 4) * 4) ^ x8;
+    int k2 = (ks + (lane >> 4) * 4 + 16) ^ x8;
+    bf16x4 lo = *(const bf16x4*)(row_start + k1);
+    bf16x4 hi = *(const bf16x4*)(row_start + k2);
+    bf16x8 r;
+    r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
+    r[4] = hi[0]; r[5] = hi[1]; r[6] = hi[2]; r[7] = hi[3];
+    return r;
+  }
+}
+
+// async global->LDS copy, 16 B per lane; LDS destination is
+// wave-uniform base + lane*16 (hardware adds the lane offset).
+typedef __attribute__((address_space(1))) const unsigned int as1_uint;
+typedef __attribute__((address_space(3))) unsigned int as3_uint;
+DEV void glds16(const void* gsrc, void* lds_base) {
+  __builtin_amdgcn_global_load_lds((as1_uint*)gsrc, (as3_uint*)lds_base,
+                                   16, 0, 0);
+}
+
 #define HIP_KERNEL_ASSERT_OK(expr)                                          \
   do {                                                                      \
     hipError_t _e = (expr);                                                 \

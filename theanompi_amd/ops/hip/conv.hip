@@ -43,15 +43,18 @@ struct GatherGemmArgs {
 template <int BM, int BN, int LAYOUT>
 __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
   constexpr int BK = 64;
-  constexpr int LDK = BK + 8;            // row pad: 16 B
   constexpr int WAVES_M = (BN == 128) ? 2 : 4;
   constexpr int WAVES_N = BN / 64;       // 2 or 1
   constexpr int TM = BM / WAVES_M;       // 64 or 32
   constexpr int FM = TM / 16;            // 4 or 2
   constexpr int FN = 64 / 16;            // wave tile N = 64 always
+  constexpr int AI = BM / 32;            // glds instrs per wave (A tile)
+  constexpr int BI = BN / 32;
 
-  __shared__ __align__(16) __bf16 As[2][BM][LDK];
-  __shared__ __align__(16) __bf16 Bs[2][BN][LDK];
+  // lane-linear tiles (glds writes base + lane*16B); content is
+  // XOR-swizzled by source-piece permutation, undone at the frag reads.
+  __shared__ __align__(16) __bf16 As[2][BM][BK];
+  __shared__ __align__(16) __bf16 Bs[2][BN][BK];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -62,46 +65,45 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
   const long gInOff = (long)g * a.groupInOff;
   const int bn0 = nTile * BN;
 
-  // --- per-thread staging descriptors (fixed across K) --------------------
-  constexpr int A_PIECES = (BM * 8) / 256;   // 16-B pieces per thread
-  constexpr int B_PIECES = (BN * 8) / 256;
-  long aRowBase[A_PIECES];
-  int aRow[A_PIECES];
-  const int p8 = (tid & 7) * 8;              // element offset of my piece
+  // --- per-lane glds descriptors (fixed across K) --------------------------
+  long aRowBase[AI];
+  int aK8x8[AI];                 // swizzled in-row byte... element offset
 #pragma unroll
-  for (int t = 0; t < A_PIECES; ++t) {
-    int row = (tid >> 3) + t * 32;
+  for (int i = 0; i < AI; ++i) {
+    int piece = (wave * AI + i) * 64 + lane;
+    int row = piece >> 3;
+    int k8 = (piece & 7) ^ (row & 7);          // source-piece swizzle
     int m = mTile * BM + row;
-    if (m >= a.M) m = a.M - 1;               // clamped rows are discarded
+    if (m >= a.M) m = a.M - 1;                 // clamped rows discarded
     int q = m % a.Q;
     int tmp = m / a.Q;
     int p = tmp % a.P;
     int n = tmp / a.P;
-    aRow[t] = row;
-    aRowBase[t] = (long)n * a.inNS + (long)p * a.inPS + (long)q * a.inQS
+    aRowBase[i] = (long)n * a.inNS + (long)p * a.inPS + (long)q * a.inQS
                   + gInOff;
+    aK8x8[i] = k8 * 8;
   }
-  const __bf16* bRowPtr[B_PIECES];
+  const __bf16* bSrc[BI];
 #pragma unroll
-  for (int t = 0; t < B_PIECES; ++t) {
-    int row = (tid >> 3) + t * 32;
+  for (int i = 0; i < BI; ++i) {
+    int piece = (wave * BI + i) * 64 + lane;
+    int row = piece >> 3;
+    int k8 = (piece & 7) ^ (row & 7);
     int n = bn0 + row;
-    if (n >= a.BNtot) n = a.BNtot - 1;       // discarded at C-write
-    bRowPtr[t] = Wg + (long)n * a.Kdf;
+    if (n >= a.BNtot) n = a.BNtot - 1;         // discarded at C-write
+    bSrc[i] = Wg + (long)n * a.Kdf + k8 * 8;
   }
 
-  // --- staging ------------------------------------------------------------
   auto stage = [&](int buf, int k0) {
 #pragma unroll
-    for (int t = 0; t < A_PIECES; ++t) {
-      int off = a.table[(k0 + p8) >> 4] + (p8 & 15);
-      pack16 v = *(const pack16*)(a.X + aRowBase[t] + off);
-      *(pack16*)&As[buf][aRow[t]][p8] = v;
+    for (int i = 0; i < AI; ++i) {
+      int off = a.table[(k0 + aK8x8[i]) >> 4] + (aK8x8[i] & 15);
+      glds16(a.X + aRowBase[i] + off,
+             &As[buf][(wave * AI + i) * 8][0]);
     }
 #pragma unroll
-    for (int t = 0; t < B_PIECES; ++t) {
-      pack16 v = *(const pack16*)(bRowPtr[t] + k0 + p8);
-      *(pack16*)&Bs[buf][(tid >> 3) + t * 32][p8] = v;
+    for (int i = 0; i < BI; ++i) {
+      glds16(bSrc[i] + k0, &Bs[buf][(wave * BI + i) * 8][0]);
     }
   };
 
@@ -128,11 +130,15 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
     for (int ks = 0; ks < BK; ks += 32) {
       bf16x8 af[FM], bf[FN];
 #pragma unroll
-      for (int i = 0; i < FM; ++i)
-        af[i] = frag_load<LAYOUT>(&As[buf][m0 + i * 16 + lrow][ks], lane);
+      for (int i = 0; i < FM; ++i) {
+        int row = m0 + i * 16 + lrow;
+        af[i] = frag_load_swz<LAYOUT>(&As[buf][row][0], row, ks, lane);
+      }
 #pragma unroll
-      for (int j = 0; j < FN; ++j)
-        bf[j] = frag_load<LAYOUT>(&Bs[buf][n0 + j * 16 + lrow][ks], lane);
+      for (int j = 0; j < FN; ++j) {
+        int row = n0 + j * 16 + lrow;
+        bf[j] = frag_load_swz<LAYOUT>(&Bs[buf][row][0], row, ks, lane);
+      }
 #pragma unroll
       for (int i = 0; i < FM; ++i)
 #pragma unroll

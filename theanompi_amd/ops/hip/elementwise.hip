@@ -128,6 +128,58 @@ __global__ void gather_bf16_k(const __bf16* __restrict__ src,
   }
 }
 
+// ---- LDS-tiled transpose (FC / 1x1-conv dgrad weight repack) --------------
+// dst[c][k] = src[k][c] for k < rows, c < cols; strides in elements.
+// 64(col)x32(row) tiles; 8-element vector loads, column gathers from LDS.
+
+__global__ void transpose_bf16_k(const __bf16* __restrict__ src,
+                                 __bf16* __restrict__ dst, int rows,
+                                 int cols, long sstride, long dstride) {
+  __shared__ __bf16 tile[32][72];    // +8 pad kills column-read conflicts
+  const int r0 = blockIdx.y * 32;
+  const int c0 = blockIdx.x * 64;
+  const int t = threadIdx.x;
+  {
+    const int row = t >> 3;          // 0..31
+    const int cp = (t & 7) * 8;      // 0..56
+    int r = r0 + row;
+    if (r < rows) {
+      int c = c0 + cp;
+      // clamp the 8-wide load inside the row
+      if (c + 8 <= cols) {
+        *(bf16x8*)&tile[row][cp] =
+            *(const bf16x8*)(src + (long)r * sstride + c);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          tile[row][cp + j] = (c + j < cols)
+                                  ? src[(long)r * sstride + c + j]
+                                  : (__bf16)0.f;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) tile[row][cp + j] = (__bf16)0.f;
+    }
+  }
+  __syncthreads();
+  {
+    const int col = t >> 2;          // 0..63 (source col = dst row)
+    const int rp = (t & 3) * 8;      // 0..24
+    int c = c0 + col;
+    if (c >= cols) return;
+    bf16x8 v;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) v[j] = tile[rp + j][col];
+    long r = r0 + rp;
+    if (r + 8 <= rows) {
+      *(bf16x8*)(dst + (long)c * dstride + r) = v;
+    } else {
+      for (int j = 0; j < 8 && r + j < rows; ++j)
+        dst[(long)c * dstride + r + j] = v[j];
+    }
+  }
+}
+
 // ---- pack / unpack (N13: bf16 wire format) --------------------------------
 
 __global__ void pack_bf16_k(const float* __restrict__ src,
@@ -211,6 +263,16 @@ hipError_t launch_dropout_bwd(const void* dy, const unsigned char* mask,
                               hipStream_t stream) {
   hipLaunchKernelGGL(dropout_bwd_k, dim3(ew_grid(n)), dim3(256), 0, stream,
                      (const __bf16*)dy, mask, (__bf16*)dx, n, p);
+  return hipGetLastError();
+}
+
+hipError_t launch_transpose_bf16(const void* src, void* dst, int rows,
+                                 int cols, long sstride, long dstride,
+                                 hipStream_t stream) {
+  dim3 grid(cdiv(cols, 64), cdiv(rows, 32));
+  hipLaunchKernelGGL(transpose_bf16_k, grid, dim3(256), 0, stream,
+                     (const __bf16*)src, (__bf16*)dst, rows, cols,
+                     sstride, dstride);
   return hipGetLastError();
 }
 
