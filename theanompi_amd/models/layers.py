@@ -547,6 +547,8 @@ class Linear(_Base):
 
     _dgrad = Conv2d._dgrad
     post_step = Conv2d.post_step
+    _alloc_wdg = Conv2d._alloc_wdg
+    repack_dgrad = Conv2d.repack_dgrad
 
     def logical_params(self):
         if self.gpu:
