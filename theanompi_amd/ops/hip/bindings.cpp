@@ -101,6 +101,7 @@ hipError_t launch_mfma_probe(const void*, const void*, float*, int,
 hipError_t launch_sgd_multi(const void*, int, long, float, int,
                             hipStream_t);
 hipError_t launch_gather_multi(const void*, int, long, hipStream_t);
+hipError_t launch_tr_probe(const int*, float*, int, hipStream_t);
 }
 
 struct SgdSegH {
@@ -589,6 +590,16 @@ static void sgd_step(torch::Tensor w, torch::Tensor g, torch::Tensor v,
 
 static long mfma_layout() { return detect_layout(); }
 
+static torch::Tensor tr_probe(torch::Tensor offs, long fill_mode) {
+  TORCH_CHECK(offs.scalar_type() == torch::kInt32 && offs.numel() == 64);
+  auto out = torch::zeros({64, 4}, torch::TensorOptions()
+                                       .dtype(torch::kFloat32)
+                                       .device(offs.device()));
+  CHECK_HIP(launch_tr_probe(offs.data_ptr<int>(), out.data_ptr<float>(),
+                            (int)fill_mode, cur_stream()));
+  return out;
+}
+
 // one launch for the whole parameter set (per optimizer phase)
 static void sgd_step_multi(std::vector<torch::Tensor> ws,
                            std::vector<torch::Tensor> gs,
@@ -676,6 +687,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("sgd_step", &sgd_step);
   m.def("mfma_layout", &mfma_layout);
+  m.def("tr_probe", &tr_probe);
   m.def("sgd_step_multi", &sgd_step_multi);
   m.def("gather_multi_make", &gather_multi_make);
   m.def("gather_multi_run", &gather_multi_run);

@@ -81,6 +81,36 @@ DEV bf16x8 frag_load_swz(const __bf16* row_start, int row, int ks,
   }
 }
 
+// Transpose-read fragment load for the wgrad TN tiles.
+//
+// ds_read_b64_tr_b16 semantics on gfx950 (empirically probed, see
+// tests/gpu tr probe): within each 16-lane group, lane l's element j
+// comes from lds[align8B(addr of lane 16g+4j+((l&15)>>2)) + (l&3)].
+// With a LINEAR [pix][64ch] row-major LDS image whose 8-element slots
+// are content-swizzled by slot^ (pix&7) (same convention as the glds
+// GEMM tiles), lane s passes
+//   A_s = pix_s*64 + ((ch0 + 4*(s&3)) ^ ((pix_s&7)*8)),
+//   pix_s = ks + 8*(s>>4) + ((s&15)>>2)
+// and two reads (pix_s, pix_s+4) give lane l the 8 k-major elements
+// (ch = ch0 + (l&15), pix = ks + (l>>4)*8 + 0..7) of an MFMA fragment
+// in FRAG_CONTIG8 order.
+typedef __attribute__((address_space(3))) bf16x4 lds_bf16x4;
+
+DEV bf16x8 frag_load_tr_swz(const __bf16* tile, int ks, int ch0,
+                            int lane) {
+  const int pix1 = ks + 8 * (lane >> 4) + ((lane & 15) >> 2);
+  const int pix2 = pix1 + 4;
+  const int co = ch0 + 4 * (lane & 3);
+  const __bf16* a1 = tile + pix1 * 64 + (co ^ ((pix1 & 7) * 8));
+  const __bf16* a2 = tile + pix2 * 64 + (co ^ ((pix2 & 7) * 8));
+  bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf16x4*)a1);
+  bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf16x4*)a2);
+  bf16x8 r;
+  r[0] = lo[0]; r[1] = lo[1]; r[2] = lo[2]; r[3] = lo[3];
+  r[4] = hi[0]; r[5] = hi[1]; r[6] = hi[2]; r[7] = hi[3];
+  return r;
+}
+
 // async global->LDS copy, 16 B per lane; LDS destination is
 // wave-uniform base + lane*16 (hardware adds the lane offset).
 typedef __attribute__((address_space(1))) const unsigned int as1_uint;

@@ -196,13 +196,17 @@ struct WgradArgs {
 
 template <int LAYOUT>
 __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
+  // TN GEMM over pixels.  Both operands stage LINEARLY ([pix][64ch],
+  // 16-B pieces, content-swizzled like the fwd tiles) and the MFMA
+  // k-major fragments come from ds_read_b64_tr_b16 (hardware transpose)
+  // -- no scalar LDS traffic anywhere (the v1 scalar transpose staging
+  // was 23% of the AlexNet step).
   constexpr int BMK = 64;   // kout tile
   constexpr int BNF = 64;   // fd tile
   constexpr int BKP = 64;   // pixel chunk
-  constexpr int LDK = BKP + 8;
 
-  __shared__ __align__(16) __bf16 DYt[2][BMK][LDK];  // [kout][pix]
-  __shared__ __align__(16) __bf16 XT[2][BNF][LDK];   // [fd][pix]
+  __shared__ __align__(16) __bf16 DYt[2][BKP][64];  // [pix][kout]
+  __shared__ __align__(16) __bf16 XT[2][BKP][64];   // [pix][fd]
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -216,45 +220,53 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   const long gInOff = (long)g * a.groupInOff;
   const long dyTotal = (long)a.M * a.Kphys;
 
-  const int pixl = tid >> 3;                // 0..31 (piece row within 32)
-  const int p8 = (tid & 7) * 8;
+  // piece descriptors: piece = (wave*4 + i)*64 + lane; row (pix) =
+  // piece>>3, source slot k8 = (piece&7) ^ (row&7)
+  int rowOf[4], k8Of[4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    int piece = (wave * 4 + i) * 64 + lane;
+    rowOf[i] = piece >> 3;
+    k8Of[i] = ((piece & 7) ^ ((piece >> 3) & 7)) * 8;
+  }
 
   auto stage = [&](int buf, int pix0) {
+    const bool interior = (pix0 + BKP <= a.M);
 #pragma unroll
-    for (int t = 0; t < 2; ++t) {           // 64 pix rows / 32 per pass
-      int pix = pixl + t * 32;
-      int pa = pix0 + pix;
-      // ---- dy piece (transposed write), zero beyond M ----
-      pack16 v;
-      if (pa < a.M) {
-        long addr = (long)pa * a.Kphys + g * a.Kgp + kout0 + p8;
-        if (addr + 8 > dyTotal) addr = dyTotal - 8;
-        v = *(const pack16*)(a.DY + addr);
-      } else {
-        v.u[0] = v.u[1] = v.u[2] = v.u[3] = 0;
-      }
-      const __bf16* e = (const __bf16*)&v;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) DYt[buf][p8 + i][pix] = e[i];
-      // ---- x gather piece (transposed write) ----
-      int pc = pa < a.M ? pa : a.M - 1;
+    for (int i = 0; i < 4; ++i) {
+      const int row = rowOf[i];
+      const int pa = pix0 + row;
+      const int pc = pa < a.M ? pa : a.M - 1;
       int q = pc % a.Q;
       int tmp = pc / a.Q;
       int p = tmp % a.P;
       int n = tmp / a.P;
-      long base = (long)n * a.inNS + (long)p * a.inPS + (long)q * a.inQS
-                  + gInOff;
-      int off = a.table[(fd0 + p8) >> 4] + (p8 & 15);
-      pack16 w = *(const pack16*)(a.X + base + off);
-      const __bf16* f = (const __bf16*)&w;
-#pragma unroll
-      for (int i = 0; i < 8; ++i) XT[buf][p8 + i][pix] = f[i];
+      long xbase = (long)n * a.inNS + (long)p * a.inPS
+                   + (long)q * a.inQS + gInOff;
+      int xoff = a.table[(fd0 + k8Of[i]) >> 4] + (k8Of[i] & 15);
+      long dyaddr = (long)pc * a.Kphys + g * a.Kgp + kout0 + k8Of[i];
+      if (dyaddr + 8 > dyTotal) dyaddr = dyTotal - 8;
+      if (interior) {
+        glds16(a.DY + dyaddr, &DYt[buf][(wave * 4 + i) * 8][0]);
+        glds16(a.X + xbase + xoff, &XT[buf][(wave * 4 + i) * 8][0]);
+      } else {
+        // tail chunk: dy rows past M must be EXACT zeros
+        pack16 v;
+        if (pa < a.M) {
+          v = *(const pack16*)(a.DY + dyaddr);
+        } else {
+          v.u[0] = v.u[1] = v.u[2] = v.u[3] = 0;
+        }
+        // LDS slot = source piece XOR (row&7)  (same as the glds path)
+        *(pack16*)&DYt[buf][row][k8Of[i] ^ ((row & 7) * 8)] = v;
+        pack16 w = *(const pack16*)(a.X + xbase + xoff);
+        *(pack16*)&XT[buf][row][k8Of[i] ^ ((row & 7) * 8)] = w;
+      }
     }
   };
 
   const int wm = wave >> 1, wn = wave & 1;  // 2x2 waves, 32x32 each
   const int m0 = wm * 32, n0 = wn * 32;
-  const int lrow = lane & 15;
 
   f32x4 acc[2][2];
 #pragma unroll
@@ -274,19 +286,21 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
       const int buf = t & 1;
 #pragma unroll
       for (int ks = 0; ks < BKP; ks += 32) {
-        bf16x8 af[2], bf[2];
+        bf16x8 af[2], bfr[2];
 #pragma unroll
         for (int i = 0; i < 2; ++i)
-          af[i] = frag_load<LAYOUT>(&DYt[buf][m0 + i * 16 + lrow][ks], lane);
+          af[i] = frag_load_tr_swz(&DYt[buf][0][0], ks, m0 + i * 16,
+                                   lane);
 #pragma unroll
         for (int j = 0; j < 2; ++j)
-          bf[j] = frag_load<LAYOUT>(&XT[buf][n0 + j * 16 + lrow][ks], lane);
+          bfr[j] = frag_load_tr_swz(&XT[buf][0][0], ks, n0 + j * 16,
+                                    lane);
 #pragma unroll
         for (int i = 0; i < 2; ++i)
 #pragma unroll
           for (int j = 0; j < 2; ++j)
             acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af[i], bf[j], acc[i][j], 0, 0, 0);
+                af[i], bfr[j], acc[i][j], 0, 0, 0);
       }
       __syncthreads();
     }
@@ -309,11 +323,6 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   }
 }
 
-// ---------------------------------------------------------------------------
-// bias grad: db[c] += sum over rows of DY[row][c]
-// Vectorized: each thread owns 8 consecutive channels (16-B loads) of a
-// row slice; consecutive threads cover consecutive channel groups so a
-// wave reads contiguous 1-KiB spans of each row.
 // ---------------------------------------------------------------------------
 
 __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,

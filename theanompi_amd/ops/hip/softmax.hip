@@ -166,7 +166,34 @@ __global__ void mfma_probe_k(const __bf16* __restrict__ A,  // [16][32]
     D[((lane >> 4) * 4 + r) * 16 + (lane & 15)] = acc[r];
 }
 
+// ds_read_b64_tr_b16 semantics probe: LDS filled with buf[i] = i (two
+// halves so bf16 stays exact); each lane reads at a host-given element
+// offset and dumps its 4 returned values.
+typedef __attribute__((address_space(3))) bf16x4 lds_bf16x4;
+
+__global__ void tr_probe_k(const int* __restrict__ offs,
+                           float* __restrict__ out, int fill_mode) {
+  __shared__ __align__(16) __bf16 buf[512];
+  int t = threadIdx.x;
+  for (int i = t; i < 512; i += 64)
+    buf[i] = (__bf16)(float)(fill_mode == 0 ? (i & 0xff) : (i >> 8));
+  __syncthreads();
+  if (t < 64) {
+    bf16x4 v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+        (lds_bf16x4*)&buf[offs[t]]);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) out[t * 4 + j] = (float)v[j];
+  }
+}
+
 extern "C" {
+
+hipError_t launch_tr_probe(const int* offs, float* out, int fill_mode,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL(tr_probe_k, dim3(1), dim3(64), 0, stream, offs, out,
+                     fill_mode);
+  return hipGetLastError();
+}
 
 hipError_t launch_softmax_xent_fwd(const void* logits, const long* labels,
                                    void* probs, float* loss, int B,
