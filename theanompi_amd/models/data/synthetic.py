@@ -73,6 +73,9 @@ class SyntheticData:
 
     def _finish(self, x_u8, y):
         """uint8 NHWC -> compute dtype, scaled to ~[-1, 1)."""
+        if self.gpu and self.dtype == torch.bfloat16:
+            from theanompi_amd import ops as _ops
+            return _ops.ext().u8_finish(x_u8, self.n_logical_c), y
         x = x_u8.to(self.dtype)
         x = x * (1.0 / 128.0) - 1.0
         x[..., self.n_logical_c:] = 0       # keep pad channels exactly 0

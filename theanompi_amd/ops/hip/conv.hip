@@ -339,10 +339,24 @@ __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
   float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   if (c8 < c8n) {
     const __bf16* base = dy + c8 * 8;
-    for (long r = sl; r < rows; r += nsl) {
-      bf16x8 v = *(const bf16x8*)(base + r * C);
+    // 4 consecutive rows per iteration: 4 independent 16-B loads in
+    // flight per thread (the 1-deep version was latency-bound)
+    long r4 = sl * 4;
+    for (; r4 + 4 <= rows; r4 += nsl * 4) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) s[j] += (float)v[j];
+      for (int u = 0; u < 4; ++u) {
+        bf16x8 v = *(const bf16x8*)(base + (r4 + u) * C);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s[j] += (float)v[j];
+      }
+    }
+    // tail rows (rows % 4): once, by the first row-lane of y-block 0
+    if (blockIdx.y == 0 && rl == 0) {
+      for (long r = rows & ~3L; r < rows; ++r) {
+        bf16x8 v = *(const bf16x8*)(base + r * C);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) s[j] += (float)v[j];
+      }
     }
   }
   __shared__ float red[8][32][8];
@@ -408,8 +422,8 @@ hipError_t launch_bias_wgrad(const void* dy, float* db, long rows, int C,
                              hipStream_t stream) {
   int c8n = C / 8;
   int gx = cdiv(c8n, 32);
-  long gy = (rows + 7) / 8;
-  long target = 1024 / gx;                 // ~1024 blocks total
+  long gy = (rows + 31) / 32;              // 4 rows per row-lane pass
+  long target = 512 / gx;                  // bounded atomics per channel
   if (gy > target) gy = target;
   if (gy < 1) gy = 1;
   dim3 grid(gx, (unsigned)gy);

@@ -128,6 +128,25 @@ __global__ void gather_bf16_k(const __bf16* __restrict__ src,
   }
 }
 
+// ---- fused loader finish: uint8 NHWC -> bf16, x/128 - 1, zero pad chans
+__global__ void u8_finish_k(const unsigned char* __restrict__ src,
+                            __bf16* __restrict__ dst, long total4, int C,
+                            int nLogC) {
+  for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < total4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
+    int c = (int)(i % C);
+    unsigned int u = *(const unsigned int*)(src + i);
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float v = (float)((u >> (8 * j)) & 0xff) * (1.f / 128.f) - 1.f;
+      o[j] = (c + j < nLogC) ? (__bf16)v : (__bf16)0.f;
+    }
+    *(bf16x4*)(dst + i) = o;
+  }
+}
+
 // ---- LDS-tiled transpose (FC / 1x1-conv dgrad weight repack) --------------
 // dst[c][k] = src[k][c] for k < rows, c < cols; strides in elements.
 // 64(col)x32(row) tiles; 8-element vector loads, column gathers from LDS.
@@ -263,6 +282,15 @@ hipError_t launch_dropout_bwd(const void* dy, const unsigned char* mask,
                               hipStream_t stream) {
   hipLaunchKernelGGL(dropout_bwd_k, dim3(ew_grid(n)), dim3(256), 0, stream,
                      (const __bf16*)dy, mask, (__bf16*)dx, n, p);
+  return hipGetLastError();
+}
+
+hipError_t launch_u8_finish(const unsigned char* src, void* dst,
+                            long total, int C, int nLogC,
+                            hipStream_t stream) {
+  long t4 = total / 4;
+  hipLaunchKernelGGL(u8_finish_k, dim3(ew_grid(t4)), dim3(256), 0, stream,
+                     src, (__bf16*)dst, t4, C, nLogC);
   return hipGetLastError();
 }
 
