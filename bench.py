@@ -41,7 +41,9 @@ def main():
     ap.add_argument("--sync-type", default="cdd",
                     choices=["cdd", "avg", "grad"])
     ap.add_argument("--strategy", default="rccl_bf16")
-    ap.add_argument("--bucket-mb", type=float, default=64)
+    ap.add_argument("--bucket-mb", type=float, default=25)
+    ap.add_argument("--no-overlap", action="store_true",
+                    help="disable comm/backward overlap in grad mode")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -70,12 +72,19 @@ def main():
                  "n_iters": 16})
     model.compile_iter_fns(args.sync_type)
 
-    from theanompi_amd.lib.exchanger import BSP_Exchanger
-    exchanger = BSP_Exchanger(
-        {"sync_type": args.sync_type, "exchange_strategy":
-             (args.strategy if on_gpu else "ar"),
-         "size": world, "rank": rank, "bucket_mb": args.bucket_mb},
-        model.params, model.vels)
+    from theanompi_amd.lib.exchanger import (BSP_Exchanger,
+                                             OverlappedGradExchanger)
+    overlapped = None
+    if args.sync_type == "grad" and not args.no_overlap:
+        overlapped = OverlappedGradExchanger(
+            model.recs, world, bucket_mb=args.bucket_mb)
+        exchanger = None
+    else:
+        exchanger = BSP_Exchanger(
+            {"sync_type": args.sync_type, "exchange_strategy":
+                 (args.strategy if on_gpu else "ar"),
+             "size": world, "rank": rank, "bucket_mb": args.bucket_mb},
+            model.params, model.vels)
 
     def barrier():
         if dist_on:
@@ -88,9 +97,15 @@ def main():
 
     def step(i):
         model.train_iter(i % model.data.n_batch_train)
-        if args.sync_type == "grad":
+        if overlapped is not None:
+            overlapped.finish()
+        elif args.sync_type == "grad":
             exchanger.set_grad_buffers(model.opt.grads())
-        exchanger.exchange()
+            exchanger.exchange()
+        elif args.sync_type == "cdd" and hasattr(model, "vel_arena"):
+            exchanger.exchange_flat_inplace([model.vel_arena])
+        else:
+            exchanger.exchange()
         model.apply_update()
 
     for i in range(args.warmup):

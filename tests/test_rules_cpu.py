@@ -115,3 +115,29 @@ def test_gosgd_cpu(tmp_path):
     assert rule.wait() == [0, 0]
     """
     _run_session(tmp_path, body)
+
+
+def test_bsp_overlap_equivalence(tmp_path):
+    """Overlapped grad exchange == plain grad exchange (same final
+    params), 2 CPU workers over gloo."""
+    outs = {}
+    for tag, overlap in (("ov", 1), ("plain", 0)):
+        d = tmp_path / tag
+        d.mkdir()
+        body = f"""
+        import os
+        os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+        from theanompi_amd import BSP
+        rule = BSP()
+        rule.init(devices=['cpu', 'cpu'],
+                  modelfile='theanompi_amd.models.test_model',
+                  modelclass='TestModel',
+                  epochs=1, sync_type='grad', overlap={overlap},
+                  seed=77, dump_dir={str(d)!r})
+        assert rule.wait() == [0, 0]
+        """
+        _run_session(tmp_path, body)
+        outs[tag] = np.load(os.path.join(str(d), "params_rank0.npz"))
+    for k in outs["ov"].files:
+        np.testing.assert_allclose(outs["ov"][k], outs["plain"][k],
+                                   atol=1e-5)

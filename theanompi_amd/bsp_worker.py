@@ -12,7 +12,8 @@ import torch
 import torch.distributed as dist
 
 from theanompi_amd.lib.base import GPUProcess, _env_opt
-from theanompi_amd.lib.exchanger import BSP_Exchanger
+from theanompi_amd.lib.exchanger import (BSP_Exchanger,
+                                         OverlappedGradExchanger)
 from theanompi_amd.lib.recorder import Recorder
 
 
@@ -30,10 +31,16 @@ def run():
     save_freq = int(_env_opt("save_freq", 0))
 
     model.compile_iter_fns(sync_type)
+    overlap = (sync_type == "grad"
+               and _env_opt("overlap", "1") not in ("0", "false"))
     config = {"sync_type": sync_type, "exchange_strategy": strategy,
               "size": proc.size, "rank": proc.rank,
-              "bucket_mb": float(_env_opt("bucket_mb", 64))}
-    exchanger = BSP_Exchanger(config, model.params, model.vels)
+              "bucket_mb": float(_env_opt("bucket_mb", 25))}
+    if overlap:
+        exchanger = OverlappedGradExchanger(
+            model.recs, proc.size, bucket_mb=config["bucket_mb"])
+    else:
+        exchanger = BSP_Exchanger(config, model.params, model.vels)
 
     proc.broadcast_params(model)
     rec = Recorder({"rank": proc.rank, "size": proc.size,
@@ -49,9 +56,15 @@ def run():
         for i in range(n_batches):
             rec.start()
             model.train_iter(i, rec)
-            if sync_type == "grad":
+            if overlap:
+                exchanger.finish(rec)
+            elif sync_type == "grad":
                 exchanger.set_grad_buffers(model.opt.grads())
-            exchanger.exchange(rec)
+                exchanger.exchange(rec)
+            elif sync_type == "cdd" and hasattr(model, "vel_arena"):
+                exchanger.exchange_flat_inplace([model.vel_arena], rec)
+            else:
+                exchanger.exchange(rec)
             rec.start()
             model.apply_update()
             rec.end("calc")

@@ -112,6 +112,9 @@ class GPUProcess:
         """Reflectively import the model class and instantiate it with the
         config dict of the reference's model contract (SURVEY.md §1 L4->L2)."""
         import importlib
+        seed = _env_opt("seed")
+        if seed is not None:
+            torch.manual_seed(int(seed))
         modelfile = modelfile or os.environ["THEANOMPI_AMD_MODELFILE"]
         modelclass = modelclass or os.environ["THEANOMPI_AMD_MODELCLASS"]
         mod = importlib.import_module(modelfile)

@@ -117,6 +117,102 @@ class BSP_Exchanger:
             recorder.end("comm")
 
 
+class OverlappedGradExchanger:
+    """BSP 'grad'-mode exchanger that overlaps the all-reduce with the
+    backward pass — headroom the reference never had (SURVEY.md §7
+    phase 4: "exchange buckets as soon as ... finishes").
+
+    Parameters are bucketed in reverse registration order (approximate
+    backward order).  A post-accumulate-grad hook fires per parameter;
+    when a bucket is complete its grads are packed to the bf16 wire
+    buffer and an async all-reduce launches while backward continues.
+    ``finish()`` waits, scales by 1/N and unpacks.  Numerically this is
+    the reference's cdd rule (averaged deltas) for SGD+momentum.
+    """
+
+    def __init__(self, recs, size, group=None, bucket_mb=25,
+                 wire_dtype=torch.bfloat16):
+        self.size = size
+        self.group = group
+        self.enabled = size > 1
+        dev = recs[0].master.device
+        wire_dtype = wire_dtype if dev.type == "cuda" else torch.float32
+        self.buckets = []
+        cur, cur_elems = [], 0
+        max_elems = int(bucket_mb * (1 << 20)) // 2
+        for r in reversed(list(recs)):          # ~backward order
+            if cur and cur_elems + r.master.numel() > max_elems:
+                self.buckets.append(cur)
+                cur, cur_elems = [], 0
+            cur.append(r)
+            cur_elems += r.master.numel()
+        if cur:
+            self.buckets.append(cur)
+        self._state = []
+        self._by_param = {}
+        for bi, group_recs in enumerate(self.buckets):
+            n = sum(r.master.numel() for r in group_recs)
+            wire = torch.empty(n, dtype=wire_dtype, device=dev)
+            offs = {}
+            off = 0
+            for r in group_recs:
+                offs[id(r.master)] = off
+                off += r.master.numel()
+            st = {"recs": group_recs, "wire": wire, "offs": offs,
+                  "ready": 0, "work": None}
+            self._state.append(st)
+            for r in group_recs:
+                self._by_param[id(r.master)] = st
+        if self.enabled:
+            for st in self._state:
+                for r in st["recs"]:
+                    r.master.register_post_accumulate_grad_hook(
+                        self._hook)
+
+    @torch.no_grad()
+    def _hook(self, param):
+        st = self._by_param.get(id(param))
+        if st is None:
+            return
+        st["ready"] += 1
+        if st["ready"] == len(st["recs"]):
+            self._launch(st)
+
+    @torch.no_grad()
+    def _launch(self, st):
+        wire = st["wire"]
+        for r in st["recs"]:
+            off = st["offs"][id(r.master)]
+            n = r.master.numel()
+            wire[off:off + n].copy_(r.master.grad.view(-1))
+        st["work"] = dist.all_reduce(wire, group=self.group,
+                                     async_op=True)
+
+    @torch.no_grad()
+    def finish(self, recorder=None):
+        """Wait for in-flight reductions, unpack averaged grads."""
+        if not self.enabled:
+            return
+        if recorder:
+            recorder.start()
+        inv = 1.0 / self.size
+        for st in self._state:
+            if st["ready"] == 0:
+                continue
+            if st["work"] is None:      # bucket partially filled: flush
+                self._launch(st)
+            st["work"].wait()
+            st["wire"].mul_(inv)
+            for r in st["recs"]:
+                off = st["offs"][id(r.master)]
+                n = r.master.numel()
+                r.master.grad.view(-1).copy_(st["wire"][off:off + n])
+            st["ready"] = 0
+            st["work"] = None
+        if recorder:
+            recorder.end("comm")
+
+
 class EASGD_Exchanger:
     """Elastic-averaging swaps (reference EASGD_Exchanger [pub], C4)."""
 

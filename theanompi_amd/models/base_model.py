@@ -78,6 +78,16 @@ class BaseModel:
                                           wd_mult=0.0))
                 self.recs.append(ParamRec(mod.beta, name + ".beta",
                                           wd_mult=0.0))
+        # velocities live in one flat arena so the BSP cdd exchange is a
+        # single zero-copy all-reduce (no per-tensor flatten copies)
+        total = sum(r.master.numel() for r in self.recs)
+        self.vel_arena = torch.zeros(total, dtype=torch.float32,
+                                     device=self.device)
+        off = 0
+        for r in self.recs:
+            n = r.master.numel()
+            r.vel = self.vel_arena[off:off + n].view(r.master.shape)
+            off += n
         # reference-parity flat views for the exchangers / checkpoints
         self.params = [r.master for r in self.recs]
         self.vels = [r.vel for r in self.recs]

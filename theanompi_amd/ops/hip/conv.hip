@@ -59,7 +59,19 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  const int mTile = blockIdx.x, nTile = blockIdx.y, g = blockIdx.z;
+  // XCD-aware m-tile remap: give each XCD a contiguous chunk of m-tiles
+  // so neighbouring tiles (overlapping conv receptive fields) share L2.
+  int mTile = blockIdx.x;
+  {
+    const int nwg = gridDim.x;
+    if (nwg >= 16) {
+      const int q = nwg >> 3, r = nwg & 7;
+      const int xcd = mTile & 7, idx = mTile >> 3;
+      mTile = (xcd < r) ? xcd * (q + 1) + idx
+                        : r * (q + 1) + (xcd - r) * q + idx;
+    }
+  }
+  const int nTile = blockIdx.y, g = blockIdx.z;
 
   const __bf16* Wg = a.W + (long)g * a.BNtot * a.Kdf;
   const long gInOff = (long)g * a.groupInOff;
@@ -424,6 +436,7 @@ hipError_t launch_bias_wgrad(const void* dy, float* db, long rows, int C,
   int gx = cdiv(c8n, 32);
   long gy = (rows + 31) / 32;              // 4 rows per row-lane pass
   long target = 512 / gx;                  // bounded atomics per channel
+  if (target > 64) target = 64;
   if (gy > target) gy = target;
   if (gy < 1) gy = 1;
   dim3 grid(gx, (unsigned)gy);
