@@ -26,6 +26,8 @@ struct GatherGemmArgs {
   int Kdf;
   int BNtot;
   int relu;
+  int splitK;
+  float* Y32;
 };
 
 struct WgradArgs {
@@ -104,6 +106,18 @@ hipError_t launch_sgd_multi(const void*, int, long, float, int,
                             hipStream_t);
 hipError_t launch_gather_multi(const void*, int, long, hipStream_t);
 hipError_t launch_tr_probe(const int*, float*, int, hipStream_t);
+hipError_t launch_splitk_finalize(const float*, const float*, void*, long,
+                                  int, int, hipStream_t);
+}
+
+// pick a K-split so small-M GEMMs (FC layers at batch<=256) still fill
+// the 256-CU chip; partials go through a zeroed fp32 workspace.
+static int pick_splitk(long baseBlocks, long nt) {
+  if (baseBlocks >= 64 || nt < 4) return 1;
+  long s = 512 / (baseBlocks ? baseBlocks : 1);
+  if (s > nt) s = nt;
+  if (s < 1) s = 1;
+  return (int)s;
 }
 
 struct SgdSegH {
@@ -265,8 +279,22 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
   a.Kdf = (int)Kdf;
   a.BNtot = (int)Kgp;
   a.relu = relu ? 1 : 0;
+  long BN = Kgp > 64 ? 128 : 64;
+  long baseBlocks = ((a.M + 127) / 128) * ((Kgp + BN - 1) / BN) * G;
+  a.splitK = pick_splitk(baseBlocks, Kdf / 64);
+  torch::Tensor y32;
+  a.Y32 = nullptr;
+  if (a.splitK > 1) {
+    y32 = torch::zeros({y.numel()}, y.options().dtype(torch::kFloat32));
+    a.Y32 = y32.data_ptr<float>();
+  }
   CHECK_HIP(launch_conv_gather_gemm(&a, (int)G, detect_layout(),
                                     cur_stream()));
+  if (a.splitK > 1) {
+    CHECK_HIP(launch_splitk_finalize(a.Y32, a.bias, y.data_ptr(),
+                                     y.numel(), (int)(G * Kgp),
+                                     a.relu, cur_stream()));
+  }
 }
 
 static void conv_dgrad(torch::Tensor dyp, torch::Tensor wdg,
@@ -305,8 +333,26 @@ static void conv_dgrad(torch::Tensor dyp, torch::Tensor wdg,
   a.Kdf = (int)Kdf2;
   a.BNtot = (int)BNtot;
   a.relu = 0;
+  a.splitK = 1;
+  a.Y32 = nullptr;
+  torch::Tensor y32;
+  if (sh == 1 && sw == 1) {      // one class = dense dx: split-K legal
+    long BN = BNtot > 64 ? 128 : 64;
+    long baseBlocks = ((a.M + 127) / 128) * ((BNtot + BN - 1) / BN) * G;
+    a.splitK = pick_splitk(baseBlocks, Kdf2 / 64);
+    if (a.splitK > 1) {
+      y32 = torch::zeros({dx.numel()},
+                         dx.options().dtype(torch::kFloat32));
+      a.Y32 = y32.data_ptr<float>();
+    }
+  }
   CHECK_HIP(launch_conv_gather_gemm(&a, (int)G, detect_layout(),
                                     cur_stream()));
+  if (a.splitK > 1) {
+    CHECK_HIP(launch_splitk_finalize(a.Y32, nullptr, dx.data_ptr(),
+                                     dx.numel(), (int)Cgp, 0,
+                                     cur_stream()));
+  }
 }
 
 static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,

@@ -38,6 +38,8 @@ struct GatherGemmArgs {
   int Kdf;                          // flat reduction size (mult of 64)
   int BNtot;                        // output channels per group (Kgp)
   int relu;
+  int splitK;                       // >1: K-slices atomic-add into Y32
+  float* __restrict__ Y32;          // fp32 workspace for split-K
 };
 
 template <int BM, int BN, int LAYOUT>
@@ -71,7 +73,9 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
                         : r * (q + 1) + (xcd - r) * q + idx;
     }
   }
-  const int nTile = blockIdx.y, g = blockIdx.z;
+  const int nTile = blockIdx.y;
+  const int g = blockIdx.z / a.splitK;
+  const int slice = blockIdx.z % a.splitK;
 
   const __bf16* Wg = a.W + (long)g * a.BNtot * a.Kdf;
   const long gInOff = (long)g * a.groupInOff;
@@ -133,11 +137,12 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
     for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int nt = a.Kdf / BK;
-  stage(0, 0);
+  stage(0, slice * BK);
   __syncthreads();
-  for (int t = 0; t < nt; ++t) {
-    if (t + 1 < nt) stage((t + 1) & 1, (t + 1) * BK);
-    const int buf = t & 1;
+  int ti = 0;
+  for (int t = slice; t < nt; t += a.splitK, ++ti) {
+    if (t + a.splitK < nt) stage((ti + 1) & 1, (t + a.splitK) * BK);
+    const int buf = ti & 1;
 #pragma unroll
     for (int ks = 0; ks < BK; ks += 32) {
       bf16x8 af[FM], bf[FN];
@@ -181,11 +186,29 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
         int nl = n0 + j * 16 + (lane & 15);
         if (bn0 + nl >= a.BNtot) continue;
         float v = acc[i][j][reg];
-        if (a.bias) v += a.bias[cOut0 + nl];
-        if (a.relu) v = v > 0.f ? v : 0.f;
-        a.Y[obase + nl] = (__bf16)v;
+        if (a.Y32) {                       // split-K: fp32 partials;
+          atomicAdd(&a.Y32[obase + nl], v);  // bias/relu in finalize
+        } else {
+          if (a.bias) v += a.bias[cOut0 + nl];
+          if (a.relu) v = v > 0.f ? v : 0.f;
+          a.Y[obase + nl] = (__bf16)v;
+        }
       }
     }
+  }
+}
+
+// split-K finalize: y = bf16(relu(Y32 + bias))
+__global__ void splitk_finalize_k(const float* __restrict__ y32,
+                                  const float* __restrict__ bias,
+                                  __bf16* __restrict__ y, long total,
+                                  int C, int relu) {
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    float v = y32[i];
+    if (bias) v += bias[i % C];
+    if (relu && v < 0.f) v = 0.f;
+    y[i] = (__bf16)v;
   }
 }
 
@@ -392,11 +415,21 @@ __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
 
 extern "C" {
 
+hipError_t launch_splitk_finalize(const float* y32, const float* bias,
+                                  void* y, long total, int C, int relu,
+                                  hipStream_t stream) {
+  long g = (total + 255) / 256;
+  if (g > 2048) g = 2048;
+  hipLaunchKernelGGL(splitk_finalize_k, dim3((int)g), dim3(256), 0,
+                     stream, y32, bias, (__bf16*)y, total, C, relu);
+  return hipGetLastError();
+}
+
 hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
                                    int layout, hipStream_t stream) {
   GatherGemmArgs a = *args;
   int BN = a.BNtot > 64 ? 128 : 64;
-  dim3 grid(cdiv(a.M, 128), cdiv(a.BNtot, BN), G);
+  dim3 grid(cdiv(a.M, 128), cdiv(a.BNtot, BN), G * a.splitK);
   dim3 block(256);
   if (BN == 128) {
     if (layout == FRAG_CONTIG8)

@@ -129,20 +129,67 @@ __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
 // Pass 3: normalize (+ optional fused ReLU).
 // ---------------------------------------------------------------------------
 
-__global__ void bn_sums_k(const __bf16* __restrict__ x, float* __restrict__ s1,
-                          float* __restrict__ s2, long rows, int C,
-                          int slices) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  int slice = blockIdx.y;
-  if (c >= C) return;
-  float a = 0.f, b = 0.f;
-  for (long r = slice; r < rows; r += slices) {
-    float v = (float)x[r * C + c];
-    a += v;
-    b += v * v;
+// 32 channel-groups x 8 row-lanes per block, 4-row ILP, LDS-reduced
+// (same shape as bias_wgrad_k; the naive per-channel-thread version was
+// the ResNet-50 bottleneck).
+__global__ void bn_sums_k(const __bf16* __restrict__ x,
+                          float* __restrict__ s1, float* __restrict__ s2,
+                          long rows, int C) {
+  const int c8n = C / 8;
+  const int cl = threadIdx.x & 31;
+  const int rl = threadIdx.x >> 5;
+  const int c8 = blockIdx.x * 32 + cl;
+  const long sl = (long)blockIdx.y * 8 + rl;
+  const long nsl = (long)gridDim.y * 8;
+  float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float b[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  if (c8 < c8n) {
+    const __bf16* base = x + c8 * 8;
+    long r4 = sl * 4;
+    for (; r4 + 4 <= rows; r4 += nsl * 4) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        bf16x8 v = *(const bf16x8*)(base + (r4 + u) * C);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = (float)v[j];
+          a[j] += f;
+          b[j] += f * f;
+        }
+      }
+    }
+    if (blockIdx.y == 0 && rl == 0) {
+      for (long r = rows & ~3L; r < rows; ++r) {
+        bf16x8 v = *(const bf16x8*)(base + r * C);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float f = (float)v[j];
+          a[j] += f;
+          b[j] += f * f;
+        }
+      }
+    }
   }
-  atomicAdd(&s1[c], a);
-  atomicAdd(&s2[c], b);
+  __shared__ float red[2][8][32][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][rl][cl][j] = a[j];
+    red[1][rl][cl][j] = b[j];
+  }
+  __syncthreads();
+  if (rl == 0 && c8 < c8n) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float sa = 0.f, sb = 0.f;
+#pragma unroll
+      for (int t = 0; t < 8; ++t) {
+        sa += red[0][t][cl][j];
+        sb += red[1][t][cl][j];
+      }
+      atomicAdd(&s1[c8 * 8 + j], sa);
+      atomicAdd(&s2[c8 * 8 + j], sb);
+    }
+  }
 }
 
 __global__ void bn_finalize_k(const float* __restrict__ s1,
@@ -172,12 +219,24 @@ __global__ void bn_apply_k(const __bf16* __restrict__ x,
                            const float* __restrict__ gamma,
                            const float* __restrict__ beta, long rows,
                            int C, int relu) {
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
-       i += (long)gridDim.x * blockDim.x) {
+  const long t4 = rows * C / 4;
+  for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < t4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
     int c = (int)(i % C);
-    float v = ((float)x[i] - mean[c]) * invstd[c] * gamma[c] + beta[c];
-    if (relu && v < 0.f) v = 0.f;
-    y[i] = (__bf16)v;
+    bf16x4 xv = *(const bf16x4*)(x + i);
+    f32x4 mv = *(const f32x4*)(mean + c);
+    f32x4 iv = *(const f32x4*)(invstd + c);
+    f32x4 gv = *(const f32x4*)(gamma + c);
+    f32x4 bv = *(const f32x4*)(beta + c);
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float v = ((float)xv[j] - mv[j]) * iv[j] * gv[j] + bv[j];
+      if (relu && v < 0.f) v = 0.f;
+      o[j] = (__bf16)v;
+    }
+    *(bf16x4*)(y + i) = o;
   }
 }
 
@@ -190,22 +249,66 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
                               const float* __restrict__ invstd,
                               float* __restrict__ sdy,
                               float* __restrict__ sdyx, long rows, int C,
-                              int slices, int relu) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  int slice = blockIdx.y;
-  if (c >= C) return;
-  float a = 0.f, b = 0.f;
-  const float m = mean[c], is = invstd[c];
-  for (long r = slice; r < rows; r += slices) {
-    long i = r * C + c;
-    float g = (float)dy[i];
-    if (relu && (float)y[i] <= 0.f) g = 0.f;
-    float xh = ((float)x[i] - m) * is;
-    a += g;
-    b += g * xh;
+                              int relu) {
+  const int c8n = C / 8;
+  const int cl = threadIdx.x & 31;
+  const int rl = threadIdx.x >> 5;
+  const int c8 = blockIdx.x * 32 + cl;
+  const long sl = (long)blockIdx.y * 8 + rl;
+  const long nsl = (long)gridDim.y * 8;
+  float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float b[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  float m[8], is[8];
+  if (c8 < c8n) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      m[j] = mean[c8 * 8 + j];
+      is[j] = invstd[c8 * 8 + j];
+    }
+    const long co = c8 * 8;
+    auto accrow = [&](long r) {
+      bf16x8 g8 = *(const bf16x8*)(dy + r * C + co);
+      bf16x8 x8 = *(const bf16x8*)(x + r * C + co);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = (float)g8[j];
+        if (relu) {
+          float yv = (float)((const __bf16*)(y + r * C + co))[j];
+          if (yv <= 0.f) g = 0.f;
+        }
+        float xh = ((float)x8[j] - m[j]) * is[j];
+        a[j] += g;
+        b[j] += g * xh;
+      }
+    };
+    long r4 = sl * 4;
+    for (; r4 + 4 <= rows; r4 += nsl * 4) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u) accrow(r4 + u);
+    }
+    if (blockIdx.y == 0 && rl == 0)
+      for (long r = rows & ~3L; r < rows; ++r) accrow(r);
   }
-  atomicAdd(&sdy[c], a);
-  atomicAdd(&sdyx[c], b);
+  __shared__ float red[2][8][32][8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    red[0][rl][cl][j] = a[j];
+    red[1][rl][cl][j] = b[j];
+  }
+  __syncthreads();
+  if (rl == 0 && c8 < c8n) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float sa = 0.f, sb = 0.f;
+#pragma unroll
+      for (int t = 0; t < 8; ++t) {
+        sa += red[0][t][cl][j];
+        sb += red[1][t][cl][j];
+      }
+      atomicAdd(&sdy[c8 * 8 + j], sa);
+      atomicAdd(&sdyx[c8 * 8 + j], sb);
+    }
+  }
 }
 
 // backward pass 2: dx = invstd*gamma*(dy - sdy/M - xhat*sdyx/M)
@@ -221,15 +324,29 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
                                __bf16* __restrict__ dx, long rows, int C,
                                int relu) {
   const float invM = 1.f / rows;
-  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
-       i += (long)gridDim.x * blockDim.x) {
+  const long t4 = rows * C / 4;
+  for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < t4;
+       i4 += (long)gridDim.x * blockDim.x) {
+    long i = i4 * 4;
     int c = (int)(i % C);
-    float g = (float)dy[i];
-    if (relu && (float)y[i] <= 0.f) g = 0.f;
-    float xh = ((float)x[i] - mean[c]) * invstd[c];
-    float d = invstd[c] * gamma[c]
-              * (g - sdy[c] * invM - xh * sdyx[c] * invM);
-    dx[i] = (__bf16)d;
+    bf16x4 gv4 = *(const bf16x4*)(dy + i);
+    bf16x4 xv4 = *(const bf16x4*)(x + i);
+    bf16x4 yv4 = *(const bf16x4*)(y + i);
+    f32x4 mv = *(const f32x4*)(mean + c);
+    f32x4 iv = *(const f32x4*)(invstd + c);
+    f32x4 ga = *(const f32x4*)(gamma + c);
+    f32x4 s1 = *(const f32x4*)(sdy + c);
+    f32x4 s2 = *(const f32x4*)(sdyx + c);
+    bf16x4 o;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      float g = (float)gv4[j];
+      if (relu && (float)yv4[j] <= 0.f) g = 0.f;
+      float xh = ((float)xv4[j] - mv[j]) * iv[j];
+      o[j] = (__bf16)(iv[j] * ga[j]
+                      * (g - s1[j] * invM - xh * s2[j] * invM));
+    }
+    *(bf16x4*)(dx + i) = o;
   }
 }
 
@@ -279,13 +396,19 @@ hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
                                const float* gamma, const float* beta,
                                long rows, int C, float momentum,
                                float eps, int relu, hipStream_t stream) {
-  int slices = 64;
-  hipLaunchKernelGGL(bn_sums_k, dim3(cdiv(C, 256), slices), dim3(256), 0,
-                     stream, (const __bf16*)x, s1, s2, rows, C, slices);
+  {
+    int gx = cdiv(C / 8, 32);
+    long gy = (rows + 31) / 32;
+    long cap = 512 / gx; if (cap > 64) cap = 64;
+    if (gy > cap) gy = cap;
+    if (gy < 1) gy = 1;
+    hipLaunchKernelGGL(bn_sums_k, dim3(gx, (unsigned)gy), dim3(256), 0,
+                       stream, (const __bf16*)x, s1, s2, rows, C);
+  }
   hipLaunchKernelGGL(bn_finalize_k, dim3(cdiv(C, 256)), dim3(256), 0,
                      stream, s1, s2, mean, invstd, rmean, rvar, rows, C,
                      momentum, eps);
-  hipLaunchKernelGGL(bn_apply_k, dim3(ngrid(rows * C)), dim3(256), 0,
+  hipLaunchKernelGGL(bn_apply_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
                      stream, (const __bf16*)x, (__bf16*)y, mean, invstd,
                      gamma, beta, rows, C, relu);
   return hipGetLastError();
@@ -309,13 +432,19 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
                          const float* gamma, float* sdy, float* sdyx,
                          void* dx, long rows, int C, int relu,
                          hipStream_t stream) {
-  int slices = 64;
-  hipLaunchKernelGGL(bn_bwd_sums_k, dim3(cdiv(C, 256), slices), dim3(256),
+  {
+    int gx = cdiv(C / 8, 32);
+    long gy = (rows + 31) / 32;
+    long cap = 512 / gx; if (cap > 64) cap = 64;
+    if (gy > cap) gy = cap;
+    if (gy < 1) gy = 1;
+    hipLaunchKernelGGL(bn_bwd_sums_k, dim3(gx, (unsigned)gy), dim3(256),
+                       0, stream, (const __bf16*)dy, (const __bf16*)x,
+                       (const __bf16*)y, mean, invstd, sdy, sdyx, rows,
+                       C, relu);
+  }
+  hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C / 4)), dim3(256),
                      0, stream, (const __bf16*)dy, (const __bf16*)x,
-                     (const __bf16*)y, mean, invstd, sdy, sdyx, rows, C,
-                     slices, relu);
-  hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C)), dim3(256), 0,
-                     stream, (const __bf16*)dy, (const __bf16*)x,
                      (const __bf16*)y, mean, invstd, gamma, sdy, sdyx,
                      (__bf16*)dx, rows, C, relu);
   return hipGetLastError();
