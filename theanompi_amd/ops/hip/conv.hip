@@ -468,8 +468,8 @@ hipError_t launch_bias_wgrad(const void* dy, float* db, long rows, int C,
   int c8n = C / 8;
   int gx = cdiv(c8n, 32);
   long gy = (rows + 31) / 32;              // 4 rows per row-lane pass
-  long target = 512 / gx;                  // bounded atomics per channel
-  if (target > 64) target = 64;
+  long target = 1024 / gx;                 // bounded atomics per channel
+  if (target > 128) target = 128;
   if (gy > target) gy = target;
   if (gy < 1) gy = 1;
   dim3 grid(gx, (unsigned)gy);

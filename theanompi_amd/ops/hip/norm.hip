@@ -269,13 +269,12 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
     auto accrow = [&](long r) {
       bf16x8 g8 = *(const bf16x8*)(dy + r * C + co);
       bf16x8 x8 = *(const bf16x8*)(x + r * C + co);
+      bf16x8 y8;
+      if (relu) y8 = *(const bf16x8*)(y + r * C + co);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = (float)g8[j];
-        if (relu) {
-          float yv = (float)((const __bf16*)(y + r * C + co))[j];
-          if (yv <= 0.f) g = 0.f;
-        }
+        if (relu && (float)y8[j] <= 0.f) g = 0.f;
         float xh = ((float)x8[j] - m[j]) * is[j];
         a[j] += g;
         b[j] += g * xh;
@@ -399,7 +398,7 @@ hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
   {
     int gx = cdiv(C / 8, 32);
     long gy = (rows + 31) / 32;
-    long cap = 512 / gx; if (cap > 64) cap = 64;
+    long cap = 1024 / gx; if (cap > 128) cap = 128;
     if (gy > cap) gy = cap;
     if (gy < 1) gy = 1;
     hipLaunchKernelGGL(bn_sums_k, dim3(gx, (unsigned)gy), dim3(256), 0,
@@ -435,7 +434,7 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
   {
     int gx = cdiv(C / 8, 32);
     long gy = (rows + 31) / 32;
-    long cap = 512 / gx; if (cap > 64) cap = 64;
+    long cap = 1024 / gx; if (cap > 128) cap = 128;
     if (gy > cap) gy = cap;
     if (gy < 1) gy = 1;
     hipLaunchKernelGGL(bn_bwd_sums_k, dim3(gx, (unsigned)gy), dim3(256),
