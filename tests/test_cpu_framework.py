@@ -224,3 +224,32 @@ def test_convplan_flat_roundtrip_grouped():
     assert flat.shape == (2, plan.Kgp, plan.Kdf)
     back = plan.flat_to_w(flat)
     assert torch.allclose(w, back)
+
+
+def test_imagenet_shard_loader(tmp_path):
+    """Shard dataset + separate loader child process."""
+    from theanompi_amd.models.data.imagenet import (ImageNetData,
+                                                    write_synthetic_shards)
+    paths = write_synthetic_shards(str(tmp_path), n_shards=3,
+                                   images_per_shard=8, hw=40)
+    data = ImageNetData(paths, paths[:1], batch_size=8, crop=32,
+                        n_classes=1000, device="cpu")
+    try:
+        assert data.n_batch_train == 3
+        x, y = data.get_train_batch(0)
+        assert x.shape == (8, 32, 32, 3)
+        assert x.dtype == torch.float32
+        assert y.shape == (8,)
+        xv, yv = data.get_val_batch(0)
+        assert xv.shape == (8, 32, 32, 3)
+    finally:
+        data.close()
+
+
+def test_cifar10_data_object():
+    from theanompi_amd.models.data.cifar10_data import Cifar10Data
+    d = Cifar10Data(batch_size=16, device="cpu", synthetic_train=64,
+                    synthetic_val=32)
+    x, y = d.get_train_batch(0)
+    assert x.shape == (16, 32, 32, 3)
+    assert d.n_batch_train == 4

@@ -10,7 +10,6 @@ import torch.nn as nn
 
 from theanompi_amd.models import layers as L
 from theanompi_amd.models.base_model import BaseModel
-from theanompi_amd.models.data.synthetic import SyntheticData
 
 
 class Cifar10_model(BaseModel):
@@ -47,10 +46,13 @@ class Cifar10_model(BaseModel):
         return m
 
     def build_data(self):
-        return SyntheticData(
-            batch_size=self.batch_size, image_hw=self.image_size,
-            n_classes=self.n_classes, device=self.device,
-            c_phys=4 if self.gpu else 3,
-            n_train_batches=int(self.config.get("n_iters", 32)),
-            n_val_batches=int(self.config.get("val_iters", 4)),
-            rank=self.rank, size=self.size, use_loader=False)
+        from theanompi_amd.models.data.cifar10_data import Cifar10Data
+        n_iters = int(self.config.get("n_iters", 32))
+        return Cifar10Data(
+            batch_size=self.batch_size, device=self.device,
+            path=self.config.get("data_path"),
+            n_classes=self.n_classes, c_phys=4,
+            rank=self.rank, size=self.size,
+            synthetic_train=self.batch_size * n_iters,
+            synthetic_val=self.batch_size
+            * int(self.config.get("val_iters", 4)))

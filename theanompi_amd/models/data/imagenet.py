@@ -1,0 +1,198 @@
+"""ImageNet shard dataset + parallel loader child process.
+
+Rebuild of ``theanompi/models/data/imagenet.py`` +
+``data/proc_load_mpi.py`` [pub].  The reference
+streamed hickle shards (256 images x 3x256x256 uint8 per file) through a
+child process spawned over an MPI intercomm, double-buffering so
+load(k+1) overlaps compute(k).
+
+MI355X-native version:
+  * shards are ``.npy`` files (uint8 [n, 256, 256, 3]) listed in a
+    train/val manifest; labels per shard in ``<name>_labels.npy``
+    (hickle is not installed in this environment; a converter from hkl
+    is one `hkl.load` + `np.save` away).
+  * one LOADER CHILD PROCESS per worker (multiprocessing, replaces
+    MPI.COMM_SELF.Spawn): receives shard paths over a queue, loads +
+    augments on the CPU (random crop to `crop`, horizontal mirror, mean
+    subtract is folded into the on-GPU u8_finish kernel), and returns
+    batches through shared pinned-memory ring slots.
+  * the parent copies pinned->HBM with hipMemcpyAsync on the side copy
+    stream and event-fences the compute stream (N18), exactly like the
+    synthetic loader.
+
+Shards are partitioned across workers per epoch with an optional
+per-epoch shuffle, as in the reference.
+"""
+
+import os
+
+import numpy as np
+import torch
+import torch.multiprocessing as tmp
+
+
+def _loader_child(task_q, slot_q, done_q, slots, lslots, crop, train):
+    """Child process: load shard -> augment -> write into a pinned slot.
+    Mirrors proc_load_mpi.py's recv(filename) / Send(batch) loop."""
+    rng = np.random.RandomState(os.getpid())
+    while True:
+        task = task_q.get()
+        if task is None:
+            break
+        path, slot = task
+        arr = np.load(path)                      # [n, H, W, 3] uint8
+        labels = np.load(path.replace(".npy", "_labels.npy"))
+        n, H, W, _ = arr.shape
+        if train:
+            y0 = rng.randint(0, H - crop + 1)
+            x0 = rng.randint(0, W - crop + 1)
+            flip = rng.rand() < 0.5
+        else:
+            y0 = (H - crop) // 2
+            x0 = (W - crop) // 2
+            flip = False
+        view = arr[:, y0:y0 + crop, x0:x0 + crop, :]
+        if flip:
+            view = view[:, :, ::-1, :]
+        dst = slots[slot]
+        m = min(n, dst.shape[0])
+        dst[:m, :, :, :3] = torch.from_numpy(np.ascontiguousarray(view[:m]))
+        dst[:m, :, :, 3:] = 0
+        lslots[slot][:m] = torch.from_numpy(labels[:m].astype(np.int64))
+        done_q.put(slot)
+
+
+class ImageNetData:
+    """Data-object side of the model contract over real shards."""
+
+    def __init__(self, train_list, val_list, batch_size, crop, n_classes,
+                 device, rank=0, size=1, shuffle=True, c_phys=4,
+                 depth=2):
+        self.train_shards = list(train_list)
+        self.val_shards = list(val_list)
+        self.batch_size = batch_size
+        self.crop = crop
+        self.n_classes = n_classes
+        self.device = torch.device(device)
+        self.gpu = self.device.type == "cuda"
+        self.rank, self.size = rank, size
+        self.shuffle = shuffle
+        self.c_phys = c_phys
+        self.epoch = 0
+        self._partition()
+
+        # pinned ring shared with the child process
+        ctx = tmp.get_context("spawn")
+        self.slots = [torch.empty((batch_size, crop, crop, c_phys),
+                                  dtype=torch.uint8) for _ in range(depth)]
+        self.lslots = [torch.empty((batch_size,), dtype=torch.int64)
+                       for _ in range(depth)]
+        # shared (not pinned): pinned host memory cannot be shared across
+        # processes portably; the h2d copy below still overlaps compute
+        # on the dedicated copy stream.
+        for t in self.slots + self.lslots:
+            t.share_memory_()
+        self.task_q = ctx.Queue()
+        self.done_q = ctx.Queue()
+        self.child = ctx.Process(
+            target=_loader_child,
+            args=(self.task_q, None, self.done_q, self.slots, self.lslots,
+                  crop, True), daemon=True)
+        self.child.start()
+        self.depth = depth
+        self._next_slot = 0
+        self._inflight = []
+        if self.gpu:
+            self.copy_stream = torch.cuda.Stream(device=self.device)
+            self.dev_bufs = [torch.empty_like(self.slots[0],
+                                              device=self.device)
+                             for _ in range(depth)]
+            self.dev_lbufs = [torch.empty_like(self.lslots[0],
+                                               device=self.device)
+                              for _ in range(depth)]
+            self.copy_events = [torch.cuda.Event() for _ in range(depth)]
+
+    def _partition(self):
+        shards = list(self.train_shards)
+        if self.shuffle:
+            rng = np.random.RandomState(1000 + self.epoch)
+            rng.shuffle(shards)
+        self.my_shards = shards[self.rank::self.size]
+        self.n_batch_train = len(self.my_shards)
+        self.n_batch_val = len(self.val_shards)
+
+    # -- contract -------------------------------------------------------------
+
+    def reset(self, mode):
+        if mode == "train":
+            self.epoch += 1
+            self._partition()
+
+    def _request(self, path, slot):
+        self.task_q.put((path, slot))
+        self._inflight.append(slot)
+
+    def get_train_batch(self, i):
+        # prime the pipeline depth
+        while len(self._inflight) < self.depth and i + len(
+                self._inflight) < self.n_batch_train:
+            s = self._next_slot
+            self._next_slot = (s + 1) % self.depth
+            self._request(self.my_shards[(i + len(self._inflight))
+                                         % self.n_batch_train], s)
+        slot = self.done_q.get()                 # child finished a shard
+        self._inflight.remove(slot)
+        x_u8, y = self.slots[slot], self.lslots[slot]
+        if self.gpu:
+            with torch.cuda.stream(self.copy_stream):
+                self.dev_bufs[slot].copy_(x_u8, non_blocking=True)
+                self.dev_lbufs[slot].copy_(y, non_blocking=True)
+                self.copy_events[slot].record(self.copy_stream)
+            torch.cuda.current_stream().wait_event(self.copy_events[slot])
+            from theanompi_amd import ops as _ops
+            x = _ops.ext().u8_finish(self.dev_bufs[slot], 3)
+            return x, self.dev_lbufs[slot].clone()
+        x = x_u8.float() * (1 / 128.0) - 1.0
+        x[..., 3:] = 0
+        return x[..., :3] if self.c_phys == 4 and not self.gpu else x, \
+            y.clone()
+
+    def get_val_batch(self, i):
+        path = self.val_shards[i % len(self.val_shards)]
+        arr = np.load(path)
+        labels = np.load(path.replace(".npy", "_labels.npy"))
+        H = arr.shape[1]
+        o = (H - self.crop) // 2
+        view = np.ascontiguousarray(
+            arr[:self.batch_size, o:o + self.crop, o:o + self.crop])
+        x = torch.from_numpy(view)
+        pad = torch.zeros((*x.shape[:3], self.c_phys), dtype=torch.uint8)
+        pad[..., :3] = x
+        y = torch.from_numpy(labels[:self.batch_size].astype(np.int64))
+        if self.gpu:
+            from theanompi_amd import ops as _ops
+            return _ops.ext().u8_finish(pad.to(self.device), 3), \
+                y.to(self.device)
+        xf = pad.float() * (1 / 128.0) - 1.0
+        return xf[..., :3], y
+
+    def close(self):
+        self.task_q.put(None)
+        self.child.join(timeout=5)
+
+
+def write_synthetic_shards(directory, n_shards=4, images_per_shard=256,
+                           hw=256, n_classes=1000, seed=0):
+    """Test/bench helper: writes ImageNet-shaped random shards."""
+    os.makedirs(directory, exist_ok=True)
+    rng = np.random.RandomState(seed)
+    paths = []
+    for s in range(n_shards):
+        arr = rng.randint(0, 256, (images_per_shard, hw, hw, 3),
+                          dtype=np.uint8)
+        lab = rng.randint(0, n_classes, (images_per_shard,))
+        p = os.path.join(directory, f"shard_{s:04d}.npy")
+        np.save(p, arr)
+        np.save(p.replace(".npy", "_labels.npy"), lab)
+        paths.append(p)
+    return paths
