@@ -42,7 +42,7 @@ struct GatherGemmArgs {
   float* __restrict__ Y32;          // fp32 workspace for split-K
 };
 
-template <int BM, int BN, int LAYOUT>
+template <int BM, int BN, int LAYOUT, int DEPTH>
 __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
   constexpr int BK = 64;
   constexpr int WAVES_M = (BN == 128) ? 2 : 4;
@@ -52,11 +52,16 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
   constexpr int FN = 64 / 16;            // wave tile N = 64 always
   constexpr int AI = BM / 32;            // glds instrs per wave (A tile)
   constexpr int BI = BN / 32;
+  constexpr int TPW = AI + BI;           // glds per wave per tile
 
   // lane-linear tiles (glds writes base + lane*16B); content is
   // XOR-swizzled by source-piece permutation, undone at the frag reads.
-  __shared__ __align__(16) __bf16 As[2][BM][BK];
-  __shared__ __align__(16) __bf16 Bs[2][BN][BK];
+  __shared__ __align__(16) __bf16 As[DEPTH][BM][BK];
+  __shared__ __align__(16) __bf16 Bs[DEPTH][BN][BK];
+  // gather table staged to LDS once: its reads then ride the lgkm
+  // counter, so they never force a vmcnt drain of in-flight glds
+  // (cdna guide §5 "three .s-level traps" (b)).
+  __shared__ int Tab[2048];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -81,9 +86,14 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
   const long gInOff = (long)g * a.groupInOff;
   const int bn0 = nTile * BN;
 
+  {  // table -> LDS (bounded by binding: Kdf/16 <= 2048)
+    const int n = a.Kdf >> 4;
+    for (int i = tid; i < n; i += 256) Tab[i] = a.table[i];
+  }
+
   // --- per-lane glds descriptors (fixed across K) --------------------------
   long aRowBase[AI];
-  int aK8x8[AI];                 // swizzled in-row byte... element offset
+  int aK8x8[AI];
 #pragma unroll
   for (int i = 0; i < AI; ++i) {
     int piece = (wave * AI + i) * 64 + lane;
@@ -109,11 +119,12 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
     if (n >= a.BNtot) n = a.BNtot - 1;         // discarded at C-write
     bSrc[i] = Wg + (long)n * a.Kdf + k8 * 8;
   }
+  __syncthreads();                             // table visible
 
   auto stage = [&](int buf, int k0) {
 #pragma unroll
     for (int i = 0; i < AI; ++i) {
-      int off = a.table[(k0 + aK8x8[i]) >> 4] + (aK8x8[i] & 15);
+      int off = Tab[(k0 + aK8x8[i]) >> 4] + (aK8x8[i] & 15);
       glds16(a.X + aRowBase[i] + off,
              &As[buf][(wave * AI + i) * 8][0]);
     }
@@ -137,12 +148,28 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
     for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int nt = a.Kdf / BK;
-  stage(0, slice * BK);
-  __syncthreads();
-  int ti = 0;
-  for (int t = slice; t < nt; t += a.splitK, ++ti) {
-    if (t + a.splitK < nt) stage((ti + 1) & 1, (t + a.splitK) * BK);
-    const int buf = ti & 1;
+  const int nti = (nt - slice + a.splitK - 1) / a.splitK;  // my K-tiles
+  // prologue: fill the ring DEPTH-1 deep
+#pragma unroll
+  for (int d = 0; d < DEPTH - 1; ++d)
+    if (d < nti) stage(d % DEPTH, (slice + (long)d * a.splitK) * BK);
+
+  for (int ti = 0; ti < nti; ++ti) {
+    const int tnext = ti + DEPTH - 1;
+    if (tnext < nti)
+      stage(tnext % DEPTH, (slice + (long)tnext * a.splitK) * BK);
+    // wait MY glds of tile ti: outstanding newer-tile instrs stay in
+    // flight (counted wait; never vmcnt(0) in the loop)
+    const int newer = (nti - 1 - ti) < (DEPTH - 1) ? (nti - 1 - ti)
+                                                   : (DEPTH - 1);
+    if (newer == 2)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(2 * TPW) : "memory");
+    else if (newer == 1)
+      asm volatile("s_waitcnt vmcnt(%0)" ::"i"(TPW) : "memory");
+    else
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();              // raw: no vmcnt drain
+    const int buf = ti % DEPTH;
 #pragma unroll
     for (int ks = 0; ks < BK; ks += 32) {
       bf16x8 af[FM], bf[FN];
@@ -163,7 +190,7 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bf[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
+    __builtin_amdgcn_s_barrier();              // readers done with buf
   }
 
   // --- epilogue: bias + relu + strided bf16 store ---------------------------
@@ -428,24 +455,32 @@ hipError_t launch_splitk_finalize(const float* y32, const float* bias,
 hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
                                    int layout, hipStream_t stream) {
   GatherGemmArgs a = *args;
+  if (a.Kdf > 2048 * 16) return hipErrorInvalidValue;  // LDS table cap
   int BN = a.BNtot > 64 ? 128 : 64;
   dim3 grid(cdiv(a.M, 128), cdiv(a.BNtot, BN), G * a.splitK);
   dim3 block(256);
+  const int depth = 2;   // 3-deep ring measured slower (1 blk/CU)
+#define GG_LAUNCH(BN_, L_, D_)                                             \
+  hipLaunchKernelGGL((conv_gather_gemm<128, BN_, L_, D_>), grid, block,    \
+                     0, stream, a)
   if (BN == 128) {
-    if (layout == FRAG_CONTIG8)
-      hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_CONTIG8>), grid,
-                         block, 0, stream, a);
-    else
-      hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_HALVES>), grid,
-                         block, 0, stream, a);
+    if (layout == FRAG_CONTIG8) {
+      if (depth == 3) GG_LAUNCH(128, FRAG_CONTIG8, 3);
+      else GG_LAUNCH(128, FRAG_CONTIG8, 2);
+    } else {
+      if (depth == 3) GG_LAUNCH(128, FRAG_HALVES, 3);
+      else GG_LAUNCH(128, FRAG_HALVES, 2);
+    }
   } else {
-    if (layout == FRAG_CONTIG8)
-      hipLaunchKernelGGL((conv_gather_gemm<128, 64, FRAG_CONTIG8>), grid,
-                         block, 0, stream, a);
-    else
-      hipLaunchKernelGGL((conv_gather_gemm<128, 64, FRAG_HALVES>), grid,
-                         block, 0, stream, a);
+    if (layout == FRAG_CONTIG8) {
+      if (depth == 3) GG_LAUNCH(64, FRAG_CONTIG8, 3);
+      else GG_LAUNCH(64, FRAG_CONTIG8, 2);
+    } else {
+      if (depth == 3) GG_LAUNCH(64, FRAG_HALVES, 3);
+      else GG_LAUNCH(64, FRAG_HALVES, 2);
+    }
   }
+#undef GG_LAUNCH
   return hipGetLastError();
 }
 
