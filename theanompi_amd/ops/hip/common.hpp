@@ -96,13 +96,14 @@ DEV bf16x8 frag_load_swz(const __bf16* row_start, int row, int ks,
 // in FRAG_CONTIG8 order.
 typedef __attribute__((address_space(3))) bf16x4 lds_bf16x4;
 
+template <int LDW>
 DEV bf16x8 frag_load_tr_swz(const __bf16* tile, int ks, int ch0,
                             int lane) {
   const int pix1 = ks + 8 * (lane >> 4) + ((lane & 15) >> 2);
   const int pix2 = pix1 + 4;
   const int co = ch0 + 4 * (lane & 3);
-  const __bf16* a1 = tile + pix1 * 64 + (co ^ ((pix1 & 7) * 8));
-  const __bf16* a2 = tile + pix2 * 64 + (co ^ ((pix2 & 7) * 8));
+  const __bf16* a1 = tile + pix1 * LDW + (co ^ ((pix1 & 7) * 8));
+  const __bf16* a2 = tile + pix2 * LDW + (co ^ ((pix2 & 7) * 8));
   bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf16x4*)a1);
   bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf16x4*)a2);
   bf16x8 r;

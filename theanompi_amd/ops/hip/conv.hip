@@ -256,19 +256,21 @@ struct WgradArgs {
   int splitK;
 };
 
-template <int LAYOUT>
+template <int LAYOUT, int BNF>
 __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
-  // TN GEMM over pixels.  Both operands stage LINEARLY ([pix][64ch],
-  // 16-B pieces, content-swizzled like the fwd tiles) and the MFMA
-  // k-major fragments come from ds_read_b64_tr_b16 (hardware transpose)
-  // -- no scalar LDS traffic anywhere (the v1 scalar transpose staging
-  // was 23% of the AlexNet step).
-  constexpr int BMK = 64;   // kout tile
-  constexpr int BNF = 64;   // fd tile
-  constexpr int BKP = 64;   // pixel chunk
-
-  __shared__ __align__(16) __bf16 DYt[2][BKP][64];  // [pix][kout]
-  __shared__ __align__(16) __bf16 XT[2][BKP][64];   // [pix][fd]
+  // TN GEMM over pixels.  Both operands stage LINEARLY ([pix][64ch]
+  // subtiles, 16-B glds pieces, content-swizzled) and the k-major MFMA
+  // fragments come from ds_read_b64_tr_b16 (hardware transpose).
+  // Counted-vmcnt double buffering, LDS-resident gather table.
+  constexpr int BMK = 64;          // kout tile
+  constexpr int BKP = 64;          // pixel chunk
+  constexpr int XI = BNF / 32;     // x-tile glds per wave
+  constexpr int DI = BMK / 32;     // dy-tile glds per wave
+  constexpr int TPW = XI + DI;
+  
+  __shared__ __align__(16) __bf16 DYt[2][BKP][64];        // [pix][kout]
+  __shared__ __align__(16) __bf16 XT[2][BKP][BNF];        // [pix][fd]
+  __shared__ int Tab[2048];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -282,89 +284,99 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   const long gInOff = (long)g * a.groupInOff;
   const long dyTotal = (long)a.M * a.Kphys;
 
-  // piece descriptors: piece = (wave*4 + i)*64 + lane; row (pix) =
-  // piece>>3, source slot k8 = (piece&7) ^ (row&7)
-  int rowOf[4], k8Of[4];
-#pragma unroll
-  for (int i = 0; i < 4; ++i) {
-    int piece = (wave * 4 + i) * 64 + lane;
-    rowOf[i] = piece >> 3;
-    k8Of[i] = ((piece & 7) ^ ((piece >> 3) & 7)) * 8;
+  {
+    const int n = a.Kdf >> 4;
+    for (int i = tid; i < n; i += 256) Tab[i] = a.table[i];
   }
+  __syncthreads();
 
+  // piece descriptors.  dy tile: piece = (wave*DI + i)*64 + lane over
+  // [BKP][64]; x tile: piece over [BKP][BNF], row = pix.
   auto stage = [&](int buf, int pix0) {
     const bool interior = (pix0 + BKP <= a.M);
+    // ---- x gather pieces ----
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int row = rowOf[i];
-      const int pa = pix0 + row;
-      const int pc = pa < a.M ? pa : a.M - 1;
+    for (int i = 0; i < XI; ++i) {
+      int piece = (wave * XI + i) * 64 + lane;
+      int row = piece / (BNF / 8);
+      int k8i = piece % (BNF / 8);
+      int k8 = ((k8i ^ row) & 7) * 8 + (k8i & ~7) * 8;   // swizzle low 3
+      int pa = pix0 + row;
+      int pc = pa < a.M ? pa : a.M - 1;
       int q = pc % a.Q;
       int tmp = pc / a.Q;
       int p = tmp % a.P;
       int n = tmp / a.P;
-      long xbase = (long)n * a.inNS + (long)p * a.inPS
-                   + (long)q * a.inQS + gInOff;
-      int xoff = a.table[(fd0 + k8Of[i]) >> 4] + (k8Of[i] & 15);
-      long dyaddr = (long)pc * a.Kphys + g * a.Kgp + kout0 + k8Of[i];
+      long base = (long)n * a.inNS + (long)p * a.inPS
+                  + (long)q * a.inQS + gInOff;
+      int off = Tab[(fd0 + k8) >> 4] + (k8 & 15);
+      glds16(a.X + base + off,
+             &XT[buf][0][0] + (long)(wave * XI + i) * 512);
+    }
+    // ---- dy pieces ----
+#pragma unroll
+    for (int i = 0; i < DI; ++i) {
+      int piece = (wave * DI + i) * 64 + lane;
+      int row = piece >> 3;
+      int k8 = ((piece & 7) ^ (row & 7)) * 8;
+      int pa = pix0 + row;
+      int pc = pa < a.M ? pa : a.M - 1;
+      long dyaddr = (long)pc * a.Kphys + g * a.Kgp + kout0 + k8;
       if (dyaddr + 8 > dyTotal) dyaddr = dyTotal - 8;
-      if (interior) {
-        glds16(a.DY + dyaddr, &DYt[buf][(wave * 4 + i) * 8][0]);
-        glds16(a.X + xbase + xoff, &XT[buf][(wave * 4 + i) * 8][0]);
+      if (interior || pa < a.M) {
+        glds16(a.DY + dyaddr, &DYt[buf][(wave * DI + i) * 8][0]);
       } else {
-        // tail chunk: dy rows past M must be EXACT zeros
+        // tail: rows past M must be EXACT zeros (summed dimension)
         pack16 v;
-        if (pa < a.M) {
-          v = *(const pack16*)(a.DY + dyaddr);
-        } else {
-          v.u[0] = v.u[1] = v.u[2] = v.u[3] = 0;
-        }
-        // LDS slot = source piece XOR (row&7)  (same as the glds path)
-        *(pack16*)&DYt[buf][row][k8Of[i] ^ ((row & 7) * 8)] = v;
-        pack16 w = *(const pack16*)(a.X + xbase + xoff);
-        *(pack16*)&XT[buf][row][k8Of[i] ^ ((row & 7) * 8)] = w;
+        v.u[0] = v.u[1] = v.u[2] = v.u[3] = 0;
+        *(pack16*)&DYt[buf][row][k8] = v;
       }
     }
   };
+  const int wm = wave >> 1, wn = wave & 1;  // 2x2 waves
+  constexpr int FM = 1;                     // 32 kout rows per wave
+  constexpr int FN = BNF / 2 / 16;          // 32(BNF=64) or 64(128)/16
+  const int m0 = wm * 32, n0 = wn * (BNF / 2);
 
-  const int wm = wave >> 1, wn = wave & 1;  // 2x2 waves, 32x32 each
-  const int m0 = wm * 32, n0 = wn * 32;
-
-  f32x4 acc[2][2];
+  f32x4 acc[FM * 2][FN];
 #pragma unroll
   for (int i = 0; i < 2; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   const int nchunks = (a.M + BKP - 1) / BKP;
-  int ci = slice;
-  if (ci < nchunks) {
-    stage(0, ci * BKP);
-    __syncthreads();
-    int t = 0;
-    for (; ci < nchunks; ci += a.splitK, ++t) {
-      int cnext = ci + a.splitK;
-      if (cnext < nchunks) stage((t + 1) & 1, cnext * BKP);
-      const int buf = t & 1;
+  const int nci = (nchunks - slice + a.splitK - 1) / a.splitK;
+  if (slice < nchunks) {
+    stage(0, slice * BKP);
+    for (int ci = 0; ci < nci; ++ci) {
+      if (ci + 1 < nci)
+        stage((ci + 1) & 1, (slice + (long)(ci + 1) * a.splitK) * BKP);
+      if (ci + 1 < nci)
+        asm volatile("s_waitcnt vmcnt(%0) lgkmcnt(0)" ::"i"(TPW)
+                     : "memory");
+      else
+        asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_s_barrier();
+      const int buf = ci & 1;
 #pragma unroll
       for (int ks = 0; ks < BKP; ks += 32) {
-        bf16x8 af[2], bfr[2];
+        bf16x8 af[2], bfr[FN];
 #pragma unroll
         for (int i = 0; i < 2; ++i)
-          af[i] = frag_load_tr_swz(&DYt[buf][0][0], ks, m0 + i * 16,
-                                   lane);
+          af[i] = frag_load_tr_swz<64>(&DYt[buf][0][0], ks,
+                                       m0 + i * 16, lane);
 #pragma unroll
-        for (int j = 0; j < 2; ++j)
-          bfr[j] = frag_load_tr_swz(&XT[buf][0][0], ks, n0 + j * 16,
-                                    lane);
+        for (int j = 0; j < FN; ++j)
+          bfr[j] = frag_load_tr_swz<BNF>(&XT[buf][0][0], ks,
+                                         n0 + j * 16, lane);
 #pragma unroll
         for (int i = 0; i < 2; ++i)
 #pragma unroll
-          for (int j = 0; j < 2; ++j)
+          for (int j = 0; j < FN; ++j)
             acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 af[i], bfr[j], acc[i][j], 0, 0, 0);
       }
-      __syncthreads();
+      __builtin_amdgcn_s_barrier();
     }
   }
 
@@ -377,7 +389,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
       if (kout >= a.Kgp) continue;
       long base = ((long)g * a.Kgp + kout) * a.Kdf + fd0;
 #pragma unroll
-      for (int j = 0; j < 2; ++j) {
+      for (int j = 0; j < FN; ++j) {
         int fd = n0 + j * 16 + (lane & 15);
         atomicAdd(&a.DW[base + fd], acc[i][j][reg]);
       }
@@ -487,14 +499,25 @@ hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
 hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
                              hipStream_t stream) {
   WgradArgs a = *args;
-  dim3 grid(cdiv(a.Kgp, 64), cdiv(a.Kdf, 64), G * a.splitK);
+  if (a.Kdf > 2048 * 16) return hipErrorInvalidValue;
+  int BNF = (a.Kdf % 128 == 0) ? 128 : 64;
+  dim3 grid(cdiv(a.Kgp, 64), cdiv(a.Kdf, BNF), G * a.splitK);
   dim3 block(256);
-  if (layout == FRAG_CONTIG8)
-    hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8>), grid, block, 0,
-                       stream, a);
-  else
-    hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES>), grid, block, 0,
-                       stream, a);
+  if (BNF == 128) {
+    if (layout == FRAG_CONTIG8)
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, 128>), grid, block,
+                         0, stream, a);
+    else
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, 128>), grid, block,
+                         0, stream, a);
+  } else {
+    if (layout == FRAG_CONTIG8)
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, 64>), grid, block,
+                         0, stream, a);
+    else
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, 64>), grid, block,
+                         0, stream, a);
+  }
   return hipGetLastError();
 }
 
