@@ -38,8 +38,11 @@ def main():
     ap.add_argument("--model", default="alexnet", choices=sorted(MODELS))
     ap.add_argument("--batch", type=int, default=0,
                     help="per-GPU batch (0 = model default)")
-    ap.add_argument("--sync-type", default="cdd",
-                    choices=["cdd", "avg", "grad"])
+    ap.add_argument("--sync-type", default="grad",
+                    choices=["cdd", "avg", "grad"],
+                    help="BSP sync rule; 'grad' == reference cdd for "
+                         "SGD+momentum (averaged deltas), with the "
+                         "all-reduce overlapped into backward")
     ap.add_argument("--strategy", default="rccl_bf16")
     ap.add_argument("--bucket-mb", type=float, default=25)
     ap.add_argument("--no-overlap", action="store_true",

@@ -86,9 +86,13 @@ def run():
     if exch._pending_send is not None:
         exch._pending_send.wait()
     store.add("go/done", 1)
+    t0 = time.time()
     while int(store.add("go/done", 0)) < proc.size:
         drain()
         time.sleep(0.001)
+        if time.time() - t0 > 300:
+            raise RuntimeError("gosgd shutdown drain timed out "
+                               "(peer likely died)")
     drain()                                   # announced-but-unreceived
     proc.barrier()
 
