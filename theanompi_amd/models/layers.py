@@ -285,12 +285,10 @@ class _HipConvFn(torch.autograd.Function):
         # weight grad (fp32, flat layout)
         dw = torch.zeros(plan.flat_shape(), dtype=torch.float32,
                          device=dy.device)
-        C.conv_wgrad(xp, dy, plan.table, dw,
+        C.conv_wgrad(xp, dy, plan.table, dw, plan.kdf_mask,
                      plan.Wp * plan.Cphys, plan.Cphys,
                      plan.stride[0], plan.stride[1],
                      plan.Cg if plan.G > 1 else 0)
-        if plan.kdf_mask is not None:
-            dw.mul_(plan.kdf_mask)   # keep padded flat slots exactly zero
         dx = None
         if ctx.needs_input_grad[0]:
             dx = layer._dgrad(dy)

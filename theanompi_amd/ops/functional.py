@@ -174,7 +174,7 @@ class _LRNFn(torch.autograd.Function):
         C = _ops.ext()
         y = torch.empty_like(x)
         scale = torch.empty(x.shape, dtype=torch.float32, device=x.device)
-        C.lrn_fwd(x, y, scale, size, alpha, beta, k)
+        C.lrn_fwd(x, y, scale, size, alpha, beta, k)   # scale = 1/denom
         ctx.save_for_backward(x, y, scale)
         ctx.meta = (size, alpha, beta, k)
         return y

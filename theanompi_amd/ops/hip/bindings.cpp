@@ -34,6 +34,7 @@ struct WgradArgs {
   const void* X;
   const void* DY;
   const int* table;
+  const float* kdfMask;
   float* DW;
   int M, P, Q;
   long inNS, inPS, inQS;
@@ -357,11 +358,13 @@ static void conv_dgrad(torch::Tensor dyp, torch::Tensor wdg,
 
 static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
                        torch::Tensor table, torch::Tensor dw,
+                       torch::Tensor kdfMask,
                        long rowStride, long colStride, long sh, long sw,
                        long groupInOff) {
   check_bf16(xp, "xp");
   check_bf16(dy, "dy");
   check_f32(dw, "dw");
+  check_f32(kdfMask, "kdfMask");
   long N = dy.size(0), P = dy.size(1), Q = dy.size(2), Kphys = dy.size(3);
   long G = dw.size(0), Kgp = dw.size(1), Kdf = dw.size(2);
   TORCH_CHECK(Kphys == G * Kgp);
@@ -371,6 +374,7 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
   a.X = xp.data_ptr();
   a.DY = dy.data_ptr();
   a.table = table.data_ptr<int>();
+  a.kdfMask = kdfMask.data_ptr<float>();
   a.DW = dw.data_ptr<float>();
   a.M = (int)(N * P * Q);
   a.P = (int)P;

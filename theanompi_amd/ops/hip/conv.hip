@@ -248,6 +248,7 @@ struct WgradArgs {
   const __bf16* __restrict__ X;     // halo-padded input
   const __bf16* __restrict__ DY;    // [M][Kphys]
   const int* __restrict__ table;    // [Kdf/16]
+  const float* __restrict__ kdfMask;  // [Kdf] 1/0: zero padded slots
   float* __restrict__ DW;           // [G][Kgp][Kdf]
   int M, P, Q;
   long inNS, inPS, inQS;
@@ -309,7 +310,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
       int n = tmp / a.P;
       long base = (long)n * a.inNS + (long)p * a.inPS
                   + (long)q * a.inQS + gInOff;
-      int off = Tab[(fd0 + k8) >> 4] + (k8 & 15);
+      int ti2 = (fd0 + k8) >> 4;
+      int tmax = (a.Kdf >> 4) - 1;
+      if (ti2 > tmax) ti2 = tmax;        // BNF tail: masked out below
+      int off = Tab[ti2] + (k8 & 15);
       glds16(a.X + base + off,
              &XT[buf][0][0] + (long)(wave * XI + i) * 512);
     }
@@ -391,7 +395,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
 #pragma unroll
       for (int j = 0; j < FN; ++j) {
         int fd = n0 + j * 16 + (lane & 15);
-        atomicAdd(&a.DW[base + fd], acc[i][j][reg]);
+        if (fd0 + fd >= a.Kdf) continue;
+        // fold the pad-slot mask in here (saves a full dw pass)
+        atomicAdd(&a.DW[base + fd],
+                  acc[i][j][reg] * a.kdfMask[fd0 + fd]);
       }
     }
   }
@@ -500,7 +507,7 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
                              hipStream_t stream) {
   WgradArgs a = *args;
   if (a.Kdf > 2048 * 16) return hipErrorInvalidValue;
-  int BNF = (a.Kdf % 128 == 0) ? 128 : 64;
+  int BNF = (a.Kdf % 128 == 0) ? 128 : 64;  // 128-tile halves occupancy; only pays on dense-Kdf layers
   dim3 grid(cdiv(a.Kgp, 64), cdiv(a.Kdf, BNF), G * a.splitK);
   dim3 block(256);
   if (BNF == 128) {
