@@ -253,3 +253,24 @@ def test_cifar10_data_object():
     x, y = d.get_train_batch(0)
     assert x.shape == (16, 32, 32, 3)
     assert d.n_batch_train == 4
+
+
+def test_exchanger_per_tensor_mode():
+    """Reference parity: tensor-by-tensor exchange."""
+    import torch.distributed as dist
+    from theanompi_amd.lib.exchanger import BSP_Exchanger
+    ps = [torch.ones(5), torch.full((3, 3), 2.0)]
+    ex = BSP_Exchanger({"sync_type": "avg", "size": 1,
+                        "per_tensor": True}, ps, None)
+    ex.exchange()            # size 1: no-op, but exercises the path
+    assert torch.allclose(ps[0], torch.ones(5))
+
+
+def test_strategy_aliases():
+    from theanompi_amd.lib.exchanger_strategy import make_strategy
+    for name in ("ar", "nccl32", "rccl32", "nccl16", "rccl_bf16",
+                 "asa", "asa32", "asa16", "asa_bf16"):
+        assert make_strategy(name) is not None
+    import pytest as _pt
+    with _pt.raises(ValueError):
+        make_strategy("bogus")
