@@ -20,6 +20,7 @@
 // fp32 atomic accumulation).
 
 #include "common.hpp"
+#include <cstdlib>
 
 // ---------------------------------------------------------------------------
 // conv_gather_gemm: fwd / dgrad / linear
@@ -43,15 +44,16 @@ struct GatherGemmArgs {
 };
 
 template <int BM, int BN, int LAYOUT, int DEPTH>
-__global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
+__global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
   constexpr int BK = 64;
-  constexpr int WAVES_M = (BN == 128) ? 2 : 4;
-  constexpr int WAVES_N = BN / 64;       // 2 or 1
+  constexpr int WAVES = (BM == 256) ? 8 : 4;     // block = WAVES*64 thr
+  constexpr int WAVES_N = (BN == 128) ? 2 : 1;
+  constexpr int WAVES_M = WAVES / WAVES_N;
   constexpr int TM = BM / WAVES_M;       // 64 or 32
   constexpr int FM = TM / 16;            // 4 or 2
   constexpr int FN = 64 / 16;            // wave tile N = 64 always
-  constexpr int AI = BM / 32;            // glds instrs per wave (A tile)
-  constexpr int BI = BN / 32;
+  constexpr int AI = BM / (8 * WAVES);   // glds instrs per wave (A tile)
+  constexpr int BI = BN / (8 * WAVES);
   constexpr int TPW = AI + BI;           // glds per wave per tile
 
   // lane-linear tiles (glds writes base + lane*16B); content is
@@ -88,7 +90,7 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
 
   {  // table -> LDS (bounded by binding: Kdf/16 <= 2048)
     const int n = a.Kdf >> 4;
-    for (int i = tid; i < n; i += 256) Tab[i] = a.table[i];
+    for (int i = tid; i < n; i += WAVES * 64) Tab[i] = a.table[i];
   }
 
   // --- per-lane glds descriptors (fixed across K) --------------------------
@@ -139,6 +141,7 @@ __global__ __launch_bounds__(256) void conv_gather_gemm(GatherGemmArgs a) {
   const int wn = (WAVES_N == 2) ? (wave & 1) : 0;
   const int m0 = wm * TM;
   const int n0 = wn * 64;
+  static_assert(WAVES_M * TM == BM && WAVES_N * 64 == BN, "tiling");
   const int lrow = lane & 15;
 
   f32x4 acc[FM][FN];
@@ -475,6 +478,25 @@ hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
                                    int layout, hipStream_t stream) {
   GatherGemmArgs a = *args;
   if (a.Kdf > 2048 * 16) return hipErrorInvalidValue;  // LDS table cap
+  // big-tile path: 256x128, 8 waves -- 2x MFMA work per staged byte,
+  // for large-M dense layers
+  static int bigTile = -1;
+  if (bigTile < 0) {
+    const char* e = getenv("THEANOMPI_GEMM_BIG");
+    bigTile = e ? atoi(e) : 0;   // needs the 8-phase schedule to pay
+  }
+  if (bigTile && a.splitK == 1 && a.M >= 4096 && a.BNtot >= 128
+      && a.Kdf >= 128) {
+    dim3 grid(cdiv(a.M, 256), cdiv(a.BNtot, 128), G);
+    dim3 block(512);
+    if (layout == FRAG_CONTIG8)
+      hipLaunchKernelGGL((conv_gather_gemm<256, 128, FRAG_CONTIG8, 2>),
+                         grid, block, 0, stream, a);
+    else
+      hipLaunchKernelGGL((conv_gather_gemm<256, 128, FRAG_HALVES, 2>),
+                         grid, block, 0, stream, a);
+    return hipGetLastError();
+  }
   int BN = a.BNtot > 64 ? 128 : 64;
   dim3 grid(cdiv(a.M, 128), cdiv(a.BNtot, BN), G * a.splitK);
   dim3 block(256);
