@@ -228,6 +228,193 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
   }
 }
 
+// ---------------------------------------------------------------------------
+// 8-phase deep-pipelined gather-GEMM (guide T3+T4+T5): 256x128 tile,
+// 8 waves, 3-tile glds ring, counted vmcnt(6) once per K-tile (never 0
+// in the loop), per-phase {ds_read || glds issue || MFMA cluster} with
+// raw barriers and setprio around the MFMA cluster.
+// ---------------------------------------------------------------------------
+
+template <int LAYOUT>
+__global__ __launch_bounds__(512) void conv_gather_gemm_8p(
+    GatherGemmArgs a) {
+  constexpr int BM = 256, BN = 128, BK = 64, WAVES = 8;
+  constexpr int AI = BM / (8 * WAVES);   // 4 glds per wave (A)
+  constexpr int BI = BN / (8 * WAVES);   // 2 glds per wave (B)
+  constexpr int TPW = AI + BI;           // 6
+
+  __shared__ __align__(16) __bf16 As[3][BM][BK];
+  __shared__ __align__(16) __bf16 Bs[3][BN][BK];
+  __shared__ int Tab[2048];
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  int mTile = blockIdx.x;
+  {
+    const int nwg = gridDim.x;
+    if (nwg >= 16) {
+      const int q = nwg >> 3, r = nwg & 7;
+      const int xcd = mTile & 7, idx = mTile >> 3;
+      mTile = (xcd < r) ? xcd * (q + 1) + idx
+                        : r * (q + 1) + (xcd - r) * q + idx;
+    }
+  }
+  const int nTile = blockIdx.y, g = blockIdx.z;
+
+  const __bf16* Wg = a.W + (long)g * a.BNtot * a.Kdf;
+  const long gInOff = (long)g * a.groupInOff;
+  const int bn0 = nTile * BN;
+
+  {
+    const int n = a.Kdf >> 4;
+    for (int i = tid; i < n; i += 512) Tab[i] = a.table[i];
+  }
+
+  long aRowBase[AI];
+  int aK8x8[AI];
+#pragma unroll
+  for (int i = 0; i < AI; ++i) {
+    int piece = (wave * AI + i) * 64 + lane;
+    int row = piece >> 3;
+    int k8 = (piece & 7) ^ (row & 7);
+    int m = mTile * BM + row;
+    if (m >= a.M) m = a.M - 1;
+    int q = m % a.Q;
+    int tmp = m / a.Q;
+    int p = tmp % a.P;
+    int n = tmp / a.P;
+    aRowBase[i] = (long)n * a.inNS + (long)p * a.inPS + (long)q * a.inQS
+                  + gInOff;
+    aK8x8[i] = k8 * 8;
+  }
+  const __bf16* bSrc[BI];
+#pragma unroll
+  for (int i = 0; i < BI; ++i) {
+    int piece = (wave * BI + i) * 64 + lane;
+    int row = piece >> 3;
+    int k8 = (piece & 7) ^ (row & 7);
+    int n = bn0 + row;
+    if (n >= a.BNtot) n = a.BNtot - 1;
+    bSrc[i] = Wg + (long)n * a.Kdf + k8 * 8;
+  }
+  __syncthreads();
+
+  // glds issue for one (tile, sub-step) pair: sub 0..2 issues 2 glds
+  // each (A0+A1, A2+A3, B0+B1)
+  auto stage2 = [&](int buf, int k0, int sub) {
+    if (sub < 2) {
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int ii = sub * 2 + i;
+        int off = Tab[(k0 + aK8x8[ii]) >> 4] + (aK8x8[ii] & 15);
+        glds16(a.X + aRowBase[ii] + off,
+               &As[buf][(wave * AI + ii) * 8][0]);
+      }
+    } else {
+#pragma unroll
+      for (int i = 0; i < BI; ++i) {
+        glds16(bSrc[i] + k0, &Bs[buf][(wave * BI + i) * 8][0]);
+      }
+    }
+  };
+
+  const int wm = wave >> 1;          // 4 x 2 wave grid
+  const int wn = wave & 1;
+  const int m0 = wm * 64;
+  const int n0 = wn * 64;
+  const int lrow = lane & 15;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  const int nt = a.Kdf / BK;
+  // prologue: tiles 0 and 1 in flight
+  for (int sub = 0; sub < 3; ++sub) stage2(0, 0, sub);
+  if (nt > 1)
+    for (int sub = 0; sub < 3; ++sub) stage2(1, BK, sub);
+  if (nt > 1)
+    asm volatile("s_waitcnt vmcnt(%0)" ::"i"(TPW) : "memory");
+  else
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < nt; ++t) {
+    const int buf = t % 3;
+    const int nbuf = (t + 2) % 3;
+    const bool more = (t + 2) < nt;
+
+    bf16x8 bf[4];                    // B frags for current ks
+#pragma unroll
+    for (int ph = 0; ph < 4; ++ph) {
+      const int ks = (ph >> 1) * 32;
+      const int f0 = (ph & 1) * 2;
+      bf16x8 af[2];
+      if ((ph & 1) == 0) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          int row = n0 + j * 16 + lrow;
+          bf[j] = frag_load_swz<LAYOUT>(&Bs[buf][row][0], row, ks, lane);
+        }
+      }
+#pragma unroll
+      for (int i = 0; i < 2; ++i) {
+        int row = m0 + (f0 + i) * 16 + lrow;
+        af[i] = frag_load_swz<LAYOUT>(&As[buf][row][0], row, ks, lane);
+      }
+      if (more && ph < 3) stage2(nbuf, (t + 2) * BK, ph);
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[f0 + i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[i], bf[j], acc[f0 + i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      if (ph == 3) {
+        // wait for tile t+1 here, one barrier ahead of its first read
+        if (more)
+          asm volatile("s_waitcnt vmcnt(%0)" ::"i"(TPW) : "memory");
+        else if (t + 1 < nt)
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+    }
+  }
+
+  const int cOut0 = g * a.BNtot + bn0;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int reg = 0; reg < 4; ++reg) {
+      int rowl = m0 + i * 16 + (lane >> 4) * 4 + reg;
+      int m = mTile * BM + rowl;
+      if (m >= a.M) continue;
+      int q = m % a.Q;
+      int tmp = m / a.Q;
+      int p = tmp % a.P;
+      int n = tmp / a.P;
+      long obase = a.outOff + (long)n * a.sN + (long)p * a.sP
+                   + (long)q * a.sQ + cOut0;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int nl = n0 + j * 16 + (lane & 15);
+        if (bn0 + nl >= a.BNtot) continue;
+        float v = acc[i][j][reg];
+        if (a.bias) v += a.bias[cOut0 + nl];
+        if (a.relu) v = v > 0.f ? v : 0.f;
+        a.Y[obase + nl] = (__bf16)v;
+      }
+    }
+  }
+}
+
 // split-K finalize: y = bf16(relu(Y32 + bias))
 __global__ void splitk_finalize_k(const float* __restrict__ y32,
                                   const float* __restrict__ bias,
@@ -489,12 +676,20 @@ hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
       && a.Kdf >= 128) {
     dim3 grid(cdiv(a.M, 256), cdiv(a.BNtot, 128), G);
     dim3 block(512);
-    if (layout == FRAG_CONTIG8)
+    if (bigTile == 2 && a.Kdf >= 3 * 64) {
+      if (layout == FRAG_CONTIG8)
+        hipLaunchKernelGGL((conv_gather_gemm_8p<FRAG_CONTIG8>), grid,
+                           block, 0, stream, a);
+      else
+        hipLaunchKernelGGL((conv_gather_gemm_8p<FRAG_HALVES>), grid,
+                           block, 0, stream, a);
+    } else if (layout == FRAG_CONTIG8) {
       hipLaunchKernelGGL((conv_gather_gemm<256, 128, FRAG_CONTIG8, 2>),
                          grid, block, 0, stream, a);
-    else
+    } else {
       hipLaunchKernelGGL((conv_gather_gemm<256, 128, FRAG_HALVES, 2>),
                          grid, block, 0, stream, a);
+    }
     return hipGetLastError();
   }
   int BN = a.BNtot > 64 ? 128 : 64;
