@@ -69,8 +69,9 @@ class BaseModel:
                 bf16 = getattr(mod, "w_bf16", None)
                 self.recs.append(ParamRec(mod.weight, name + ".W",
                                           bf16_out=bf16))
-                self.recs.append(ParamRec(mod.bias, name + ".b",
-                                          wd_mult=0.0))
+                if mod.bias is not None:
+                    self.recs.append(ParamRec(mod.bias, name + ".b",
+                                              wd_mult=0.0))
                 if mod.gpu:
                     self._post_step_layers.append(mod)
             elif isinstance(mod, L.BatchNorm2d):

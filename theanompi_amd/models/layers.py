@@ -246,6 +246,21 @@ class ConvPlan:
 # GPU autograd Function for conv / linear
 # ---------------------------------------------------------------------------
 
+class _EmptyF32Cache:
+    def __init__(self):
+        self._t = {}
+
+    def get(self, device):
+        t = self._t.get(device)
+        if t is None:
+            t = torch.empty(0, dtype=torch.float32, device=device)
+            self._t[device] = t
+        return t
+
+
+_EMPTY_F32 = _EmptyF32Cache()
+
+
 class _HipConvFn(torch.autograd.Function):
 
     @staticmethod
@@ -258,7 +273,10 @@ class _HipConvFn(torch.autograd.Function):
         M = N * plan.P * plan.Q
         y = torch.empty((N, plan.P, plan.Q, plan.Kphys),
                         dtype=torch.bfloat16, device=x.device)
-        C.conv_fwd(xp, layer.w_bf16, layer.b_f32, plan.table, y,
+        C.conv_fwd(xp, layer.w_bf16,
+                   layer.b_f32 if layer.b_f32 is not None else
+                   _EMPTY_F32.get(x.device),
+                   plan.table, y,
                    plan.Wp * plan.Cphys,            # pixel row stride (h)
                    plan.Cphys,                      # pixel col stride (w)
                    plan.stride[0], plan.stride[1],
@@ -280,8 +298,10 @@ class _HipConvFn(torch.autograd.Function):
             C.relu_bwd(dy, y, dym)
             dy = dym
         # bias grad
-        db = torch.zeros_like(layer.b_f32)
-        C.bias_wgrad(dy, db)
+        db = None
+        if layer.b_f32 is not None:
+            db = torch.zeros_like(layer.b_f32)
+            C.bias_wgrad(dy, db)
         # weight grad (fp32, flat layout)
         dw = torch.zeros(plan.flat_shape(), dtype=torch.float32,
                          device=dy.device)
@@ -337,9 +357,10 @@ class Conv2d(_Base):
 
     def __init__(self, in_shape, K, R, S=None, stride=1, pad=0, groups=1,
                  relu=False, init=("he",), bias_init=0.0, device="cpu",
-                 batch=None, needs_dgrad=True):
+                 batch=None, needs_dgrad=True, has_bias=True):
         super().__init__()
         self.needs_dgrad = needs_dgrad
+        self.has_bias = has_bias
         S = S if S is not None else R
         stride = (stride, stride) if isinstance(stride, int) else stride
         pad = (pad, pad) if isinstance(pad, int) else pad
@@ -366,14 +387,16 @@ class Conv2d(_Base):
             flat = self.plan.w_to_flat(wl.to(self.device_t))
             self.weight = nn.Parameter(flat)
             bp = torch.zeros(self.plan.Kphys, device=self.device_t)
-            for g in range(groups):
-                bp[g * self.plan.Kgp:
-                   g * self.plan.Kgp + self.plan.Kg] = b0[g * self.plan.Kg:
-                                                          (g + 1) * self.plan.Kg]
-            self.bias = nn.Parameter(bp)
+            if has_bias:
+                for g in range(groups):
+                    bp[g * self.plan.Kgp:
+                       g * self.plan.Kgp
+                       + self.plan.Kg] = b0[g * self.plan.Kg:
+                                            (g + 1) * self.plan.Kg]
+            self.bias = nn.Parameter(bp) if has_bias else None
             self.register_buffer("w_bf16",
                                  flat.to(torch.bfloat16).contiguous())
-            self.b_f32 = self.bias.data  # alias used by kernels
+            self.b_f32 = self.bias.data if has_bias else None
             if not needs_dgrad:      # first layer: dx never computed
                 self.plan.dgrad_plans = []
                 self.plan._repack_maps = []
@@ -384,7 +407,7 @@ class Conv2d(_Base):
             self.post_step()          # fill w_bf16 + dgrad repack copies
         else:
             self.weight = nn.Parameter(w0)
-            self.bias = nn.Parameter(b0)
+            self.bias = nn.Parameter(b0) if has_bias else None
             P = (H + 2 * pad[0] - R) // stride[0] + 1
             Q = (W + 2 * pad[1] - S) // stride[1] + 1
             self.out_shape = (P, Q, K)
@@ -456,21 +479,24 @@ class Conv2d(_Base):
             return
         C = _ops.ext()
         self.w_bf16.copy_(self.weight.data)   # fp32 -> bf16 cast copy
-        self.b_f32 = self.bias.data
+        if self.bias is not None:
+            self.b_f32 = self.bias.data
         self.repack_dgrad(C)
 
     def logical_params(self):
         if self.gpu:
-            return [
-                ("W", lambda: self.plan.flat_to_w(self.weight.data)
-                                  [:, :self.Cg_logical],
-                 self._set_w),
-                ("b", lambda: self._get_b(), self._set_b),
-            ]
-        return [("W", lambda: self.weight.data,
-                 lambda v: self.weight.data.copy_(v)),
-                ("b", lambda: self.bias.data,
-                 lambda v: self.bias.data.copy_(v))]
+            out = [("W", lambda: self.plan.flat_to_w(self.weight.data)
+                                     [:, :self.Cg_logical],
+                    self._set_w)]
+            if self.bias is not None:
+                out.append(("b", lambda: self._get_b(), self._set_b))
+            return out
+        out = [("W", lambda: self.weight.data,
+                lambda v: self.weight.data.copy_(v))]
+        if self.bias is not None:
+            out.append(("b", lambda: self.bias.data,
+                        lambda v: self.bias.data.copy_(v)))
+        return out
 
     def _get_b(self):
         plan = self.plan

@@ -19,7 +19,8 @@ class _ConvBN(nn.Module):
         super().__init__()
         self.conv = L.Conv2d((hw, hw, cin), k, r, stride=stride, pad=pad,
                              relu=False, device=dev, batch=B,
-                             needs_dgrad=needs_dgrad, init=("he",))
+                             needs_dgrad=needs_dgrad, init=("he",),
+                             has_bias=False)    # BN supplies the shift
         self.bn = L.BatchNorm2d(self.conv.out_shape[2], relu=relu,
                                 device=dev)
         self.out_hw = self.conv.out_shape[0]

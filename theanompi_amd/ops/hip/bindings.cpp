@@ -250,7 +250,7 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
   check_bf16(xp, "xp");
   check_bf16(w, "w");
   check_bf16(y, "y");
-  check_f32(bias, "bias");
+  if (bias.numel()) check_f32(bias, "bias");
   TORCH_CHECK(table.scalar_type() == torch::kInt32 && table.is_cuda());
   TORCH_CHECK(w.dim() == 3 && xp.dim() == 4 && y.dim() == 4);
   long N = y.size(0), P = y.size(1), Q = y.size(2), Kphys = y.size(3);
@@ -263,7 +263,7 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
   GatherGemmArgs a;
   a.X = xp.data_ptr();
   a.W = w.data_ptr();
-  a.bias = bias.data_ptr<float>();
+  a.bias = bias.numel() ? bias.data_ptr<float>() : nullptr;
   a.table = table.data_ptr<int>();
   a.Y = y.data_ptr();
   a.M = (int)(N * P * Q);
