@@ -28,6 +28,7 @@ struct GatherGemmArgs {
   int relu;
   int splitK;
   float* Y32;
+  long xTotal;
 };
 
 struct WgradArgs {
@@ -41,6 +42,7 @@ struct WgradArgs {
   int groupInOff;
   int Kdf, Kgp, Kphys;
   int splitK;
+  long xTotal;
 };
 
 extern "C" {
@@ -280,6 +282,7 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
   a.Kdf = (int)Kdf;
   a.BNtot = (int)Kgp;
   a.relu = relu ? 1 : 0;
+  a.xTotal = xp.numel();
   long BN = Kgp > 64 ? 128 : 64;
   long baseBlocks = ((a.M + 127) / 128) * ((Kgp + BN - 1) / BN) * G;
   a.splitK = pick_splitk(baseBlocks, Kdf / 64);
@@ -334,6 +337,7 @@ static void conv_dgrad(torch::Tensor dyp, torch::Tensor wdg,
   a.Kdf = (int)Kdf2;
   a.BNtot = (int)BNtot;
   a.relu = 0;
+  a.xTotal = dyp.numel();
   a.splitK = 1;
   a.Y32 = nullptr;
   torch::Tensor y32;
@@ -386,6 +390,7 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
   a.Kdf = (int)Kdf;
   a.Kgp = (int)Kgp;
   a.Kphys = (int)Kphys;
+  a.xTotal = xp.numel();
   long base_blocks = ((Kgp + 63) / 64) * ((Kdf + 63) / 64) * G;
   long splitK = 1024 / (base_blocks ? base_blocks : 1);
   long maxSplit = (a.M + 63) / 64;

@@ -41,6 +41,7 @@ struct GatherGemmArgs {
   int relu;
   int splitK;                       // >1: K-slices atomic-add into Y32
   float* __restrict__ Y32;          // fp32 workspace for split-K
+  long xTotal;                      // X elements (gather tail clamp)
 };
 
 template <int BM, int BN, int LAYOUT, int DEPTH>
@@ -127,8 +128,12 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
 #pragma unroll
     for (int i = 0; i < AI; ++i) {
       int off = Tab[(k0 + aK8x8[i]) >> 4] + (aK8x8[i] & 15);
-      glds16(a.X + aRowBase[i] + off,
-             &As[buf][(wave * AI + i) * 8][0]);
+      long ga = aRowBase[i] + off;
+      // span padding may point past the last pixel of the last image
+      // (stem layers); the values are multiplied by zero weights, the
+      // ADDRESS must stay in bounds.
+      if (ga + 8 > a.xTotal) ga = a.xTotal - 8;
+      glds16(a.X + ga, &As[buf][(wave * AI + i) * 8][0]);
     }
 #pragma unroll
     for (int i = 0; i < BI; ++i) {
@@ -308,8 +313,9 @@ __global__ __launch_bounds__(512) void conv_gather_gemm_8p(
       for (int i = 0; i < 2; ++i) {
         int ii = sub * 2 + i;
         int off = Tab[(k0 + aK8x8[ii]) >> 4] + (aK8x8[ii] & 15);
-        glds16(a.X + aRowBase[ii] + off,
-               &As[buf][(wave * AI + ii) * 8][0]);
+        long ga = aRowBase[ii] + off;
+        if (ga + 8 > a.xTotal) ga = a.xTotal - 8;
+        glds16(a.X + ga, &As[buf][(wave * AI + ii) * 8][0]);
       }
     } else {
 #pragma unroll
@@ -445,6 +451,7 @@ struct WgradArgs {
   int groupInOff;
   int Kdf, Kgp, Kphys;
   int splitK;
+  long xTotal;                      // X elements (gather tail clamp)
 };
 
 template <int LAYOUT, int BNF>
@@ -504,6 +511,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
       int tmax = (a.Kdf >> 4) - 1;
       if (ti2 > tmax) ti2 = tmax;        // BNF tail: masked out below
       int off = Tab[ti2] + (k8 & 15);
+      if (base + off + 8 > a.xTotal) base = a.xTotal - 8 - off;
       glds16(a.X + base + off,
              &XT[buf][0][0] + (long)(wave * XI + i) * 512);
     }
