@@ -47,6 +47,10 @@ def main():
     ap.add_argument("--bucket-mb", type=float, default=25)
     ap.add_argument("--no-overlap", action="store_true",
                     help="disable comm/backward overlap in grad mode")
+    ap.add_argument("--graph", action="store_true",
+                    help="capture the whole train step in a hipGraph "
+                         "(single-GPU; device-seeded dropout keeps "
+                         "masks fresh across replays)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -111,6 +115,33 @@ def main():
             exchanger.exchange()
         model.apply_update()
 
+    graph = None
+    if args.graph:
+        assert on_gpu and world == 1, "--graph: single-GPU only for now"
+        assert args.sync_type == "grad"
+        model.opt.make_persistent()
+        x0, y0 = model.data.get_train_batch(0)
+        x_st, y_st = x0.clone(), y0.clone()
+
+        def graph_body():
+            loss, err = model._fwd_bwd(x_st, y_st)
+            model.opt.step()
+            model._refresh_layers()
+            return loss, err
+
+        for _ in range(3):
+            graph_body()
+        torch.cuda.synchronize()
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            loss_t, err_t = graph_body()
+
+        def step(i):                       # noqa: F811 (graph replay)
+            xi, yi = model.data.get_train_batch(i % 4)
+            x_st.copy_(xi)
+            y_st.copy_(yi)
+            graph.replay()
+
     for i in range(args.warmup):
         step(i)
     barrier()
@@ -154,6 +185,7 @@ def main():
                 "parallelism": f"dp{world}",
                 "sync_type": args.sync_type,
                 "strategy": args.strategy if on_gpu else "ar",
+                "hipgraph": bool(args.graph),
             },
         }
         print(json.dumps(out), flush=True)

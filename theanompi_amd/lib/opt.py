@@ -40,10 +40,30 @@ class SGDMomentum:
         self.momentum = momentum
         self.weight_decay = weight_decay
         self._gpu = any(r.master.is_cuda for r in self.recs)
+        self.persistent_grads = False   # graph mode: stable grad buffers
+        self._desc = None               # cached kernel descriptor
+
+    def make_persistent(self):
+        """Pre-allocate .grad buffers with stable addresses and build the
+        fused-update descriptor once (required for hipGraph capture; also
+        saves the per-step descriptor upload in eager mode).  Changing lr
+        afterwards requires rebuild (adjust_hyperp calls invalidate())."""
+        self.persistent_grads = True
+        for r in self.recs:
+            if r.master.grad is None:
+                r.master.grad = torch.zeros_like(r.master.data)
+        self._desc = None
+
+    def invalidate(self):
+        self._desc = None
 
     # -- helpers --------------------------------------------------------------
 
     def zero_grad(self):
+        if self.persistent_grads:
+            for r in self.recs:
+                r.master.grad.zero_()
+            return
         for r in self.recs:
             if r.master.grad is not None:
                 r.master.grad = None
@@ -124,9 +144,31 @@ class SGDMomentum:
             C.sgd_step_multi(ws, gs, vs, outs, lrs, wds,
                              self.momentum, phase)
 
+    def _cached_step(self, phase):
+        C = _ops.ext()
+        if self._desc is None:
+            ws, gs, vs, outs, lrs, wds = [], [], [], [], [], []
+            empty = torch.empty(0, dtype=torch.bfloat16,
+                                device=self.recs[0].master.device)
+            for r in self.recs:
+                ws.append(r.master.data.view(-1))
+                gs.append(r.master.grad.view(-1))
+                vs.append(r.vel.view(-1))
+                outs.append(r.bf16_out.view(-1)
+                            if r.bf16_out is not None else empty)
+                lrs.append(self.lr * r.lr_mult)
+                wds.append(self.weight_decay * r.wd_mult)
+            self._desc = C.sgd_multi_make(ws, gs, vs, outs, lrs, wds)
+            self._nseg = len(ws)
+            self._maxn4 = max(w.numel() // 4 for w in ws)
+        C.sgd_multi_run(self._desc, self._nseg, self._maxn4,
+                        self.momentum, phase)
+
     def step(self):
         """Full fused step: v = mu*v - lr*(g + wd*w); w += v."""
-        if self._gpu:
+        if self._gpu and self.persistent_grads:
+            self._cached_step(0)
+        elif self._gpu:
             self._batched_gpu(0, lambda r: r.master.grad)
         else:
             for r in self.recs:

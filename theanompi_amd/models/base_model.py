@@ -220,6 +220,8 @@ class BaseModel:
         self.current_lr = self.lr * (0.1 ** decays)
         if self.opt:
             self.opt.lr = self.current_lr
+            if getattr(self.opt, "invalidate", None):
+                self.opt.invalidate()
 
     def lr_steps(self):
         return (20, 40, 60)

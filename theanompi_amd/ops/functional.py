@@ -260,15 +260,24 @@ class _AddFn(torch.autograd.Function):
 
 
 class _DropoutFn(torch.autograd.Function):
-    _seed_counter = [12345]
+    _seed_buf = {}
+
+    @staticmethod
+    def _seed(device):
+        t = _DropoutFn._seed_buf.get(device)
+        if t is None:
+            t = torch.tensor([12345], dtype=torch.int64, device=device)
+            _DropoutFn._seed_buf[device] = t
+        return t
 
     @staticmethod
     def forward(ctx, x, p):
         C = _ops.ext()
         y = torch.empty_like(x)
         mask = torch.empty(x.shape, dtype=torch.uint8, device=x.device)
-        _DropoutFn._seed_counter[0] += 1
-        C.dropout_fwd(x, y, mask, float(p), _DropoutFn._seed_counter[0])
+        # device-resident seed (bumped in-kernel) keeps masks fresh under
+        # hipGraph replay
+        C.dropout_fwd(x, y, mask, float(p), _DropoutFn._seed(x.device))
         ctx.save_for_backward(mask)
         ctx.p = p
         return y

@@ -59,7 +59,8 @@ hipError_t launch_relu_bwd(const void*, const void*, void*, long,
 hipError_t launch_add_relu(const void*, const void*, void*, long, int,
                            hipStream_t);
 hipError_t launch_dropout_fwd(const void*, void*, unsigned char*, long,
-                              float, unsigned long long, hipStream_t);
+                              float, const unsigned long long*,
+                              hipStream_t);
 hipError_t launch_dropout_bwd(const void*, const unsigned char*, void*, long,
                               float, hipStream_t);
 hipError_t launch_gather_bf16(const void*, const int*, void*, long,
@@ -432,13 +433,14 @@ static void add_relu(torch::Tensor a, torch::Tensor b, torch::Tensor y,
 }
 
 static void dropout_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor mask,
-                        double p, long seed) {
+                        double p, torch::Tensor seed) {
   check_bf16(x, "x");
   TORCH_CHECK(mask.scalar_type() == torch::kUInt8);
-  CHECK_HIP(launch_dropout_fwd(x.data_ptr(), y.data_ptr(),
-                               mask.data_ptr<unsigned char>(), x.numel(),
-                               (float)p, (unsigned long long)seed,
-                               cur_stream()));
+  TORCH_CHECK(seed.scalar_type() == torch::kInt64 && seed.is_cuda());
+  CHECK_HIP(launch_dropout_fwd(
+      x.data_ptr(), y.data_ptr(), mask.data_ptr<unsigned char>(),
+      x.numel(), (float)p,
+      (const unsigned long long*)seed.data_ptr(), cur_stream()));
 }
 
 static void dropout_bwd(torch::Tensor dy, torch::Tensor mask,
@@ -702,6 +704,39 @@ static void sgd_step_multi(std::vector<torch::Tensor> ws,
   // synchronously w.r.t. the host, so letting it go out of scope is safe.
 }
 
+// cached variant for graph capture / persistent grads: descriptor
+// built once (device tensor), replayed with one launch.
+static torch::Tensor sgd_multi_make(std::vector<torch::Tensor> ws,
+                                    std::vector<torch::Tensor> gs,
+                                    std::vector<torch::Tensor> vs,
+                                    std::vector<torch::Tensor> outs,
+                                    std::vector<double> lrs,
+                                    std::vector<double> wds) {
+  int n = (int)ws.size();
+  std::vector<SgdSegH> segs(n);
+  for (int i = 0; i < n; ++i) {
+    long nn = ws[i].numel();
+    segs[i] = {ws[i].data_ptr<float>(), gs[i].data_ptr<float>(),
+               vs[i].data_ptr<float>(),
+               outs[i].numel() ? outs[i].data_ptr() : nullptr,
+               nn / 4, nn, (float)lrs[i], (float)wds[i]};
+  }
+  auto buf = torch::empty({(long)(n * sizeof(SgdSegH))},
+                          torch::TensorOptions()
+                              .dtype(torch::kUInt8)
+                              .device(ws[0].device()));
+  hipMemcpyAsync(buf.data_ptr(), segs.data(), n * sizeof(SgdSegH),
+                 hipMemcpyHostToDevice, cur_stream());
+  hipStreamSynchronize(cur_stream());
+  return buf;
+}
+
+static void sgd_multi_run(torch::Tensor desc, long nseg, long max_n4,
+                          double mu, long phase) {
+  CHECK_HIP(launch_sgd_multi(desc.data_ptr(), (int)nseg, max_n4,
+                             (float)mu, (int)phase, cur_stream()));
+}
+
 // static repack set: build the descriptor buffer once...
 static torch::Tensor gather_multi_make(std::vector<torch::Tensor> srcs,
                                        std::vector<torch::Tensor> maps,
@@ -759,6 +794,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_layout", &mfma_layout);
   m.def("tr_probe", &tr_probe);
   m.def("sgd_step_multi", &sgd_step_multi);
+  m.def("sgd_multi_make", &sgd_multi_make);
+  m.def("sgd_multi_run", &sgd_multi_run);
   m.def("gather_multi_make", &gather_multi_make);
   m.def("gather_multi_run", &gather_multi_run);
 }

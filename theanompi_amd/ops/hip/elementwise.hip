@@ -95,9 +95,14 @@ DEV unsigned int mix32(unsigned long long x) {
 __global__ void dropout_fwd_k(const __bf16* __restrict__ x,
                               __bf16* __restrict__ y,
                               unsigned char* __restrict__ mask, long n,
-                              float p, unsigned long long seed) {
+                              float p,
+                              const unsigned long long* __restrict__ seedp) {
+  // seed comes from DEVICE memory so a hipGraph replay (which freezes
+  // kernel args) still draws fresh masks: the companion bump kernel
+  // advances the counter inside the graph.
   const float scale = 1.f / (1.f - p);
   const unsigned int thr = (unsigned int)(p * 4294967296.f);
+  const unsigned long long seed = *seedp;
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (long)gridDim.x * blockDim.x) {
     unsigned int u = mix32(seed * 0x9E3779B97F4A7C15ULL + (unsigned long long)i);
@@ -105,6 +110,10 @@ __global__ void dropout_fwd_k(const __bf16* __restrict__ x,
     mask[i] = keep;
     y[i] = keep ? (__bf16)((float)x[i] * scale) : (__bf16)0.f;
   }
+}
+
+__global__ void seed_bump_k(unsigned long long* seed) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *seed += 1;
 }
 
 __global__ void dropout_bwd_k(const __bf16* __restrict__ dy,
@@ -270,10 +279,13 @@ hipError_t launch_add_relu(const void* p, const void* q, void* y, long n,
 }
 
 hipError_t launch_dropout_fwd(const void* x, void* y, unsigned char* mask,
-                              long n, float p, unsigned long long seed,
+                              long n, float p,
+                              const unsigned long long* seedp,
                               hipStream_t stream) {
+  hipLaunchKernelGGL(seed_bump_k, dim3(1), dim3(64), 0, stream,
+                     (unsigned long long*)seedp);
   hipLaunchKernelGGL(dropout_fwd_k, dim3(ew_grid(n)), dim3(256), 0, stream,
-                     (const __bf16*)x, (__bf16*)y, mask, n, p, seed);
+                     (const __bf16*)x, (__bf16*)y, mask, n, p, seedp);
   return hipGetLastError();
 }
 
