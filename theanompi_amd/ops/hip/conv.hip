@@ -201,34 +201,75 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
     __builtin_amdgcn_s_barrier();              // readers done with buf
   }
 
-  // --- epilogue: bias + relu + strided bf16 store ---------------------------
+  // --- epilogue ------------------------------------------------------------
   const int cOut0 = g * a.BNtot + bn0;
+  if (a.Y32) {
+    // split-K: fp32 atomic partials (bias/relu in the finalize kernel)
+#pragma unroll
+    for (int i = 0; i < FM; ++i) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int rowl = m0 + i * 16 + (lane >> 4) * 4 + reg;
+        int m = mTile * BM + rowl;
+        if (m >= a.M) continue;
+        int q = m % a.Q;
+        int tmp = m / a.Q;
+        int p = tmp % a.P;
+        int n = tmp / a.P;
+        long obase = a.outOff + (long)n * a.sN + (long)p * a.sP
+                     + (long)q * a.sQ + cOut0;
+#pragma unroll
+        for (int j = 0; j < FN; ++j) {
+          int nl = n0 + j * 16 + (lane & 15);
+          if (bn0 + nl >= a.BNtot) continue;
+          atomicAdd(&a.Y32[obase + nl], acc[i][j][reg]);
+        }
+      }
+    }
+    return;
+  }
+  // direct path: stage the bf16 tile through LDS (reusing the staging
+  // buffers) so global stores are full coalesced rows — the natural
+  // per-lane layout writes 32-B scattered runs, which made low-K convs
+  // store-bound.
+  __syncthreads();
+  __bf16* Yt = &As[0][0][0];               // [BM][BN] view, 16/32 KiB
 #pragma unroll
   for (int i = 0; i < FM; ++i) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       int rowl = m0 + i * 16 + (lane >> 4) * 4 + reg;
-      int m = mTile * BM + rowl;
-      if (m >= a.M) continue;
-      int q = m % a.Q;
-      int tmp = m / a.Q;
-      int p = tmp % a.P;
-      int n = tmp / a.P;
-      long obase = a.outOff + (long)n * a.sN + (long)p * a.sP
-                   + (long)q * a.sQ + cOut0;
 #pragma unroll
       for (int j = 0; j < FN; ++j) {
         int nl = n0 + j * 16 + (lane & 15);
-        if (bn0 + nl >= a.BNtot) continue;
         float v = acc[i][j][reg];
-        if (a.Y32) {                       // split-K: fp32 partials;
-          atomicAdd(&a.Y32[obase + nl], v);  // bias/relu in finalize
-        } else {
-          if (a.bias) v += a.bias[cOut0 + nl];
-          if (a.relu) v = v > 0.f ? v : 0.f;
-          a.Y[obase + nl] = (__bf16)v;
-        }
+        if (a.bias) v += a.bias[cOut0 + nl];
+        if (a.relu) v = v > 0.f ? v : 0.f;
+        Yt[rowl * BN + nl] = (__bf16)v;
       }
+    }
+  }
+  __syncthreads();
+  // each lane streams 16-B pieces of consecutive rows out to global
+  constexpr int PIECES = (BM * BN) / 8;    // 8 bf16 per piece
+  constexpr int THREADS = BM * 2;
+  const int nvalid = a.BNtot - bn0;        // columns that exist (< BN ok)
+  for (int pc = tid; pc < PIECES; pc += THREADS) {
+    int rowl = pc / (BN / 8);
+    int c8 = (pc % (BN / 8)) * 8;
+    int m = mTile * BM + rowl;
+    if (m >= a.M || c8 >= nvalid) continue;
+    int q = m % a.Q;
+    int tmp = m / a.Q;
+    int p = tmp % a.P;
+    int n = tmp / a.P;
+    long obase = a.outOff + (long)n * a.sN + (long)p * a.sP
+                 + (long)q * a.sQ + cOut0 + c8;
+    if (c8 + 8 <= nvalid) {
+      *(pack16*)(a.Y + obase) = *(const pack16*)(Yt + rowl * BN + c8);
+    } else {
+      for (int e = 0; e < nvalid - c8; ++e)
+        a.Y[obase + e] = Yt[rowl * BN + c8 + e];
     }
   }
 }
