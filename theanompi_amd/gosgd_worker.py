@@ -10,6 +10,7 @@ staging, so a push never blocks the sender's GPU stream (reference used
 MPI Isend/Irecv + probe, C5).
 """
 
+import os
 import random
 import time
 
@@ -104,6 +105,10 @@ def run():
         n += 1
     if proc.verbose and n:
         rec.val_error(uidx, *(stats / n).tolist())
+    save_dir = _env_opt("save_dir", "")
+    if save_dir and proc.rank == 0:
+        model.save(save_dir, model.epoch)
+        rec.save(os.path.join(save_dir, "inforec.pkl"))
     proc.finalize()
     return model, rec
 
