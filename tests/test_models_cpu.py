@@ -97,3 +97,30 @@ def test_checkpoint_cross_rule(tmp_path):
     m2.load(str(tmp_path), epoch=2)
     for a, b in zip(ref, m2.params):
         assert torch.allclose(a, b, atol=1e-6)
+
+
+@pytest.mark.gpu
+def test_convergence_gpu():
+    """End-to-end learning check: on a LEARNABLE synthetic task (class
+    templates + noise) the bf16 GPU pipeline must reach ~0 train error
+    — exercises every kernel's numerics jointly, not just per-op
+    tolerances."""
+    import torch as T
+    from theanompi_amd.models.cifar10 import Cifar10_model
+    T.manual_seed(0)
+    m = make("theanompi_amd.models.cifar10", "Cifar10_model", 128,
+             device="cuda:0")
+    d = m.data
+    templates = T.randint(0, 256, (10, 32, 32, 4), dtype=T.int16)
+    templates[..., 3:] = 0
+    with T.no_grad():
+        y_all = d.y.cpu()
+        noise = T.randint(-60, 60, d.x.shape, dtype=T.int16)
+        newx = (templates[y_all] + noise).clamp(0, 255).to(T.uint8)
+        newx[..., 3:] = 0
+        d.x.copy_(newx.to(d.x.device))
+    err = 1.0
+    for i in range(200):
+        loss, err = m.train_iter(i % d.n_batch_train)
+        m.apply_update()
+    assert float(err) < 0.05, f"did not learn: err={float(err)}"
