@@ -141,3 +141,29 @@ def test_bsp_overlap_equivalence(tmp_path):
     for k in outs["ov"].files:
         np.testing.assert_allclose(outs["ov"][k], outs["plain"][k],
                                    atol=1e-5)
+
+
+def test_strategy_equivalence(tmp_path):
+    """'ar' and 'asa' produce identical results (same seeded init/data),
+    2 CPU workers."""
+    outs = {}
+    for strat in ("ar", "asa"):
+        d = tmp_path / strat
+        d.mkdir()
+        body = f"""
+        import os
+        os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+        from theanompi_amd import BSP
+        rule = BSP()
+        rule.init(devices=['cpu', 'cpu'],
+                  modelfile='theanompi_amd.models.test_model',
+                  modelclass='TestModel',
+                  epochs=1, sync_type='cdd', exchange_strategy={strat!r},
+                  seed=55, dump_dir={str(d)!r})
+        assert rule.wait() == [0, 0]
+        """
+        _run_session(tmp_path, body)
+        outs[strat] = np.load(os.path.join(str(d), "params_rank0.npz"))
+    for k in outs["ar"].files:
+        np.testing.assert_allclose(outs["ar"][k], outs["asa"][k],
+                                   atol=2e-5)

@@ -8,6 +8,7 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
 #include <cstring>
 #include <vector>
 
@@ -117,7 +118,12 @@ hipError_t launch_splitk_finalize(const float*, const float*, void*, long,
 // pick a K-split so small-M GEMMs (FC layers at batch<=256) still fill
 // the 256-CU chip; partials go through a zeroed fp32 workspace.
 static int pick_splitk(long baseBlocks, long nt) {
-  if (baseBlocks >= 64 || nt < 4) return 1;
+  static long thresh = -1;
+  if (thresh < 0) {
+    const char* e = getenv("THEANOMPI_SPLITK_T");
+    thresh = e ? atol(e) : 128;   // swept 32/64/128/192: 128 best
+  }
+  if (baseBlocks >= thresh || nt < 4) return 1;
   long s = 512 / (baseBlocks ? baseBlocks : 1);
   if (s > nt) s = nt;
   if (s < 1) s = 1;
