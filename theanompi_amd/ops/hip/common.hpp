@@ -62,10 +62,11 @@ DEV bf16x8 frag_load(const __bf16* row_ptr, int lane) {
 // GEMM tiles: LDS slot (row, p8) holds the source 8-element piece
 // p8 ^ (row & 7), so reads XOR the same pattern back (guide T2/rule 21:
 // swizzle on the SOURCE address + the read, LDS destination linear).
-template <int LAYOUT>
+template <int LAYOUT, int BK = 64>
 DEV bf16x8 frag_load_swz(const __bf16* row_start, int row, int ks,
                          int lane) {
-  const int x8 = (row & 7) * 8;
+  // swizzle span = the row's BK elements (BK/8 eight-element slots)
+  const int x8 = (row & (BK / 8 - 1)) * 8;
   if constexpr (LAYOUT == FRAG_CONTIG8) {
     int koff = ks + (lane >> 4) * 8;
     return *(const bf16x8*)(row_start + (koff ^ x8));

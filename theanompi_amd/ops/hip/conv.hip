@@ -44,17 +44,17 @@ struct GatherGemmArgs {
   long xTotal;                      // X elements (gather tail clamp)
 };
 
-template <int BM, int BN, int LAYOUT, int DEPTH>
+template <int BM, int BN, int LAYOUT, int DEPTH, int BK = 64>
 __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
-  constexpr int BK = 64;
   constexpr int WAVES = (BM == 256) ? 8 : 4;     // block = WAVES*64 thr
   constexpr int WAVES_N = (BN == 128) ? 2 : 1;
   constexpr int WAVES_M = WAVES / WAVES_N;
   constexpr int TM = BM / WAVES_M;       // 64 or 32
   constexpr int FM = TM / 16;            // 4 or 2
   constexpr int FN = 64 / 16;            // wave tile N = 64 always
-  constexpr int AI = BM / (8 * WAVES);   // glds instrs per wave (A tile)
-  constexpr int BI = BN / (8 * WAVES);
+  constexpr int PPR = BK / 8;            // 16-B pieces per tile row
+  constexpr int AI = BM * BK / (512 * WAVES);  // glds per wave (A tile)
+  constexpr int BI = BN * BK / (512 * WAVES);
   constexpr int TPW = AI + BI;           // glds per wave per tile
 
   // lane-linear tiles (glds writes base + lane*16B); content is
@@ -100,8 +100,8 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
 #pragma unroll
   for (int i = 0; i < AI; ++i) {
     int piece = (wave * AI + i) * 64 + lane;
-    int row = piece >> 3;
-    int k8 = (piece & 7) ^ (row & 7);          // source-piece swizzle
+    int row = piece / PPR;
+    int k8 = (piece % PPR) ^ (row & (PPR - 1));  // source-piece swizzle
     int m = mTile * BM + row;
     if (m >= a.M) m = a.M - 1;                 // clamped rows discarded
     int q = m % a.Q;
@@ -116,8 +116,8 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
 #pragma unroll
   for (int i = 0; i < BI; ++i) {
     int piece = (wave * BI + i) * 64 + lane;
-    int row = piece >> 3;
-    int k8 = (piece & 7) ^ (row & 7);
+    int row = piece / PPR;
+    int k8 = (piece % PPR) ^ (row & (PPR - 1));
     int n = bn0 + row;
     if (n >= a.BNtot) n = a.BNtot - 1;         // discarded at C-write
     bSrc[i] = Wg + (long)n * a.Kdf + k8 * 8;
@@ -133,11 +133,13 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
       // (stem layers); the values are multiplied by zero weights, the
       // ADDRESS must stay in bounds.
       if (ga + 8 > a.xTotal) ga = a.xTotal - 8;
-      glds16(a.X + ga, &As[buf][(wave * AI + i) * 8][0]);
+      glds16(a.X + ga,
+             &As[buf][0][0] + (wave * AI + i) * 512);
     }
 #pragma unroll
     for (int i = 0; i < BI; ++i) {
-      glds16(bSrc[i] + k0, &Bs[buf][(wave * BI + i) * 8][0]);
+      glds16(bSrc[i] + k0,
+             &Bs[buf][0][0] + (wave * BI + i) * 512);
     }
   };
 
@@ -184,12 +186,14 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
 #pragma unroll
       for (int i = 0; i < FM; ++i) {
         int row = m0 + i * 16 + lrow;
-        af[i] = frag_load_swz<LAYOUT>(&As[buf][row][0], row, ks, lane);
+        af[i] = frag_load_swz<LAYOUT, BK>(&As[buf][row][0], row, ks,
+                                          lane);
       }
 #pragma unroll
       for (int j = 0; j < FN; ++j) {
         int row = n0 + j * 16 + lrow;
-        bf[j] = frag_load_swz<LAYOUT>(&Bs[buf][row][0], row, ks, lane);
+        bf[j] = frag_load_swz<LAYOUT, BK>(&Bs[buf][row][0], row, ks,
+                                          lane);
       }
 #pragma unroll
       for (int i = 0; i < FM; ++i)
@@ -231,7 +235,36 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
   // direct path: stage the bf16 tile through LDS (reusing the staging
   // buffers) so global stores are full coalesced rows — the natural
   // per-lane layout writes 32-B scattered runs, which made low-K convs
-  // store-bound.
+  // store-bound.  (Falls back to scattered stores when the staging
+  // buffers are smaller than the Y tile, e.g. the BK=32 variant.)
+  constexpr bool LDSEP = (2 * BK) >= BN;
+  if constexpr (!LDSEP) {
+#pragma unroll
+    for (int i = 0; i < FM; ++i) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int rowl = m0 + i * 16 + (lane >> 4) * 4 + reg;
+        int m = mTile * BM + rowl;
+        if (m >= a.M) continue;
+        int q = m % a.Q;
+        int tmp = m / a.Q;
+        int p = tmp % a.P;
+        int n = tmp / a.P;
+        long obase = a.outOff + (long)n * a.sN + (long)p * a.sP
+                     + (long)q * a.sQ + cOut0;
+#pragma unroll
+        for (int j = 0; j < FN; ++j) {
+          int nl = n0 + j * 16 + (lane & 15);
+          if (bn0 + nl >= a.BNtot) continue;
+          float v = acc[i][j][reg];
+          if (a.bias) v += a.bias[cOut0 + nl];
+          if (a.relu) v = v > 0.f ? v : 0.f;
+          a.Y[obase + nl] = (__bf16)v;
+        }
+      }
+    }
+    return;
+  }
   __syncthreads();
   __bf16* Yt = &As[0][0][0];               // [BM][BN] view, 16/32 KiB
 #pragma unroll
@@ -744,6 +777,33 @@ hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
   int BN = a.BNtot > 64 ? 128 : 64;
   dim3 grid(cdiv(a.M, 128), cdiv(a.BNtot, BN), G * a.splitK);
   dim3 block(256);
+  static int bk32 = -1;
+  if (bk32 < 0) {
+    const char* e = getenv("THEANOMPI_GEMM_BK32");
+    bk32 = e ? atoi(e) : 0;
+  }
+  if (bk32 && a.Kdf % 64 == 0) {
+    if (BN == 128) {
+      if (layout == FRAG_CONTIG8)
+        hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_CONTIG8, 2,
+                                             32>),
+                           grid, block, 0, stream, a);
+      else
+        hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_HALVES, 2,
+                                             32>),
+                           grid, block, 0, stream, a);
+    } else {
+      if (layout == FRAG_CONTIG8)
+        hipLaunchKernelGGL((conv_gather_gemm<128, 64, FRAG_CONTIG8, 2,
+                                             32>),
+                           grid, block, 0, stream, a);
+      else
+        hipLaunchKernelGGL((conv_gather_gemm<128, 64, FRAG_HALVES, 2,
+                                             32>),
+                           grid, block, 0, stream, a);
+    }
+    return hipGetLastError();
+  }
   const int depth = 2;   // 3-deep ring measured slower (1 blk/CU)
 #define GG_LAUNCH(BN_, L_, D_)                                             \
   hipLaunchKernelGGL((conv_gather_gemm<128, BN_, L_, D_>), grid, block,    \
