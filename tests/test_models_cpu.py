@@ -124,3 +124,13 @@ def test_convergence_gpu():
         loss, err = m.train_iter(i % d.n_batch_train)
         m.apply_update()
     assert float(err) < 0.05, f"did not learn: err={float(err)}"
+
+
+def test_zoo_alias_paths():
+    """Reference-layout import paths keep working."""
+    from theanompi_amd.models.lasagne_model_zoo.vgg import VGG16
+    from theanompi_amd.models.lasagne_model_zoo.resnet50 import ResNet50
+    from theanompi_amd.models.keras_model_zoo.wresnet import WResNet
+    assert VGG16.name == "vgg16"
+    assert ResNet50.name == "resnet50"
+    assert WResNet.name == "wresnet"
