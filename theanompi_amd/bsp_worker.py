@@ -52,6 +52,8 @@ def run():
 
     for epoch in range(n_epochs):
         rec.start_epoch()
+        if epoch > 0:
+            model.reset_iter("train")   # per-epoch shard reshuffle
         n_batches = model.data.n_batch_train
         for i in range(n_batches):
             rec.start()

@@ -111,6 +111,8 @@ class ImageNetData:
                                                device=self.device)
                               for _ in range(depth)]
             self.copy_events = [torch.cuda.Event() for _ in range(depth)]
+            self.done_events = [torch.cuda.Event() for _ in range(depth)]
+            self._slot_used = [False] * depth
 
     def _partition(self):
         shards = list(self.train_shards)
@@ -144,14 +146,22 @@ class ImageNetData:
         self._inflight.remove(slot)
         x_u8, y = self.slots[slot], self.lslots[slot]
         if self.gpu:
+            compute = torch.cuda.current_stream()
             with torch.cuda.stream(self.copy_stream):
+                if self._slot_used[slot]:
+                    # don't overwrite a device slot the compute stream
+                    # may still be reading
+                    self.copy_stream.wait_event(self.done_events[slot])
                 self.dev_bufs[slot].copy_(x_u8, non_blocking=True)
                 self.dev_lbufs[slot].copy_(y, non_blocking=True)
                 self.copy_events[slot].record(self.copy_stream)
-            torch.cuda.current_stream().wait_event(self.copy_events[slot])
+            compute.wait_event(self.copy_events[slot])
             from theanompi_amd import ops as _ops
             x = _ops.ext().u8_finish(self.dev_bufs[slot], 3)
-            return x, self.dev_lbufs[slot].clone()
+            yd = self.dev_lbufs[slot].clone()
+            self.done_events[slot].record(compute)
+            self._slot_used[slot] = True
+            return x, yd
         x = x_u8.float() * (1 / 128.0) - 1.0
         x[..., 3:] = 0
         return x[..., :3] if self.c_phys == 4 and not self.gpu else x, \
