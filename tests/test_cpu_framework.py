@@ -274,3 +274,24 @@ def test_strategy_aliases():
     import pytest as _pt
     with _pt.raises(ValueError):
         make_strategy("bogus")
+
+
+def test_imagenet_epoch_reshuffle(tmp_path):
+    """reset('train') advances the epoch and reshuffles the shard order
+    (per-epoch shard permutation)."""
+    from theanompi_amd.models.data.imagenet import (ImageNetData,
+                                                    write_synthetic_shards)
+    paths = write_synthetic_shards(str(tmp_path), n_shards=6,
+                                   images_per_shard=4, hw=40)
+    d = ImageNetData(paths, paths[:1], batch_size=4, crop=32,
+                     n_classes=10, device="cpu", shuffle=True)
+    try:
+        order0 = list(d.my_shards)
+        d.reset("train")
+        order1 = list(d.my_shards)
+        assert sorted(order0) == sorted(order1)     # same shard set
+        assert order0 != order1                     # new permutation
+        x, y = d.get_train_batch(0)                 # still serves batches
+        assert x.shape[0] == 4
+    finally:
+        d.close()
