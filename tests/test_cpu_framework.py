@@ -295,3 +295,54 @@ def test_imagenet_epoch_reshuffle(tmp_path):
         assert x.shape[0] == 4
     finally:
         d.close()
+
+
+class _DictStore:
+    def __init__(self):
+        self.d = {}
+
+    def set(self, k, v):
+        self.d[k] = v
+
+    def get(self, k):
+        return self.d[k]
+
+    def check(self, keys):
+        return all(k in self.d for k in keys)
+
+
+def test_heartbeat_detects_dead_rank():
+    """A peer whose counter stops advancing trips the watchdog within
+    the timeout; live peers never do."""
+    import time
+    from theanompi_amd.lib.heartbeat import Heartbeat
+    store = _DictStore()
+    dead = []
+    a = Heartbeat(store, 0, 2, timeout=1.0, interval=0.1,
+                  on_dead=dead.append)
+    b = Heartbeat(store, 1, 2, timeout=1.0, interval=0.1,
+                  on_dead=lambda r: dead.append(("b", r)))
+    a.start()
+    b.start()
+    time.sleep(0.8)
+    assert not dead                    # both beating -> no trigger
+    b.stop()                           # rank 1 "dies"
+    t0 = time.time()
+    while not dead and time.time() - t0 < 5:
+        time.sleep(0.05)
+    a.stop()
+    assert dead == [1]                 # a saw rank 1 go silent
+
+
+def test_heartbeat_no_false_abort_in_bsp(tmp_path):
+    """heartbeat=<timeout> as a rule kwarg reaches the workers and a
+    healthy 2-rank BSP run completes normally."""
+    from theanompi_amd import BSP
+    rule = BSP()
+    rule.init(devices=["cpu", "cpu"],
+              modelfile="theanompi_amd.models.test_model",
+              modelclass="TestModel", epochs=1, sync_type="cdd",
+              exchange_strategy="ar", seed=11, heartbeat=20,
+              dump_dir=str(tmp_path))
+    codes = rule.wait()
+    assert all(c == 0 for c in codes)

@@ -11,6 +11,7 @@ import os
 import torch
 import torch.distributed as dist
 
+from theanompi_amd.lib.heartbeat import maybe_start
 from theanompi_amd.lib.base import GPUProcess, _env_opt
 from theanompi_amd.lib.exchanger import (BSP_Exchanger,
                                          OverlappedGradExchanger)
@@ -21,6 +22,7 @@ def run():
     proc = GPUProcess()
     device = proc.init_device()
     proc.init_comm()
+    hb = maybe_start(proc)
     model = proc.build_model()
 
     sync_type = _env_opt("sync_type", "cdd")
@@ -98,6 +100,8 @@ def run():
                 for i, p in enumerate(model.params)}
         np.savez(os.path.join(dump_dir, f"params_rank{proc.rank}.npz"),
                  **arrs)
+    if hb:
+        hb.stop()
     proc.finalize()
     return model, rec
 

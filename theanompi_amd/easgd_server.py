@@ -13,6 +13,7 @@ import time
 
 import torch
 
+from theanompi_amd.lib.heartbeat import maybe_start
 from theanompi_amd.lib.base import GPUProcess, _env_opt
 from theanompi_amd.lib.exchanger import EASGD_Exchanger
 from theanompi_amd.lib.recorder import Recorder
@@ -22,6 +23,7 @@ def run():
     proc = GPUProcess()
     proc.init_device()
     proc.init_comm()
+    hb = maybe_start(proc)
     model = proc.build_model()
     model.compile_iter_fns("avg")        # server never trains; needs params
 
@@ -85,6 +87,8 @@ def run():
         print(f"[easgd server] done: {uidx} worker iters in "
               f"{time.time() - t0:.1f}s", flush=True)
     proc.barrier()
+    if hb:
+        hb.stop()
     proc.finalize()
     return model, rec
 

@@ -7,6 +7,7 @@ apply  w_i -= alpha*(w_i - w~).
 
 import torch
 
+from theanompi_amd.lib.heartbeat import maybe_start
 from theanompi_amd.lib.base import GPUProcess, _env_opt
 from theanompi_amd.lib.exchanger import EASGD_Exchanger
 from theanompi_amd.lib.recorder import Recorder
@@ -16,6 +17,7 @@ def run():
     proc = GPUProcess()
     proc.init_device()
     proc.init_comm()
+    hb = maybe_start(proc)
     model = proc.build_model()
     model.compile_iter_fns("avg")        # full local SGD steps
 
@@ -55,6 +57,8 @@ def run():
             running = False
 
     proc.barrier()
+    if hb:
+        hb.stop()
     proc.finalize()
     return model, rec
 

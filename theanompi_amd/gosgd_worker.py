@@ -17,6 +17,7 @@ import time
 import torch
 import torch.distributed as dist
 
+from theanompi_amd.lib.heartbeat import maybe_start
 from theanompi_amd.lib.base import GPUProcess, _env_opt
 from theanompi_amd.lib.exchanger import GOSGD_Exchanger
 from theanompi_amd.lib.recorder import Recorder
@@ -26,6 +27,7 @@ def run():
     proc = GPUProcess()
     proc.init_device()
     proc.init_comm()
+    hb = maybe_start(proc)
     model = proc.build_model()
     model.compile_iter_fns("avg")
 
@@ -109,6 +111,8 @@ def run():
     if save_dir and proc.rank == 0:
         model.save(save_dir, model.epoch)
         rec.save(os.path.join(save_dir, "inforec.pkl"))
+    if hb:
+        hb.stop()
     proc.finalize()
     return model, rec
 

@@ -1,0 +1,120 @@
+"""Rank-heartbeat watchdog — the failure-detection subsystem SURVEY.md
+§5.3 asks for (the reference has none: a dead MPI rank wedges the whole
+job forever).
+
+Each rank runs a daemon thread that bumps a monotone counter under
+``hb/<rank>`` on the rendezvous TCPStore every ``interval`` seconds and
+watches every peer's counter.  If a peer's counter stops advancing for
+``timeout`` seconds the watchdog prints a diagnostic and aborts the
+process with exit code 13 — a clean, immediate failure instead of a
+hung collective, so benchmarks and long-running jobs never wedge.
+
+Opt-in: pass ``heartbeat=<timeout_seconds>`` to a rule's ``init`` (any
+rule kwarg becomes a worker env opt); 0/absent disables it.  The store
+is the only transport — no extra sockets, no collectives, safe beside
+RCCL/gloo traffic.
+"""
+
+import os
+import sys
+import threading
+import time
+
+
+class Heartbeat:
+    """Store-backed liveness monitor for one worker process."""
+
+    EXIT_CODE = 13
+
+    def __init__(self, store, rank, size, timeout, interval=None,
+                 on_dead=None):
+        self.store = store
+        self.rank = rank
+        self.size = size
+        self.timeout = float(timeout)
+        self.interval = interval if interval else max(0.5,
+                                                      self.timeout / 8.0)
+        self.on_dead = on_dead          # test hook; default aborts
+        self._stop = threading.Event()
+        self._thread = None
+        self._beat = 0
+        # last counter value + local wall time we saw it change, per peer
+        self._seen = {r: (-1, time.monotonic())
+                      for r in range(size) if r != rank}
+
+    # -- lifecycle ------------------------------------------------------------
+
+    def start(self):
+        if self.size < 2 or self.timeout <= 0:
+            return self
+        self._pulse()
+        self._thread = threading.Thread(target=self._run, daemon=True,
+                                        name=f"hb-rank{self.rank}")
+        self._thread.start()
+        return self
+
+    def stop(self):
+        self._stop.set()
+        if self._thread is not None:
+            self._thread.join(timeout=2 * self.interval)
+            self._thread = None
+
+    def __enter__(self):
+        return self.start()
+
+    def __exit__(self, *exc):
+        self.stop()
+        return False
+
+    # -- internals ------------------------------------------------------------
+
+    def _pulse(self):
+        self._beat += 1
+        self.store.set(f"hb/{self.rank}", str(self._beat))
+
+    def _peer_beat(self, r):
+        key = f"hb/{r}"
+        if not self.store.check([key]):
+            return -1                   # peer not registered yet
+        return int(self.store.get(key))
+
+    def _run(self):
+        while not self._stop.wait(self.interval):
+            self._pulse()
+            now = time.monotonic()
+            for r, (last, t_last) in list(self._seen.items()):
+                cur = self._peer_beat(r)
+                if cur != last:
+                    self._seen[r] = (cur, now)
+                elif last >= 0 and now - t_last > self.timeout:
+                    self._dead(r, now - t_last)
+                    return
+
+    def _dead(self, r, silent_s):
+        msg = (f"[heartbeat] rank {self.rank}: rank {r} silent for "
+               f"{silent_s:.1f}s (> {self.timeout:.1f}s) — aborting "
+               f"so the job does not wedge\n")
+        sys.stderr.write(msg)
+        sys.stderr.flush()
+        if self.on_dead is not None:
+            self.on_dead(r)
+            return
+        os._exit(self.EXIT_CODE)
+
+
+def maybe_start(proc):
+    """Start a watchdog from worker env opts; returns it (or None).
+
+    Workers call this right after ``init_comm()``:
+        hb = maybe_start(proc)
+        ...
+        if hb: hb.stop()
+    """
+    from theanompi_amd.lib.base import _env_opt
+    timeout = float(_env_opt("heartbeat", 0) or 0)
+    if timeout <= 0 or proc.size < 2:
+        return None
+    store = proc.control_store()
+    if store is None:
+        return None
+    return Heartbeat(store, proc.rank, proc.size, timeout).start()
