@@ -408,3 +408,36 @@ def test_testmodel_gpu_trains():
         m.apply_update()
         losses.append(float(loss))
     assert losses[-1] < losses[0], losses[:3] + losses[-3:]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("relu,tiny", [(True, False), (False, False),
+                                       (False, True), (True, True)])
+def test_batchnorm_bwd_xhat_from_y(relu, tiny):
+    """BN backward reconstructs xhat from the saved bf16 output when all
+    |gamma| are safe (skipping the x read); a tiny gamma must fall back
+    to the exact x path.  Both paths vs the fp32 reference."""
+    from theanompi_amd.ops import functional as OF
+    torch.manual_seed(9)
+    x = q(torch.randn(16, 7, 7, 64) * 1.5 - 0.2)
+    gamma = torch.randn(64).abs() + 0.5
+    if tiny:
+        gamma[3] = 1e-4                 # trips the allok=0 fallback
+    beta = torch.randn(64) * 0.3
+    d = dev()
+    gg = gamma.to(d).requires_grad_()
+    bg = beta.to(d).requires_grad_()
+    xg = x.to(d).to(torch.bfloat16).requires_grad_()
+    rm_g, rv_g = torch.zeros(64, device=d), torch.ones(64, device=d)
+    y = OF.batchnorm(xg, gg, bg, rm_g, rv_g, True, 0.1, 1e-5, relu=relu)
+    gr = gamma.clone().requires_grad_()
+    br = beta.clone().requires_grad_()
+    xr = x.clone().requires_grad_()
+    yr = OF.batchnorm_ref(xr, gr, br, torch.zeros(64), torch.ones(64),
+                          True, 0.1, 1e-5, relu=relu)
+    dy = q(torch.randn_like(yr))
+    y.backward(dy.to(d).to(torch.bfloat16))
+    yr.backward(dy)
+    assert rel_err(xg.grad, xr.grad) < 5e-2
+    assert rel_err(gg.grad.cpu(), gr.grad) < 5e-2
+    assert rel_err(bg.grad.cpu(), br.grad) < 5e-2

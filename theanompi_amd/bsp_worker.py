@@ -16,6 +16,7 @@ from theanompi_amd.lib.base import GPUProcess, _env_opt
 from theanompi_amd.lib.exchanger import (BSP_Exchanger,
                                          OverlappedGradExchanger)
 from theanompi_amd.lib.recorder import Recorder
+from theanompi_amd.models.base_model import _rtx
 
 
 def run():
@@ -60,15 +61,17 @@ def run():
         for i in range(n_batches):
             rec.start()
             model.train_iter(i, rec)
-            if overlap:
-                exchanger.finish(rec)
-            elif sync_type == "grad":
-                exchanger.set_grad_buffers(model.opt.grads())
-                exchanger.exchange(rec)
-            elif sync_type == "cdd" and hasattr(model, "vel_arena"):
-                exchanger.exchange_flat_inplace([model.vel_arena], rec)
-            else:
-                exchanger.exchange(rec)
+            with _rtx("exchange"):
+                if overlap:
+                    exchanger.finish(rec)
+                elif sync_type == "grad":
+                    exchanger.set_grad_buffers(model.opt.grads())
+                    exchanger.exchange(rec)
+                elif sync_type == "cdd" and hasattr(model, "vel_arena"):
+                    exchanger.exchange_flat_inplace([model.vel_arena],
+                                                    rec)
+                else:
+                    exchanger.exchange(rec)
             rec.start()
             model.apply_update()
             rec.end("calc")

@@ -5,12 +5,35 @@ provide ``build_module()`` + hyperparameters; anything else satisfying
 the duck-type contract also works (the workers only use the contract).
 """
 
+import os
+
 import torch
 
 from theanompi_amd.lib import helper_funcs
 from theanompi_amd.lib.opt import ParamRec, SGDMomentum
 from theanompi_amd.models import layers as L
 from theanompi_amd.ops import functional as OF
+
+
+_ROCTX = os.environ.get("THEANOMPI_AMD_ROCTX", "0") not in ("0", "")
+
+
+class _rtx:
+    """roctx range around an iteration phase — visible in
+    rocprofv3 --marker-trace; torch.cuda.nvtx maps to roctx on ROCm.
+    Enabled with THEANOMPI_AMD_ROCTX=1, zero-cost otherwise."""
+
+    def __init__(self, name):
+        self.name = name
+
+    def __enter__(self):
+        if _ROCTX:
+            torch.cuda.nvtx.range_push(self.name)
+
+    def __exit__(self, *exc):
+        if _ROCTX:
+            torch.cuda.nvtx.range_pop()
+        return False
 
 
 class BaseModel:
@@ -161,10 +184,12 @@ class BaseModel:
     # -- contract: train ------------------------------------------------------
 
     def train_iter(self, count, recorder=None):
-        x, y = self.data.get_train_batch(count)
+        with _rtx("load"):
+            x, y = self.data.get_train_batch(count)
         if recorder:
             recorder.end("wait")
-        loss, err = self._fwd_bwd(x, y)
+        with _rtx("fwd_bwd"):
+            loss, err = self._fwd_bwd(x, y)
         if self.sync_type == "cdd":
             self.opt.phase1()
         elif self.sync_type == "avg":
@@ -187,6 +212,14 @@ class BaseModel:
     def apply_update(self):
         """Called by the worker after the exchange (phase 2 of cdd; no-op
         weight math for avg; full fused step for grad mode)."""
+        ctx = _rtx("update")
+        ctx.__enter__()
+        try:
+            return self._apply_update_inner()
+        finally:
+            ctx.__exit__()
+
+    def _apply_update_inner(self):
         if self.sync_type == "cdd":
             self.opt.phase2()
         elif self.sync_type == "grad":

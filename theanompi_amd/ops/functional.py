@@ -201,20 +201,20 @@ class _BatchNormFn(torch.autograd.Function):
         C.bn_fwd(x, gamma, beta, y, save_mean, save_invstd,
                  running_mean, running_var, float(momentum), float(eps),
                  bool(training), bool(relu))
-        ctx.save_for_backward(x, gamma, save_mean, save_invstd, y)
+        ctx.save_for_backward(x, gamma, beta, save_mean, save_invstd, y)
         ctx.relu = relu
         ctx.training = training
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, gamma, save_mean, save_invstd, y = ctx.saved_tensors
+        x, gamma, beta, save_mean, save_invstd, y = ctx.saved_tensors
         C = _ops.ext()
         dx = torch.empty_like(x)
         dgamma = torch.empty_like(gamma)
         dbeta = torch.empty_like(gamma)
-        C.bn_bwd(dy.contiguous(), x, gamma, save_mean, save_invstd,
-                 y, dx, dgamma, dbeta, bool(ctx.relu))
+        C.bn_bwd(dy.contiguous(), x, gamma, beta, save_mean,
+                 save_invstd, y, dx, dgamma, dbeta, bool(ctx.relu))
         return dx, dgamma, dbeta, None, None, None, None, None, None
 
 

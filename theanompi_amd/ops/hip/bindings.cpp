@@ -97,8 +97,9 @@ hipError_t launch_bn_fwd_eval(const void*, void*, float*, float*,
                               const float*, long, int, float, int,
                               hipStream_t);
 hipError_t launch_bn_bwd(const void*, const void*, const void*, const float*,
-                         const float*, const float*, float*, float*, void*,
-                         long, int, int, hipStream_t);
+                         const float*, const float*, const float*, float*,
+                         int*, float*, float*, void*, long, int, int,
+                         hipStream_t);
 hipError_t launch_softmax_xent_fwd(const void*, const long*, void*, float*,
                                    int, int, int, hipStream_t);
 hipError_t launch_softmax_xent_bwd(const void*, const long*, void*, int, int,
@@ -606,18 +607,25 @@ static void bn_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
 }
 
 static void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma,
-                   torch::Tensor save_mean, torch::Tensor save_invstd,
-                   torch::Tensor y, torch::Tensor dx, torch::Tensor dgamma,
+                   torch::Tensor beta, torch::Tensor save_mean,
+                   torch::Tensor save_invstd, torch::Tensor y,
+                   torch::Tensor dx, torch::Tensor dgamma,
                    torch::Tensor dbeta, bool relu) {
   check_bf16(dy, "dy");
   long C = x.size(-1);
   long rows = x.numel() / C;
   auto sdy = torch::zeros_like(dgamma);
   auto sdyx = torch::zeros_like(dgamma);
+  auto rgamma = torch::empty_like(gamma);
+  auto allok = torch::ones({1}, torch::dtype(torch::kInt32)
+                                    .device(gamma.device()));
   CHECK_HIP(launch_bn_bwd(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
                           save_mean.data_ptr<float>(),
                           save_invstd.data_ptr<float>(),
-                          gamma.data_ptr<float>(), sdy.data_ptr<float>(),
+                          gamma.data_ptr<float>(),
+                          beta.data_ptr<float>(),
+                          rgamma.data_ptr<float>(),
+                          allok.data_ptr<int>(), sdy.data_ptr<float>(),
                           sdyx.data_ptr<float>(), dx.data_ptr(), rows,
                           (int)C, relu ? 1 : 0, cur_stream()));
   dgamma.copy_(sdyx);

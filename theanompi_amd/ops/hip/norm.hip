@@ -240,6 +240,25 @@ __global__ void bn_apply_k(const __bf16* __restrict__ x,
   }
 }
 
+// Backward prep: per-channel 1/gamma plus a single "all channels safe"
+// flag.  When every |gamma| is comfortably away from 0 the backward
+// kernels reconstruct xhat from the SAVED bf16 OUTPUT
+// (xhat = (y - beta) / gamma; relu-masked lanes contribute 0 anyway)
+// and skip the x read entirely — one fewer full-tensor pass in each of
+// bn_bwd_sums / bn_bwd_apply, which are memory-bound (ResNet-50: BN is
+// ~31% of the step).  A tiny |gamma| would amplify y's bf16 rounding,
+// so that case keeps the exact x path (uniform per-layer branch).
+__global__ void bn_gamma_prep_k(const float* __restrict__ gamma,
+                                float* __restrict__ rgamma,
+                                int* __restrict__ allok, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float g = gamma[c];
+  bool ok = fabsf(g) > 1e-2f;
+  rgamma[c] = ok ? 1.f / g : 0.f;
+  if (!ok) atomicAnd(allok, 0);
+}
+
 // backward pass 1: per-channel  sum(dy)  and  sum(dy * xhat)
 // (dy masked by fused relu via y>0 when relu=1)
 __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
@@ -247,6 +266,9 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
                               const __bf16* __restrict__ y,
                               const float* __restrict__ mean,
                               const float* __restrict__ invstd,
+                              const float* __restrict__ beta,
+                              const float* __restrict__ rgamma,
+                              const int* __restrict__ allok,
                               float* __restrict__ sdy,
                               float* __restrict__ sdyx, long rows, int C,
                               int relu) {
@@ -258,24 +280,29 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
   const long nsl = (long)gridDim.y * 8;
   float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float b[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+  const bool useY = (*allok != 0);
   float m[8], is[8];
   if (c8 < c8n) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      m[j] = mean[c8 * 8 + j];
-      is[j] = invstd[c8 * 8 + j];
+      // y path: m := beta, is := 1/gamma  (xhat = (y - beta) / gamma)
+      m[j] = useY ? beta[c8 * 8 + j] : mean[c8 * 8 + j];
+      is[j] = useY ? rgamma[c8 * 8 + j] : invstd[c8 * 8 + j];
     }
     const long co = c8 * 8;
     auto accrow = [&](long r) {
       bf16x8 g8 = *(const bf16x8*)(dy + r * C + co);
-      bf16x8 x8 = *(const bf16x8*)(x + r * C + co);
+      bf16x8 s8;      // xhat source: y (no x traffic) or x (exact)
+      if (useY) s8 = *(const bf16x8*)(y + r * C + co);
+      else      s8 = *(const bf16x8*)(x + r * C + co);
       bf16x8 y8;
-      if (relu) y8 = *(const bf16x8*)(y + r * C + co);
+      if (relu && !useY) y8 = *(const bf16x8*)(y + r * C + co);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = (float)g8[j];
-        if (relu && (float)y8[j] <= 0.f) g = 0.f;
-        float xh = ((float)x8[j] - m[j]) * is[j];
+        float yv = useY ? (float)s8[j] : (relu ? (float)y8[j] : 1.f);
+        if (relu && yv <= 0.f) g = 0.f;
+        float xh = ((float)s8[j] - m[j]) * is[j];
         a[j] += g;
         b[j] += g * xh;
       }
@@ -318,21 +345,33 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
                                const float* __restrict__ mean,
                                const float* __restrict__ invstd,
                                const float* __restrict__ gamma,
+                               const float* __restrict__ beta,
+                               const float* __restrict__ rgamma,
+                               const int* __restrict__ allok,
                                const float* __restrict__ sdy,
                                const float* __restrict__ sdyx,
                                __bf16* __restrict__ dx, long rows, int C,
                                int relu) {
   const float invM = 1.f / rows;
   const long t4 = rows * C / 4;
+  // xhat from y — skip the x read.  Only valid without fused relu:
+  // the mean-subtraction terms need the true xhat at MASKED positions
+  // too, and y=0 there cannot reconstruct it.  (bn_bwd_sums_k has no
+  // such restriction: its xhat is always multiplied by the masked g.)
+  const bool useY = (*allok != 0) && !relu;
   for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < t4;
        i4 += (long)gridDim.x * blockDim.x) {
     long i = i4 * 4;
     int c = (int)(i % C);
     bf16x4 gv4 = *(const bf16x4*)(dy + i);
-    bf16x4 xv4 = *(const bf16x4*)(x + i);
     bf16x4 yv4 = *(const bf16x4*)(y + i);
-    f32x4 mv = *(const f32x4*)(mean + c);
-    f32x4 iv = *(const f32x4*)(invstd + c);
+    bf16x4 xv4;
+    if (!useY) xv4 = *(const bf16x4*)(x + i);
+    f32x4 mv = useY ? *(const f32x4*)(beta + c)
+                    : *(const f32x4*)(mean + c);
+    f32x4 iv = useY ? *(const f32x4*)(rgamma + c)
+                    : *(const f32x4*)(invstd + c);
+    f32x4 ist = *(const f32x4*)(invstd + c);
     f32x4 ga = *(const f32x4*)(gamma + c);
     f32x4 s1 = *(const f32x4*)(sdy + c);
     f32x4 s2 = *(const f32x4*)(sdyx + c);
@@ -341,8 +380,9 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
     for (int j = 0; j < 4; ++j) {
       float g = (float)gv4[j];
       if (relu && (float)yv4[j] <= 0.f) g = 0.f;
-      float xh = ((float)xv4[j] - mv[j]) * iv[j];
-      o[j] = (__bf16)(iv[j] * ga[j]
+      float src = useY ? (float)yv4[j] : (float)xv4[j];
+      float xh = (src - mv[j]) * iv[j];
+      o[j] = (__bf16)(ist[j] * ga[j]
                       * (g - s1[j] * invM - xh * s2[j] * invM));
     }
     *(bf16x4*)(dx + i) = o;
@@ -428,9 +468,12 @@ hipError_t launch_bn_fwd_eval(const void* x, void* y, float* mean,
 
 hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
                          const float* mean, const float* invstd,
-                         const float* gamma, float* sdy, float* sdyx,
-                         void* dx, long rows, int C, int relu,
-                         hipStream_t stream) {
+                         const float* gamma, const float* beta,
+                         float* rgamma, int* allok, float* sdy,
+                         float* sdyx, void* dx, long rows, int C,
+                         int relu, hipStream_t stream) {
+  hipLaunchKernelGGL(bn_gamma_prep_k, dim3(cdiv(C, 256)), dim3(256), 0,
+                     stream, gamma, rgamma, allok, C);
   {
     int gx = cdiv(C / 8, 32);
     long gy = (rows + 31) / 32;
@@ -439,13 +482,13 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
     if (gy < 1) gy = 1;
     hipLaunchKernelGGL(bn_bwd_sums_k, dim3(gx, (unsigned)gy), dim3(256),
                        0, stream, (const __bf16*)dy, (const __bf16*)x,
-                       (const __bf16*)y, mean, invstd, sdy, sdyx, rows,
-                       C, relu);
+                       (const __bf16*)y, mean, invstd, beta, rgamma,
+                       allok, sdy, sdyx, rows, C, relu);
   }
   hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C / 4)), dim3(256),
                      0, stream, (const __bf16*)dy, (const __bf16*)x,
-                     (const __bf16*)y, mean, invstd, gamma, sdy, sdyx,
-                     (__bf16*)dx, rows, C, relu);
+                     (const __bf16*)y, mean, invstd, gamma, beta, rgamma,
+                     allok, sdy, sdyx, (__bf16*)dx, rows, C, relu);
   return hipGetLastError();
 }
 
