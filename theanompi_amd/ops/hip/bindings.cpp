@@ -614,8 +614,8 @@ static void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma,
   check_bf16(dy, "dy");
   long C = x.size(-1);
   long rows = x.numel() / C;
-  auto sdy = torch::zeros_like(dgamma);
-  auto sdyx = torch::zeros_like(dgamma);
+  auto sdy = torch::empty_like(dgamma);     // zeroed by the prep kernel
+  auto sdyx = torch::empty_like(dgamma);
   auto rgamma = torch::empty_like(gamma);
   auto allok = torch::ones({1}, torch::dtype(torch::kInt32)
                                     .device(gamma.device()));

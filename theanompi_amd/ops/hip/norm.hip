@@ -250,9 +250,13 @@ __global__ void bn_apply_k(const __bf16* __restrict__ x,
 // so that case keeps the exact x path (uniform per-layer branch).
 __global__ void bn_gamma_prep_k(const float* __restrict__ gamma,
                                 float* __restrict__ rgamma,
-                                int* __restrict__ allok, int C) {
+                                int* __restrict__ allok,
+                                float* __restrict__ sdy,
+                                float* __restrict__ sdyx, int C) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
+  sdy[c] = 0.f;                 // accumulators for the sums pass
+  sdyx[c] = 0.f;
   float g = gamma[c];
   bool ok = fabsf(g) > 1e-2f;
   rgamma[c] = ok ? 1.f / g : 0.f;
@@ -290,30 +294,52 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
       is[j] = useY ? rgamma[c8 * 8 + j] : invstd[c8 * 8 + j];
     }
     const long co = c8 * 8;
-    auto accrow = [&](long r) {
+    // Two tight per-path row loops behind one runtime-uniform branch:
+    // conditional dual-source loads in a single shared loop defeated
+    // the unroll-4 load pipelining (measured 4x slower).
+    auto accrow_y = [&](long r) {   // xhat from y: dy + y loads only
       bf16x8 g8 = *(const bf16x8*)(dy + r * C + co);
-      bf16x8 s8;      // xhat source: y (no x traffic) or x (exact)
-      if (useY) s8 = *(const bf16x8*)(y + r * C + co);
-      else      s8 = *(const bf16x8*)(x + r * C + co);
-      bf16x8 y8;
-      if (relu && !useY) y8 = *(const bf16x8*)(y + r * C + co);
+      bf16x8 y8 = *(const bf16x8*)(y + r * C + co);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = (float)g8[j];
-        float yv = useY ? (float)s8[j] : (relu ? (float)y8[j] : 1.f);
+        float yv = (float)y8[j];
         if (relu && yv <= 0.f) g = 0.f;
-        float xh = ((float)s8[j] - m[j]) * is[j];
+        float xh = (yv - m[j]) * is[j];
+        a[j] += g;
+        b[j] += g * xh;
+      }
+    };
+    auto accrow_x = [&](long r) {   // exact path (tiny-gamma fallback)
+      bf16x8 g8 = *(const bf16x8*)(dy + r * C + co);
+      bf16x8 x8 = *(const bf16x8*)(x + r * C + co);
+      bf16x8 y8;
+      if (relu) y8 = *(const bf16x8*)(y + r * C + co);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = (float)g8[j];
+        if (relu && (float)y8[j] <= 0.f) g = 0.f;
+        float xh = ((float)x8[j] - m[j]) * is[j];
         a[j] += g;
         b[j] += g * xh;
       }
     };
     long r4 = sl * 4;
-    for (; r4 + 4 <= rows; r4 += nsl * 4) {
+    if (useY) {
+      for (; r4 + 4 <= rows; r4 += nsl * 4) {
 #pragma unroll
-      for (int u = 0; u < 4; ++u) accrow(r4 + u);
+        for (int u = 0; u < 4; ++u) accrow_y(r4 + u);
+      }
+      if (blockIdx.y == 0 && rl == 0)
+        for (long r = rows & ~3L; r < rows; ++r) accrow_y(r);
+    } else {
+      for (; r4 + 4 <= rows; r4 += nsl * 4) {
+#pragma unroll
+        for (int u = 0; u < 4; ++u) accrow_x(r4 + u);
+      }
+      if (blockIdx.y == 0 && rl == 0)
+        for (long r = rows & ~3L; r < rows; ++r) accrow_x(r);
     }
-    if (blockIdx.y == 0 && rl == 0)
-      for (long r = rows & ~3L; r < rows; ++r) accrow(r);
   }
   __shared__ float red[2][8][32][8];
 #pragma unroll
@@ -359,19 +385,41 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
   // too, and y=0 there cannot reconstruct it.  (bn_bwd_sums_k has no
   // such restriction: its xhat is always multiplied by the masked g.)
   const bool useY = (*allok != 0) && !relu;
+  if (useY) {           // non-relu only: dy + y loads, no x traffic
+    for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < t4;
+         i4 += (long)gridDim.x * blockDim.x) {
+      long i = i4 * 4;
+      int c = (int)(i % C);
+      bf16x4 gv4 = *(const bf16x4*)(dy + i);
+      bf16x4 yv4 = *(const bf16x4*)(y + i);
+      f32x4 bv = *(const f32x4*)(beta + c);
+      f32x4 rg = *(const f32x4*)(rgamma + c);
+      f32x4 ist = *(const f32x4*)(invstd + c);
+      f32x4 ga = *(const f32x4*)(gamma + c);
+      f32x4 s1 = *(const f32x4*)(sdy + c);
+      f32x4 s2 = *(const f32x4*)(sdyx + c);
+      bf16x4 o;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float g = (float)gv4[j];
+        float xh = ((float)yv4[j] - bv[j]) * rg[j];
+        o[j] = (__bf16)(ist[j] * ga[j]
+                        * (g - s1[j] * invM - xh * s2[j] * invM));
+      }
+      *(bf16x4*)(dx + i) = o;
+    }
+    return;
+  }
   for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < t4;
        i4 += (long)gridDim.x * blockDim.x) {
     long i = i4 * 4;
     int c = (int)(i % C);
     bf16x4 gv4 = *(const bf16x4*)(dy + i);
-    bf16x4 yv4 = *(const bf16x4*)(y + i);
-    bf16x4 xv4;
-    if (!useY) xv4 = *(const bf16x4*)(x + i);
-    f32x4 mv = useY ? *(const f32x4*)(beta + c)
-                    : *(const f32x4*)(mean + c);
-    f32x4 iv = useY ? *(const f32x4*)(rgamma + c)
-                    : *(const f32x4*)(invstd + c);
-    f32x4 ist = *(const f32x4*)(invstd + c);
+    bf16x4 xv4 = *(const bf16x4*)(x + i);
+    bf16x4 yv4;
+    if (relu) yv4 = *(const bf16x4*)(y + i);
+    f32x4 mv = *(const f32x4*)(mean + c);
+    f32x4 iv = *(const f32x4*)(invstd + c);
     f32x4 ga = *(const f32x4*)(gamma + c);
     f32x4 s1 = *(const f32x4*)(sdy + c);
     f32x4 s2 = *(const f32x4*)(sdyx + c);
@@ -380,9 +428,8 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
     for (int j = 0; j < 4; ++j) {
       float g = (float)gv4[j];
       if (relu && (float)yv4[j] <= 0.f) g = 0.f;
-      float src = useY ? (float)yv4[j] : (float)xv4[j];
-      float xh = (src - mv[j]) * iv[j];
-      o[j] = (__bf16)(ist[j] * ga[j]
+      float xh = ((float)xv4[j] - mv[j]) * iv[j];
+      o[j] = (__bf16)(iv[j] * ga[j]
                       * (g - s1[j] * invM - xh * s2[j] * invM));
     }
     *(bf16x4*)(dx + i) = o;
@@ -473,7 +520,7 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
                          float* sdyx, void* dx, long rows, int C,
                          int relu, hipStream_t stream) {
   hipLaunchKernelGGL(bn_gamma_prep_k, dim3(cdiv(C, 256)), dim3(256), 0,
-                     stream, gamma, rgamma, allok, C);
+                     stream, gamma, rgamma, allok, sdy, sdyx, C);
   {
     int gx = cdiv(C / 8, 32);
     long gy = (rows + 31) / 32;
