@@ -264,7 +264,12 @@ __global__ void bn_gamma_prep_k(const float* __restrict__ gamma,
 }
 
 // backward pass 1: per-channel  sum(dy)  and  sum(dy * xhat)
-// (dy masked by fused relu via y>0 when relu=1)
+// (dy masked by fused relu via y>0 when relu=1).  Templated on the
+// xhat source and launched BOTH ways; the instantiation that does not
+// match the device-side allok flag exits immediately.  A single kernel
+// carrying both loop bodies spilled 68 B/lane of scratch and ran 31%
+// slower than the x path it was meant to beat.
+template <int USEY, int RELU>
 __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
                               const __bf16* __restrict__ x,
                               const __bf16* __restrict__ y,
@@ -274,8 +279,9 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
                               const float* __restrict__ rgamma,
                               const int* __restrict__ allok,
                               float* __restrict__ sdy,
-                              float* __restrict__ sdyx, long rows, int C,
-                              int relu) {
+                              float* __restrict__ sdyx, long rows,
+                              int C) {
+  if ((*allok != 0) != (USEY != 0)) return;   // other path owns this op
   const int c8n = C / 8;
   const int cl = threadIdx.x & 31;
   const int rl = threadIdx.x >> 5;
@@ -284,62 +290,41 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
   const long nsl = (long)gridDim.y * 8;
   float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float b[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
-  const bool useY = (*allok != 0);
   float m[8], is[8];
   if (c8 < c8n) {
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      // y path: m := beta, is := 1/gamma  (xhat = (y - beta) / gamma)
-      m[j] = useY ? beta[c8 * 8 + j] : mean[c8 * 8 + j];
-      is[j] = useY ? rgamma[c8 * 8 + j] : invstd[c8 * 8 + j];
+      // USEY: m := beta, is := 1/gamma  (xhat = (y - beta) / gamma)
+      m[j] = USEY ? beta[c8 * 8 + j] : mean[c8 * 8 + j];
+      is[j] = USEY ? rgamma[c8 * 8 + j] : invstd[c8 * 8 + j];
     }
     const long co = c8 * 8;
-    // Two tight per-path row loops behind one runtime-uniform branch:
-    // conditional dual-source loads in a single shared loop defeated
-    // the unroll-4 load pipelining (measured 4x slower).
-    auto accrow_y = [&](long r) {   // xhat from y: dy + y loads only
+    auto accrow = [&](long r) {
       bf16x8 g8 = *(const bf16x8*)(dy + r * C + co);
-      bf16x8 y8 = *(const bf16x8*)(y + r * C + co);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        float g = (float)g8[j];
-        float yv = (float)y8[j];
-        if (relu && yv <= 0.f) g = 0.f;
-        float xh = (yv - m[j]) * is[j];
-        a[j] += g;
-        b[j] += g * xh;
-      }
-    };
-    auto accrow_x = [&](long r) {   // exact path (tiny-gamma fallback)
-      bf16x8 g8 = *(const bf16x8*)(dy + r * C + co);
-      bf16x8 x8 = *(const bf16x8*)(x + r * C + co);
+      bf16x8 s8;            // xhat source: saved output or input
+      if (USEY) s8 = *(const bf16x8*)(y + r * C + co);
+      else      s8 = *(const bf16x8*)(x + r * C + co);
       bf16x8 y8;
-      if (relu) y8 = *(const bf16x8*)(y + r * C + co);
+      if (!USEY && RELU) y8 = *(const bf16x8*)(y + r * C + co);
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = (float)g8[j];
-        if (relu && (float)y8[j] <= 0.f) g = 0.f;
-        float xh = ((float)x8[j] - m[j]) * is[j];
+        if (RELU) {
+          float yv = USEY ? (float)s8[j] : (float)y8[j];
+          g *= (float)(yv > 0.f);       // VALU mask, no select chain
+        }
+        float xh = ((float)s8[j] - m[j]) * is[j];
         a[j] += g;
         b[j] += g * xh;
       }
     };
     long r4 = sl * 4;
-    if (useY) {
-      for (; r4 + 4 <= rows; r4 += nsl * 4) {
+    for (; r4 + 4 <= rows; r4 += nsl * 4) {
 #pragma unroll
-        for (int u = 0; u < 4; ++u) accrow_y(r4 + u);
-      }
-      if (blockIdx.y == 0 && rl == 0)
-        for (long r = rows & ~3L; r < rows; ++r) accrow_y(r);
-    } else {
-      for (; r4 + 4 <= rows; r4 += nsl * 4) {
-#pragma unroll
-        for (int u = 0; u < 4; ++u) accrow_x(r4 + u);
-      }
-      if (blockIdx.y == 0 && rl == 0)
-        for (long r = rows & ~3L; r < rows; ++r) accrow_x(r);
+      for (int u = 0; u < 4; ++u) accrow(r4 + u);
     }
+    if (blockIdx.y == 0 && rl == 0)
+      for (long r = rows & ~3L; r < rows; ++r) accrow(r);
   }
   __shared__ float red[2][8][32][8];
 #pragma unroll
@@ -363,8 +348,6 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
   }
 }
 
-// backward pass 2: dx = invstd*gamma*(dy - sdy/M - xhat*sdyx/M)
-// and dgamma = sdyx, dbeta = sdy (copied by one thread per channel).
 __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
                                const __bf16* __restrict__ x,
                                const __bf16* __restrict__ y,
@@ -527,10 +510,18 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
     long cap = 1024 / gx; if (cap > 128) cap = 128;
     if (gy > cap) gy = cap;
     if (gy < 1) gy = 1;
-    hipLaunchKernelGGL(bn_bwd_sums_k, dim3(gx, (unsigned)gy), dim3(256),
-                       0, stream, (const __bf16*)dy, (const __bf16*)x,
-                       (const __bf16*)y, mean, invstd, beta, rgamma,
-                       allok, sdy, sdyx, rows, C, relu);
+    auto launch_pair = [&](auto k1, auto k0) {
+      hipLaunchKernelGGL(k1, dim3(gx, (unsigned)gy), dim3(256), 0,
+                         stream, (const __bf16*)dy, (const __bf16*)x,
+                         (const __bf16*)y, mean, invstd, beta, rgamma,
+                         allok, sdy, sdyx, rows, C);
+      hipLaunchKernelGGL(k0, dim3(gx, (unsigned)gy), dim3(256), 0,
+                         stream, (const __bf16*)dy, (const __bf16*)x,
+                         (const __bf16*)y, mean, invstd, beta, rgamma,
+                         allok, sdy, sdyx, rows, C);
+    };
+    if (relu) launch_pair(bn_bwd_sums_k<1, 1>, bn_bwd_sums_k<0, 1>);
+    else      launch_pair(bn_bwd_sums_k<1, 0>, bn_bwd_sums_k<0, 0>);
   }
   hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C / 4)), dim3(256),
                      0, stream, (const __bf16*)dy, (const __bf16*)x,
