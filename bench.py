@@ -27,6 +27,7 @@ MODELS = {
     "resnet50": ("theanompi_amd.models.resnet50", "ResNet50", 256),
     "vgg16": ("theanompi_amd.models.vgg16", "VGG16", 64),
     "cifar10": ("theanompi_amd.models.cifar10", "Cifar10_model", 128),
+    "wresnet": ("theanompi_amd.models.wresnet", "WResNet", 128),
 }
 
 
