@@ -617,8 +617,18 @@ static void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma,
   auto sdy = torch::empty_like(dgamma);     // zeroed by the prep kernel
   auto sdyx = torch::empty_like(dgamma);
   auto rgamma = torch::empty_like(gamma);
-  auto allok = torch::ones({1}, torch::dtype(torch::kInt32)
-                                    .device(gamma.device()));
+  // THEANOMPI_BN_YPATH=0 forces the exact x path (A/B knob): allok
+  // starts at 0, the prep kernel only ever clears it, so the y-path
+  // instantiation always exits.
+  static const bool ypath_on = [] {
+    const char* e = getenv("THEANOMPI_BN_YPATH");
+    return !(e && e[0] == '0');
+  }();
+  auto allok = ypath_on
+      ? torch::ones({1}, torch::dtype(torch::kInt32)
+                             .device(gamma.device()))
+      : torch::zeros({1}, torch::dtype(torch::kInt32)
+                              .device(gamma.device()));
   CHECK_HIP(launch_bn_bwd(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
                           save_mean.data_ptr<float>(),
                           save_invstd.data_ptr<float>(),
