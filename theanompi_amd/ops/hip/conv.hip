@@ -860,7 +860,7 @@ hipError_t launch_bias_wgrad(const void* dy, float* db, long rows, int C,
   int gx = cdiv(c8n, 32);
   long gy = (rows + 31) / 32;              // 4 rows per row-lane pass
   long target = 1024 / gx;                 // bounded atomics per channel
-  if (target > 512) target = 512;          // ~1024 blocks fills 256 CUs
+  if (target > 128) target = 128;
   if (gy > target) gy = target;
   if (gy < 1) gy = 1;
   dim3 grid(gx, (unsigned)gy);

@@ -468,9 +468,7 @@ hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
   {
     int gx = cdiv(C / 8, 32);
     long gy = (rows + 31) / 32;
-    // fill the 256-CU chip: ~1024 blocks total (the old 128-block cap
-    // left half the CUs idle on C=256 layers); atomics stay bounded.
-    long cap = 1024 / gx; if (cap > 512) cap = 512;
+    long cap = 1024 / gx; if (cap > 128) cap = 128;
     if (gy > cap) gy = cap;
     if (gy < 1) gy = 1;
     hipLaunchKernelGGL(bn_sums_k, dim3(gx, (unsigned)gy), dim3(256), 0,
@@ -509,9 +507,12 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
   {
     int gx = cdiv(C / 8, 32);
     long gy = (rows + 31) / 32;
-    // fill the 256-CU chip: ~1024 blocks total (the old 128-block cap
-    // left half the CUs idle on C=256 layers); atomics stay bounded.
-    long cap = 1024 / gx; if (cap > 512) cap = 512;
+    // measured: 512 blocks only pay off for the relu y-path layers
+    // (smaller C, long rows); elsewhere the fixed LDS-reduce + atomic
+    // cost of extra blocks outweighs the parallelism.
+    long cap = 1024 / gx;
+    long lim = relu ? 512 : 128;
+    if (cap > lim) cap = lim;
     if (gy > cap) gy = cap;
     if (gy < 1) gy = 1;
     auto launch_pair = [&](auto k1, auto k0) {

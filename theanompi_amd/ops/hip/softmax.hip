@@ -328,7 +328,7 @@ extern "C" {
 hipError_t launch_sgd_multi(const void* segs_dev, int nseg, long max_n4,
                             float mu, int phase, hipStream_t stream) {
   long gx = (max_n4 + 255) / 256;
-  if (gx > 512) gx = 512;
+  if (gx > 1024) gx = 1024;    // big FC segments dominate; fill the chip
   if (gx < 1) gx = 1;
   dim3 grid((unsigned)gx, nseg);
   hipLaunchKernelGGL(sgd_multi_k, grid, dim3(256), 0, stream,
