@@ -346,3 +346,30 @@ def test_heartbeat_no_false_abort_in_bsp(tmp_path):
               dump_dir=str(tmp_path))
     codes = rule.wait()
     assert all(c == 0 for c in codes)
+
+
+def test_imagenet_mean_image(tmp_path):
+    """Dataset mean-image subtraction ((x - mean)/128, the reference's
+    img_mean path) on the CPU branch."""
+    import numpy as np
+    from theanompi_amd.models.data.imagenet import (ImageNetData,
+                                                    write_synthetic_shards)
+    paths = write_synthetic_shards(str(tmp_path), n_shards=2,
+                                   images_per_shard=4, hw=40)
+    mean = np.full((32, 32, 3), 100.0, dtype=np.float32)
+    mp = str(tmp_path / "mean.npy")
+    np.save(mp, mean)
+    d = ImageNetData(paths, paths[:1], batch_size=4, crop=32,
+                     n_classes=10, device="cpu", shuffle=False,
+                     mean_path=mp)
+    d2 = ImageNetData(paths, paths[:1], batch_size=4, crop=32,
+                      n_classes=10, device="cpu", shuffle=False)
+    try:
+        xm, _ = d.get_val_batch(0)
+        x0, _ = d2.get_val_batch(0)
+        # (u8 - 100)/128  vs  u8/128 - 1: differ by (128-100)/128
+        assert torch.allclose(xm - x0, torch.full_like(xm, 28.0 / 128.0),
+                              atol=1e-5)
+    finally:
+        d.close()
+        d2.close()

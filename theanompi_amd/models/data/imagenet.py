@@ -67,7 +67,7 @@ class ImageNetData:
 
     def __init__(self, train_list, val_list, batch_size, crop, n_classes,
                  device, rank=0, size=1, shuffle=True, c_phys=4,
-                 depth=2):
+                 depth=2, mean_path=None):
         self.train_shards = list(train_list)
         self.val_shards = list(val_list)
         self.batch_size = batch_size
@@ -79,6 +79,19 @@ class ImageNetData:
         self.shuffle = shuffle
         self.c_phys = c_phys
         self.epoch = 0
+        # optional dataset mean image ([crop, crop, 3] fp32 .npy, the
+        # reference's img_mean): subtracted on-GPU inside u8_finish so
+        # the host path still ships uint8.
+        self.mean_dev = None
+        self.mean_host = None
+        if mean_path is not None:
+            m = np.load(mean_path).astype(np.float32)
+            assert m.shape == (crop, crop, 3), m.shape
+            mp = np.zeros((crop, crop, c_phys), dtype=np.float32)
+            mp[..., :3] = m
+            self.mean_host = torch.from_numpy(mp)
+            if self.device.type == "cuda":
+                self.mean_dev = self.mean_host.to(self.device)
         self._partition()
 
         # pinned ring shared with the child process
@@ -157,12 +170,16 @@ class ImageNetData:
                 self.copy_events[slot].record(self.copy_stream)
             compute.wait_event(self.copy_events[slot])
             from theanompi_amd import ops as _ops
-            x = _ops.ext().u8_finish(self.dev_bufs[slot], 3)
+            x = _ops.ext().u8_finish(self.dev_bufs[slot], 3,
+                                     self.mean_dev)
             yd = self.dev_lbufs[slot].clone()
             self.done_events[slot].record(compute)
             self._slot_used[slot] = True
             return x, yd
-        x = x_u8.float() * (1 / 128.0) - 1.0
+        if self.mean_host is not None:
+            x = (x_u8.float() - self.mean_host) * (1 / 128.0)
+        else:
+            x = x_u8.float() * (1 / 128.0) - 1.0
         x[..., 3:] = 0
         return x[..., :3] if self.c_phys == 4 and not self.gpu else x, \
             y.clone()
@@ -181,9 +198,13 @@ class ImageNetData:
         y = torch.from_numpy(labels[:self.batch_size].astype(np.int64))
         if self.gpu:
             from theanompi_amd import ops as _ops
-            return _ops.ext().u8_finish(pad.to(self.device), 3), \
+            return _ops.ext().u8_finish(pad.to(self.device), 3,
+                                        self.mean_dev), \
                 y.to(self.device)
-        xf = pad.float() * (1 / 128.0) - 1.0
+        if self.mean_host is not None:
+            xf = (pad.float() - self.mean_host) * (1 / 128.0)
+        else:
+            xf = pad.float() * (1 / 128.0) - 1.0
         return xf[..., :3], y
 
     def close(self):

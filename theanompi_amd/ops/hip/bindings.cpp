@@ -68,7 +68,8 @@ hipError_t launch_gather_bf16(const void*, const int*, void*, long,
                               hipStream_t);
 hipError_t launch_transpose_bf16(const void*, void*, int, int, long, long,
                                  hipStream_t);
-hipError_t launch_u8_finish(const unsigned char*, void*, long, int, int,
+hipError_t launch_u8_finish(const unsigned char*, void*, const float*,
+                            long, long, int, int,
                             hipStream_t);
 hipError_t launch_pack_bf16(const float*, void*, long, hipStream_t);
 hipError_t launch_unpack_bf16(const void*, float*, long, hipStream_t);
@@ -469,15 +470,25 @@ static void gather_bf16(torch::Tensor src, torch::Tensor map,
                                dst.data_ptr(), dst.numel(), cur_stream()));
 }
 
-static torch::Tensor u8_finish(torch::Tensor src, long nLogC) {
+static torch::Tensor u8_finish(torch::Tensor src, long nLogC,
+                               c10::optional<torch::Tensor> meanimg) {
   TORCH_CHECK(src.scalar_type() == torch::kUInt8 && src.is_cuda()
               && src.is_contiguous());
   auto dst = torch::empty(src.sizes(),
                           src.options().dtype(torch::kBFloat16));
   long C = src.size(-1);
+  long perImg = src.numel() / src.size(0);
+  const float* mp = nullptr;
+  if (meanimg.has_value()) {
+    auto& m = meanimg.value();
+    TORCH_CHECK(m.scalar_type() == torch::kFloat32 && m.is_cuda()
+                && m.is_contiguous() && m.numel() == perImg,
+                "mean image must be fp32 [crop, crop, C] on device");
+    mp = m.data_ptr<float>();
+  }
   CHECK_HIP(launch_u8_finish(src.data_ptr<unsigned char>(),
-                             dst.data_ptr(), src.numel(), (int)C,
-                             (int)nLogC, cur_stream()));
+                             dst.data_ptr(), mp, src.numel(), perImg,
+                             (int)C, (int)nLogC, cur_stream()));
   return dst;
 }
 
@@ -801,7 +812,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_bwd", &dropout_bwd);
   m.def("gather_bf16", &gather_bf16);
   m.def("transpose_bf16", &transpose_bf16);
-  m.def("u8_finish", &u8_finish);
+  m.def("u8_finish", &u8_finish, py::arg("src"), py::arg("nLogC"),
+        py::arg("meanimg") = py::none());
   m.def("pack_bf16", &pack_bf16);
   m.def("unpack_bf16", &unpack_bf16);
   m.def("maxpool_fwd", &maxpool_fwd);

@@ -138,18 +138,26 @@ __global__ void gather_bf16_k(const __bf16* __restrict__ src,
 }
 
 // ---- fused loader finish: uint8 NHWC -> bf16, x/128 - 1, zero pad chans
+// uint8 NHWC -> bf16 on device.  Default normalisation (255-level
+// images): x/128 - 1.  With a mean image (fp32, one per-pixel value
+// per (h, w, c) of the crop, the reference's subtracted dataset mean
+// [pub data/proc_load_mpi.py]): (x - mean) / 128 — the subtraction
+// happens HERE so the host still ships uint8.
 __global__ void u8_finish_k(const unsigned char* __restrict__ src,
-                            __bf16* __restrict__ dst, long total4, int C,
-                            int nLogC) {
+                            __bf16* __restrict__ dst,
+                            const float* __restrict__ meanimg,
+                            long total4, long perImg, int C, int nLogC) {
   for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < total4;
        i4 += (long)gridDim.x * blockDim.x) {
     long i = i4 * 4;
     int c = (int)(i % C);
     unsigned int u = *(const unsigned int*)(src + i);
+    f32x4 mi = {128.f, 128.f, 128.f, 128.f};
+    if (meanimg) mi = *(const f32x4*)(meanimg + (i % perImg));
     bf16x4 o;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
-      float v = (float)((u >> (8 * j)) & 0xff) * (1.f / 128.f) - 1.f;
+      float v = ((float)((u >> (8 * j)) & 0xff) - mi[j]) * (1.f / 128.f);
       o[j] = (c + j < nLogC) ? (__bf16)v : (__bf16)0.f;
     }
     *(bf16x4*)(dst + i) = o;
@@ -298,11 +306,11 @@ hipError_t launch_dropout_bwd(const void* dy, const unsigned char* mask,
 }
 
 hipError_t launch_u8_finish(const unsigned char* src, void* dst,
-                            long total, int C, int nLogC,
-                            hipStream_t stream) {
+                            const float* meanimg, long total, long perImg,
+                            int C, int nLogC, hipStream_t stream) {
   long t4 = total / 4;
   hipLaunchKernelGGL(u8_finish_k, dim3(ew_grid(t4)), dim3(256), 0, stream,
-                     src, (__bf16*)dst, t4, C, nLogC);
+                     src, (__bf16*)dst, meanimg, t4, perImg, C, nLogC);
   return hipGetLastError();
 }
 
