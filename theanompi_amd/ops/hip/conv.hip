@@ -678,17 +678,22 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
 
 // ---------------------------------------------------------------------------
 
+// block = CLW channel-group lanes x (256/CLW) row-lanes; LDS-reduce
+// the row lanes so each block issues at most 256 atomics (the naive
+// version serialized on ~1M atomics over a few hundred addresses).
+// CLW adapts to the layer width: the fixed 32-wide split idled 1/2 to
+// 3/4 of every block on narrow layers (C 64-128 — most of GoogLeNet),
+// dividing effective load bandwidth by the idle fraction.
+template <int CLW>
 __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
                              float* __restrict__ db, long rows, int C) {
-  // block = 32 channel-groups x 8 row-lanes; LDS-reduce the row lanes so
-  // each block issues at most 256 atomics (the naive version serialized
-  // on ~1M atomics over a few hundred addresses).
+  constexpr int RLN = 256 / CLW;
   const int c8n = C / 8;
-  const int cl = threadIdx.x & 31;          // channel-group lane
-  const int rl = threadIdx.x >> 5;          // row lane (0..7)
-  const int c8 = blockIdx.x * 32 + cl;
-  const long sl = (long)blockIdx.y * 8 + rl;
-  const long nsl = (long)gridDim.y * 8;
+  const int cl = threadIdx.x & (CLW - 1);   // channel-group lane
+  const int rl = threadIdx.x / CLW;         // row lane (0..RLN-1)
+  const int c8 = blockIdx.x * CLW + cl;
+  const long sl = (long)blockIdx.y * RLN + rl;
+  const long nsl = (long)gridDim.y * RLN;
   float s[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   if (c8 < c8n) {
     const __bf16* base = dy + c8 * 8;
@@ -712,7 +717,7 @@ __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
       }
     }
   }
-  __shared__ float red[8][32][8];
+  __shared__ float red[RLN][CLW][8];        // 16 KB for any CLW
 #pragma unroll
   for (int j = 0; j < 8; ++j) red[rl][cl][j] = s[j];
   __syncthreads();
@@ -721,11 +726,12 @@ __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
     for (int j = 0; j < 8; ++j) {
       float a = 0.f;
 #pragma unroll
-      for (int t = 0; t < 8; ++t) a += red[t][cl][j];
+      for (int t = 0; t < RLN; ++t) a += red[t][cl][j];
       atomicAdd(&db[c8 * 8 + j], a);
     }
   }
 }
+
 
 // ---------------------------------------------------------------------------
 // launchers
@@ -857,16 +863,26 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
 hipError_t launch_bias_wgrad(const void* dy, float* db, long rows, int C,
                              hipStream_t stream) {
   int c8n = C / 8;
-  int gx = cdiv(c8n, 32);
-  long gy = (rows + 31) / 32;              // 4 rows per row-lane pass
+  int clw = c8n >= 32 ? 32 : (c8n >= 16 ? 16 : 8);
+  int gx = cdiv(c8n, clw);
+  int rln = 256 / clw;
+  long gy = (rows + rln * 4 - 1) / (rln * 4);
   long target = 1024 / gx;                 // bounded atomics per channel
   if (target > 128) target = 128;
   if (gy > target) gy = target;
   if (gy < 1) gy = 1;
   dim3 grid(gx, (unsigned)gy);
-  hipLaunchKernelGGL(bias_wgrad_k, grid, dim3(256), 0, stream,
-                     (const __bf16*)dy, db, rows, C);
+  if (clw == 32)
+    hipLaunchKernelGGL(bias_wgrad_k<32>, grid, dim3(256), 0, stream,
+                       (const __bf16*)dy, db, rows, C);
+  else if (clw == 16)
+    hipLaunchKernelGGL(bias_wgrad_k<16>, grid, dim3(256), 0, stream,
+                       (const __bf16*)dy, db, rows, C);
+  else
+    hipLaunchKernelGGL(bias_wgrad_k<8>, grid, dim3(256), 0, stream,
+                       (const __bf16*)dy, db, rows, C);
   return hipGetLastError();
 }
+
 
 }  // extern "C"

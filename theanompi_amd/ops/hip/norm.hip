@@ -132,15 +132,17 @@ __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
 // 32 channel-groups x 8 row-lanes per block, 4-row ILP, LDS-reduced
 // (same shape as bias_wgrad_k; the naive per-channel-thread version was
 // the ResNet-50 bottleneck).
+template <int CLW>     // channel-lane width adapts to narrow layers
 __global__ void bn_sums_k(const __bf16* __restrict__ x,
                           float* __restrict__ s1, float* __restrict__ s2,
                           long rows, int C) {
+  constexpr int RLN = 256 / CLW;
   const int c8n = C / 8;
-  const int cl = threadIdx.x & 31;
-  const int rl = threadIdx.x >> 5;
-  const int c8 = blockIdx.x * 32 + cl;
-  const long sl = (long)blockIdx.y * 8 + rl;
-  const long nsl = (long)gridDim.y * 8;
+  const int cl = threadIdx.x & (CLW - 1);
+  const int rl = threadIdx.x / CLW;
+  const int c8 = blockIdx.x * CLW + cl;
+  const long sl = (long)blockIdx.y * RLN + rl;
+  const long nsl = (long)gridDim.y * RLN;
   float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float b[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   if (c8 < c8n) {
@@ -170,7 +172,7 @@ __global__ void bn_sums_k(const __bf16* __restrict__ x,
       }
     }
   }
-  __shared__ float red[2][8][32][8];
+  __shared__ float red[2][RLN][CLW][8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     red[0][rl][cl][j] = a[j];
@@ -182,7 +184,7 @@ __global__ void bn_sums_k(const __bf16* __restrict__ x,
     for (int j = 0; j < 8; ++j) {
       float sa = 0.f, sb = 0.f;
 #pragma unroll
-      for (int t = 0; t < 8; ++t) {
+      for (int t = 0; t < RLN; ++t) {
         sa += red[0][t][cl][j];
         sb += red[1][t][cl][j];
       }
@@ -269,7 +271,7 @@ __global__ void bn_gamma_prep_k(const float* __restrict__ gamma,
 // match the device-side allok flag exits immediately.  A single kernel
 // carrying both loop bodies spilled 68 B/lane of scratch and ran 31%
 // slower than the x path it was meant to beat.
-template <int USEY, int RELU>
+template <int USEY, int RELU, int CLW>
 __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
                               const __bf16* __restrict__ x,
                               const __bf16* __restrict__ y,
@@ -282,12 +284,13 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
                               float* __restrict__ sdyx, long rows,
                               int C) {
   if ((*allok != 0) != (USEY != 0)) return;   // other path owns this op
+  constexpr int RLN = 256 / CLW;
   const int c8n = C / 8;
-  const int cl = threadIdx.x & 31;
-  const int rl = threadIdx.x >> 5;
-  const int c8 = blockIdx.x * 32 + cl;
-  const long sl = (long)blockIdx.y * 8 + rl;
-  const long nsl = (long)gridDim.y * 8;
+  const int cl = threadIdx.x & (CLW - 1);
+  const int rl = threadIdx.x / CLW;
+  const int c8 = blockIdx.x * CLW + cl;
+  const long sl = (long)blockIdx.y * RLN + rl;
+  const long nsl = (long)gridDim.y * RLN;
   float a[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float b[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
   float m[8], is[8];
@@ -326,7 +329,7 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
     if (blockIdx.y == 0 && rl == 0)
       for (long r = rows & ~3L; r < rows; ++r) accrow(r);
   }
-  __shared__ float red[2][8][32][8];
+  __shared__ float red[2][RLN][CLW][8];
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     red[0][rl][cl][j] = a[j];
@@ -338,7 +341,7 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
     for (int j = 0; j < 8; ++j) {
       float sa = 0.f, sb = 0.f;
 #pragma unroll
-      for (int t = 0; t < 8; ++t) {
+      for (int t = 0; t < RLN; ++t) {
         sa += red[0][t][cl][j];
         sb += red[1][t][cl][j];
       }
@@ -466,13 +469,24 @@ hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
                                long rows, int C, float momentum,
                                float eps, int relu, hipStream_t stream) {
   {
-    int gx = cdiv(C / 8, 32);
-    long gy = (rows + 31) / 32;
+    int c8n = C / 8;
+    int clw = c8n >= 32 ? 32 : (c8n >= 16 ? 16 : 8);
+    int gx = cdiv(c8n, clw);
+    int rln = 256 / clw;
+    long gy = (rows + rln * 4 - 1) / (rln * 4);
     long cap = 1024 / gx; if (cap > 128) cap = 128;
     if (gy > cap) gy = cap;
     if (gy < 1) gy = 1;
-    hipLaunchKernelGGL(bn_sums_k, dim3(gx, (unsigned)gy), dim3(256), 0,
-                       stream, (const __bf16*)x, s1, s2, rows, C);
+    dim3 g(gx, (unsigned)gy);
+    if (clw == 32)
+      hipLaunchKernelGGL(bn_sums_k<32>, g, dim3(256), 0, stream,
+                         (const __bf16*)x, s1, s2, rows, C);
+    else if (clw == 16)
+      hipLaunchKernelGGL(bn_sums_k<16>, g, dim3(256), 0, stream,
+                         (const __bf16*)x, s1, s2, rows, C);
+    else
+      hipLaunchKernelGGL(bn_sums_k<8>, g, dim3(256), 0, stream,
+                         (const __bf16*)x, s1, s2, rows, C);
   }
   hipLaunchKernelGGL(bn_finalize_k, dim3(cdiv(C, 256)), dim3(256), 0,
                      stream, s1, s2, mean, invstd, rmean, rvar, rows, C,
@@ -505,8 +519,11 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
   hipLaunchKernelGGL(bn_gamma_prep_k, dim3(cdiv(C, 256)), dim3(256), 0,
                      stream, gamma, rgamma, allok, sdy, sdyx, C);
   {
-    int gx = cdiv(C / 8, 32);
-    long gy = (rows + 31) / 32;
+    int c8n = C / 8;
+    int clw = c8n >= 32 ? 32 : (c8n >= 16 ? 16 : 8);
+    int gx = cdiv(c8n, clw);
+    int rln = 256 / clw;
+    long gy = (rows + rln * 4 - 1) / (rln * 4);
     // measured: 512 blocks only pay off for the relu y-path layers
     // (smaller C, long rows); elsewhere the fixed LDS-reduce + atomic
     // cost of extra blocks outweighs the parallelism.
@@ -515,18 +532,27 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
     if (cap > lim) cap = lim;
     if (gy > cap) gy = cap;
     if (gy < 1) gy = 1;
+    dim3 g(gx, (unsigned)gy);
     auto launch_pair = [&](auto k1, auto k0) {
-      hipLaunchKernelGGL(k1, dim3(gx, (unsigned)gy), dim3(256), 0,
-                         stream, (const __bf16*)dy, (const __bf16*)x,
-                         (const __bf16*)y, mean, invstd, beta, rgamma,
-                         allok, sdy, sdyx, rows, C);
-      hipLaunchKernelGGL(k0, dim3(gx, (unsigned)gy), dim3(256), 0,
-                         stream, (const __bf16*)dy, (const __bf16*)x,
-                         (const __bf16*)y, mean, invstd, beta, rgamma,
-                         allok, sdy, sdyx, rows, C);
+      hipLaunchKernelGGL(k1, g, dim3(256), 0, stream, (const __bf16*)dy,
+                         (const __bf16*)x, (const __bf16*)y, mean,
+                         invstd, beta, rgamma, allok, sdy, sdyx, rows,
+                         C);
+      hipLaunchKernelGGL(k0, g, dim3(256), 0, stream, (const __bf16*)dy,
+                         (const __bf16*)x, (const __bf16*)y, mean,
+                         invstd, beta, rgamma, allok, sdy, sdyx, rows,
+                         C);
     };
-    if (relu) launch_pair(bn_bwd_sums_k<1, 1>, bn_bwd_sums_k<0, 1>);
-    else      launch_pair(bn_bwd_sums_k<1, 0>, bn_bwd_sums_k<0, 0>);
+    auto pick = [&](auto tag) {
+      constexpr int W = decltype(tag)::value;
+      if (relu) launch_pair(bn_bwd_sums_k<1, 1, W>,
+                            bn_bwd_sums_k<0, 1, W>);
+      else      launch_pair(bn_bwd_sums_k<1, 0, W>,
+                            bn_bwd_sums_k<0, 0, W>);
+    };
+    if (clw == 32)      pick(std::integral_constant<int, 32>{});
+    else if (clw == 16) pick(std::integral_constant<int, 16>{});
+    else                pick(std::integral_constant<int, 8>{});
   }
   hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C / 4)), dim3(256),
                      0, stream, (const __bf16*)dy, (const __bf16*)x,
