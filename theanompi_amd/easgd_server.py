@@ -59,7 +59,6 @@ def run():
             flag = 1 if uidx + tau < n_total else 0
             store.set(f"ea/resp/{w}/{seq[w]}", str(flag))
             exch.server_exchange(w)          # recv params / send center
-            model.refresh_all()
             uidx += tau
             seq[w] += 1
             if flag == 0:
@@ -70,6 +69,7 @@ def run():
         if uidx - last_val >= val_freq:
             last_val = uidx
             epoch += 1
+            model.refresh_all()      # bf16 working copies, for val only
             stats = torch.zeros(3)
             n = 0
             for j in range(model.data.n_batch_val):
