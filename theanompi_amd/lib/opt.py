@@ -34,11 +34,15 @@ class ParamRec:
 class SGDMomentum:
     """Momentum SGD over ParamRecs, usable under every rule."""
 
-    def __init__(self, recs, lr=0.01, momentum=0.9, weight_decay=0.0):
+    def __init__(self, recs, lr=0.01, momentum=0.9, weight_decay=0.0,
+                 grad_arena=None):
         self.recs = list(recs)
         self.lr = lr
         self.momentum = momentum
         self.weight_decay = weight_decay
+        # flat pre-zeroed buffer the conv/FC backward kernels atomically
+        # accumulate into (one fill per step instead of one per op)
+        self.grad_arena = grad_arena
         self._gpu = any(r.master.is_cuda for r in self.recs)
         self.persistent_grads = False   # graph mode: stable grad buffers
         self._desc = None               # cached kernel descriptor
@@ -64,6 +68,8 @@ class SGDMomentum:
             for r in self.recs:
                 r.master.grad.zero_()
             return
+        if self.grad_arena is not None:
+            self.grad_arena.zero_()      # ONE fill for every conv/FC grad
         for r in self.recs:
             if r.master.grad is not None:
                 r.master.grad = None

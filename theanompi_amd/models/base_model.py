@@ -112,6 +112,26 @@ class BaseModel:
             n = r.master.numel()
             r.vel = self.vel_arena[off:off + n].view(r.master.shape)
             off += n
+        # pre-zeroed grad arena: the GPU conv/FC backward accumulates
+        # atomically and would otherwise pay one fill launch per op
+        # (~14% of the GoogLeNet step); instead zero_grad clears this
+        # arena once and backward hands out slices.
+        self.grad_arena = None
+        if self.gpu:
+            self.grad_arena = torch.zeros(total, dtype=torch.float32,
+                                          device=self.device)
+            off = 0
+            views = {}
+            for r in self.recs:
+                n = r.master.numel()
+                views[id(r.master)] = \
+                    self.grad_arena[off:off + n].view(r.master.shape)
+                off += n
+            for mod in self.module.modules():
+                if isinstance(mod, (L.Conv2d, L.Linear)):
+                    mod._dw_buf = views.get(id(mod.weight))
+                    if mod.bias is not None:
+                        mod._db_buf = views.get(id(mod.bias))
         # reference-parity flat views for the exchangers / checkpoints
         self.params = [r.master for r in self.recs]
         self.vels = [r.vel for r in self.recs]
@@ -168,6 +188,7 @@ class BaseModel:
         optimizer and warm the kernels (instant)."""
         self.sync_type = sync_type
         self.opt = SGDMomentum(self.recs, lr=self.current_lr,
+                               grad_arena=self.grad_arena,
                                momentum=self.momentum,
                                weight_decay=self.weight_decay)
 

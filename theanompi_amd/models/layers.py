@@ -160,6 +160,10 @@ class ConvPlan:
                     "Kdf2": base, "e2f2": e2f2,
                     "table": torch.from_numpy(tab2).to(device),
                 })
+        # every input pixel's class has a contributing plan -> the dgrad
+        # launches write ALL of dx and the buffer needn't be pre-zeroed
+        self.dgrad_full_cover = all(p is not None
+                                    for p in self.dgrad_plans)
         # repack maps (dest flat [G,Kdf2,Cgp] <- src flat [G,Kdf,Kgp])
         self._repack_maps = None
 
@@ -297,14 +301,22 @@ class _HipConvFn(torch.autograd.Function):
             dym = torch.empty_like(dy)
             C.relu_bwd(dy, y, dym)
             dy = dym
-        # bias grad
+        # bias grad.  The arena slices (_dw_buf/_db_buf) are pre-zeroed
+        # by zero_grad; they are only safe to hand to autograd while
+        # .grad is None (otherwise AccumulateGrad would add the slice
+        # into itself), so manual grad accumulation falls back to a
+        # fresh zeroed buffer.
         db = None
         if layer.b_f32 is not None:
-            db = torch.zeros_like(layer.b_f32)
+            db = getattr(layer, "_db_buf", None)
+            if db is None or layer.bias.grad is not None:
+                db = torch.zeros_like(layer.b_f32)
             C.bias_wgrad(dy, db)
         # weight grad (fp32, flat layout)
-        dw = torch.zeros(plan.flat_shape(), dtype=torch.float32,
-                         device=dy.device)
+        dw = getattr(layer, "_dw_buf", None)
+        if dw is None or layer.weight.grad is not None:
+            dw = torch.zeros(plan.flat_shape(), dtype=torch.float32,
+                             device=dy.device)
         C.conv_wgrad(xp, dy, plan.table, dw, plan.kdf_mask,
                      plan.Wp * plan.Cphys, plan.Cphys,
                      plan.stride[0], plan.stride[1],
@@ -427,8 +439,9 @@ class Conv2d(_Base):
         plan = self.plan
         N = dy.shape[0]
         H, W = plan.NHWC[1], plan.NHWC[2]
-        dx = torch.zeros((N, H, W, plan.Cphys), dtype=torch.bfloat16,
-                         device=dy.device)
+        alloc = torch.empty if plan.dgrad_full_cover else torch.zeros
+        dx = alloc((N, H, W, plan.Cphys), dtype=torch.bfloat16,
+                   device=dy.device)
         for plan2, wdg in zip(plan.dgrad_plans, self._wdg):
             if plan2 is None:
                 continue

@@ -135,7 +135,8 @@ class _MaxPool2dFn(torch.autograd.Function):
         (argmax,) = ctx.saved_tensors
         xshape, kernel, stride, pad = ctx.meta
         C = _ops.ext()
-        dx = torch.zeros(xshape, dtype=dy.dtype, device=dy.device)
+        # maxpool_bwd_k writes every dx element - no pre-zero needed
+        dx = torch.empty(xshape, dtype=dy.dtype, device=dy.device)
         C.maxpool_bwd(dy.contiguous(), argmax, dx,
                       kernel[0], kernel[1], stride[0], stride[1],
                       pad[0], pad[1])
@@ -162,7 +163,7 @@ class _AvgPool2dFn(torch.autograd.Function):
     def backward(ctx, dy):
         xshape, kernel, stride, pad = ctx.meta
         C = _ops.ext()
-        dx = torch.zeros(xshape, dtype=dy.dtype, device=dy.device)
+        dx = torch.empty(xshape, dtype=dy.dtype, device=dy.device)
         C.avgpool_bwd(dy.contiguous(), dx, kernel[0], kernel[1],
                       stride[0], stride[1], pad[0], pad[1])
         return dx, None, None, None

@@ -496,13 +496,14 @@ __global__ __launch_bounds__(512) void conv_gather_gemm_8p(
 }
 
 // split-K finalize: y = bf16(relu(Y32 + bias))
-__global__ void splitk_finalize_k(const float* __restrict__ y32,
+__global__ void splitk_finalize_k(float* __restrict__ y32,
                                   const float* __restrict__ bias,
                                   __bf16* __restrict__ y, long total,
                                   int C, int relu) {
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     float v = y32[i];
+    y32[i] = 0.f;     // leave the cached workspace zeroed for reuse
     if (bias) v += bias[i % C];
     if (relu && v < 0.f) v = 0.f;
     y[i] = (__bf16)v;
@@ -739,7 +740,7 @@ __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
 
 extern "C" {
 
-hipError_t launch_splitk_finalize(const float* y32, const float* bias,
+hipError_t launch_splitk_finalize(float* y32, const float* bias,
                                   void* y, long total, int C, int relu,
                                   hipStream_t stream) {
   long g = (total + 255) / 256;
