@@ -174,19 +174,22 @@ class _LRNFn(torch.autograd.Function):
     def forward(ctx, x, size, alpha, beta, k):
         C = _ops.ext()
         y = torch.empty_like(x)
-        scale = torch.empty(x.shape, dtype=torch.float32, device=x.device)
+        # 1/denom in bf16: halves the save traffic, and backward
+        # reconstructs x from y so only (y, scale) need saving at all
+        scale = torch.empty(x.shape, dtype=torch.bfloat16,
+                            device=x.device)
         C.lrn_fwd(x, y, scale, size, alpha, beta, k)   # scale = 1/denom
-        ctx.save_for_backward(x, y, scale)
+        ctx.save_for_backward(y, scale)
         ctx.meta = (size, alpha, beta, k)
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, scale = ctx.saved_tensors
+        y, scale = ctx.saved_tensors
         size, alpha, beta, k = ctx.meta
         C = _ops.ext()
-        dx = torch.empty_like(x)
-        C.lrn_bwd(dy.contiguous(), x, y, scale, dx, size, alpha, beta, k)
+        dx = torch.empty_like(y)
+        C.lrn_bwd(dy.contiguous(), y, scale, dx, size, alpha, beta, k)
         return dx, None, None, None, None
 
 

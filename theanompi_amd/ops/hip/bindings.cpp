@@ -84,11 +84,11 @@ hipError_t launch_avgpool_fwd(const void*, void*, int, int, int, int, int,
 hipError_t launch_avgpool_bwd(const void*, void*, int, int, int, int, int,
                               int, int, int, int, int, int, int,
                               hipStream_t);
-hipError_t launch_lrn_fwd(const void*, void*, float*, long, int, int, float,
+hipError_t launch_lrn_fwd(const void*, void*, void*, long, int, int, float,
                           float, float, hipStream_t);
-hipError_t launch_lrn_bwd(const void*, const void*, const void*,
-                          const float*, void*, long, int, int, float, float,
-                          float, hipStream_t);
+hipError_t launch_lrn_bwd(const void*, const void*, const void*, void*,
+                          long, int, int, float, float, float,
+                          hipStream_t);
 hipError_t launch_bn_fwd_train(const void*, void*, float*, float*, float*,
                                float*, float*, float*, const float*,
                                const float*, long, int, float, float, int,
@@ -571,25 +571,25 @@ static void avgpool_bwd(torch::Tensor dy, torch::Tensor dx, long kh,
 static void lrn_fwd(torch::Tensor x, torch::Tensor y, torch::Tensor scale,
                     long size, double alpha, double beta, double k) {
   check_bf16(x, "x");
-  check_f32(scale, "scale");
+  check_bf16(scale, "scale");          // 1/denom saved in bf16
   long C = x.size(-1);
   long rows = x.numel() / C;
-  CHECK_HIP(launch_lrn_fwd(x.data_ptr(), y.data_ptr(),
-                           scale.data_ptr<float>(), rows, (int)C,
-                           (int)size, (float)alpha, (float)beta, (float)k,
-                           cur_stream()));
+  CHECK_HIP(launch_lrn_fwd(x.data_ptr(), y.data_ptr(), scale.data_ptr(),
+                           rows, (int)C, (int)size, (float)alpha,
+                           (float)beta, (float)k, cur_stream()));
 }
 
-static void lrn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor y,
+static void lrn_bwd(torch::Tensor dy, torch::Tensor y,
                     torch::Tensor scale, torch::Tensor dx, long size,
                     double alpha, double beta, double k) {
   check_bf16(dy, "dy");
-  long C = x.size(-1);
-  long rows = x.numel() / C;
-  CHECK_HIP(launch_lrn_bwd(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
-                           scale.data_ptr<float>(), dx.data_ptr(), rows,
-                           (int)C, (int)size, (float)alpha, (float)beta,
-                           (float)k, cur_stream()));
+  check_bf16(scale, "scale");
+  long C = y.size(-1);
+  long rows = y.numel() / C;
+  CHECK_HIP(launch_lrn_bwd(dy.data_ptr(), y.data_ptr(), scale.data_ptr(),
+                           dx.data_ptr(), rows, (int)C, (int)size,
+                           (float)alpha, (float)beta, (float)k,
+                           cur_stream()));
 }
 
 static void bn_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,

@@ -10,7 +10,7 @@
 
 __global__ void lrn_fwd_k(const __bf16* __restrict__ x,
                           __bf16* __restrict__ y,
-                          float* __restrict__ scale, long rows, int C,
+                          __bf16* __restrict__ scale, long rows, int C,
                           int size, float alpha, float beta, float k) {
   // one thread per (row, 4-channel group); the +-4 window around the
   // group is covered by three aligned 8-B loads (size <= 9).
@@ -50,7 +50,7 @@ __global__ void lrn_fwd_k(const __bf16* __restrict__ x,
       }
       float sc = k + a * s;
       float inv = 1.f / sc;                // saved: bwd needs 1/scale
-      scale[o + j] = inv;
+      scale[o + j] = (__bf16)inv;          // bf16 save halves the write
       // inv^beta; beta = 0.75 fast path: sqrt(inv)*sqrt(sqrt(inv))
       float pw = (beta == 0.75f)
                      ? __fsqrt_rn(inv) * __fsqrt_rn(__fsqrt_rn(inv))
@@ -62,10 +62,11 @@ __global__ void lrn_fwd_k(const __bf16* __restrict__ x,
 }
 
 // dx_c = dy_c*scale_c^-b - (2*a*b/n) * x_c * sum_{j in win(c)} dy_j*y_j/scale_j
+// x is reconstructed from the saved output (x = y / scale^beta), so
+// the backward reads dy + y + bf16 scale only — no x pass.
 __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
-                          const __bf16* __restrict__ x,
                           const __bf16* __restrict__ y,
-                          const float* __restrict__ scale,
+                          const __bf16* __restrict__ scale,
                           __bf16* __restrict__ dx, long rows, int C,
                           int size, float alpha, float beta, float k) {
   const int half = size / 2;
@@ -78,30 +79,31 @@ __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
     long row = i4 / C4;
     long b0 = row * C;
     // t[w] = dy*y/scale over the 12-channel neighbourhood
-    float t[12], xv[12], dyv[4];
+    float t[12], yv[4], dyv[4];
 #pragma unroll
     for (int b = 0; b < 3; ++b) {
       int cb = c4 + b - 1;
       if (cb >= 0 && cb < C4) {
         bf16x4 d = *(const bf16x4*)(dy + b0 + cb * 4);
         bf16x4 yy = *(const bf16x4*)(y + b0 + cb * 4);
-        bf16x4 xx = *(const bf16x4*)(x + b0 + cb * 4);
-        f32x4 sc = *(const f32x4*)(scale + b0 + cb * 4);  // = 1/scale
+        bf16x4 sc = *(const bf16x4*)(scale + b0 + cb * 4); // = 1/scale
 #pragma unroll
         for (int j = 0; j < 4; ++j) {
-          t[b * 4 + j] = (float)d[j] * (float)yy[j] * sc[j];
-          xv[b * 4 + j] = (float)xx[j];
-          if (b == 1) dyv[j] = (float)d[j];
+          t[b * 4 + j] = (float)d[j] * (float)yy[j] * (float)sc[j];
+          if (b == 1) {
+            dyv[j] = (float)d[j];
+            yv[j] = (float)yy[j];
+          }
         }
       } else {
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          t[b * 4 + j] = 0.f;
-          xv[b * 4 + j] = 0.f;
-        }
+        for (int j = 0; j < 4; ++j) t[b * 4 + j] = 0.f;
       }
     }
-    const f32x4 scc = *(const f32x4*)(scale + b0 + c4 * 4);
+    bf16x4 sccb = *(const bf16x4*)(scale + b0 + c4 * 4);
+    f32x4 scc;
+#pragma unroll
+    for (int j = 0; j < 4; ++j) scc[j] = (float)sccb[j];
     long o = b0 + c4 * 4;
     bf16x4 out;
 #pragma unroll
@@ -115,7 +117,8 @@ __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
       float pw = (beta == 0.75f)
                      ? __fsqrt_rn(inv) * __fsqrt_rn(__fsqrt_rn(inv))
                      : __powf(inv, beta);
-      float d = dyv[j] * pw - coef * xv[j] * s;
+      float xv = yv[j] * __frcp_rn(pw);    // x = y / scale^beta
+      float d = dyv[j] * pw - coef * xv * s;
       out[j] = (__bf16)d;
     }
     *(bf16x4*)(dx + o) = out;
@@ -442,22 +445,22 @@ __global__ void bn_eval_stats_k(const float* __restrict__ rmean,
 
 extern "C" {
 
-hipError_t launch_lrn_fwd(const void* x, void* y, float* scale, long rows,
+hipError_t launch_lrn_fwd(const void* x, void* y, void* scale, long rows,
                           int C, int size, float alpha, float beta,
                           float k, hipStream_t stream) {
   hipLaunchKernelGGL(lrn_fwd_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
-                     stream, (const __bf16*)x, (__bf16*)y, scale, rows, C,
-                     size, alpha, beta, k);
+                     stream, (const __bf16*)x, (__bf16*)y,
+                     (__bf16*)scale, rows, C, size, alpha, beta, k);
   return hipGetLastError();
 }
 
-hipError_t launch_lrn_bwd(const void* dy, const void* x, const void* y,
-                          const float* scale, void* dx, long rows, int C,
+hipError_t launch_lrn_bwd(const void* dy, const void* y,
+                          const void* scale, void* dx, long rows, int C,
                           int size, float alpha, float beta, float k,
                           hipStream_t stream) {
   hipLaunchKernelGGL(lrn_bwd_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
-                     stream, (const __bf16*)dy, (const __bf16*)x,
-                     (const __bf16*)y, scale, (__bf16*)dx, rows, C, size,
+                     stream, (const __bf16*)dy, (const __bf16*)y,
+                     (const __bf16*)scale, (__bf16*)dx, rows, C, size,
                      alpha, beta, k);
   return hipGetLastError();
 }
