@@ -405,7 +405,11 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
   a.Kphys = (int)Kphys;
   a.xTotal = xp.numel();
   long base_blocks = ((Kgp + 63) / 64) * ((Kdf + 63) / 64) * G;
-  long splitK = 1024 / (base_blocks ? base_blocks : 1);
+  static const long wgrad_target = [] {
+    const char* e = getenv("THEANOMPI_WGRAD_T");
+    return e ? atol(e) : 1024L;          // total-block target for splitK
+  }();
+  long splitK = wgrad_target / (base_blocks ? base_blocks : 1);
   long maxSplit = (a.M + 63) / 64;
   if (splitK < 1) splitK = 1;
   if (splitK > maxSplit) splitK = maxSplit;
