@@ -405,10 +405,14 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
   a.Kphys = (int)Kphys;
   a.xTotal = xp.numel();
   long base_blocks = ((Kgp + 63) / 64) * ((Kdf + 63) / 64) * G;
-  static const long wgrad_target = [] {
+  static const long wgrad_env = [] {
     const char* e = getenv("THEANOMPI_WGRAD_T");
-    return e ? atol(e) : 1024L;          // total-block target for splitK
+    return e ? atol(e) : 0L;             // total-block target override
   }();
+  // grouped convs shrink the per-group GEMM (few base blocks) and
+  // measured ~1.5% faster end-to-end with a 2048-block target;
+  // ungrouped layers measured best at 1024 (swept on hardware).
+  long wgrad_target = wgrad_env ? wgrad_env : (G > 1 ? 2048 : 1024);
   long splitK = wgrad_target / (base_blocks ? base_blocks : 1);
   long maxSplit = (a.M + 63) / 64;
   if (splitK < 1) splitK = 1;
