@@ -441,3 +441,21 @@ def test_batchnorm_bwd_xhat_from_y(relu, tiny):
     assert rel_err(xg.grad, xr.grad) < 5e-2
     assert rel_err(gg.grad.cpu(), gr.grad) < 5e-2
     assert rel_err(bg.grad.cpu(), br.grad) < 5e-2
+
+
+@pytest.mark.gpu
+def test_relu_bwd_sliced_dy():
+    """relu_bwd consumes a channel slice of a concat buffer directly
+    (the inception backward split) — no .contiguous() copy."""
+    from theanompi_amd import ops as _ops
+    C = _ops.ext()
+    d = dev()
+    torch.manual_seed(11)
+    full = q(torch.randn(4, 6, 6, 96)).to(d).to(torch.bfloat16)
+    y = q(torch.randn(4, 6, 6, 32)).to(d).to(torch.bfloat16)
+    sl = full[..., 32:64]                   # non-contiguous channel slice
+    assert not sl.is_contiguous()
+    out = torch.empty(sl.shape, dtype=torch.bfloat16, device=d)
+    C.relu_bwd(sl, y, out)
+    ref = torch.where(y.float() > 0, sl.float(), torch.zeros(()).to(d))
+    assert rel_err(out, ref) == 0.0 or rel_err(out, ref) < 1e-6

@@ -265,6 +265,16 @@ class _EmptyF32Cache:
 _EMPTY_F32 = _EmptyF32Cache()
 
 
+def _row_strided(t):
+    """Last-dim-contiguous view whose rows sit a uniform stride apart
+    (e.g. a channel slice of a concat buffer)."""
+    if t.dim() != 4 or t.stride(3) != 1:
+        return False
+    ld = t.stride(2)
+    return (ld >= t.shape[3] and t.stride(1) == ld * t.shape[2]
+            and t.stride(0) == t.stride(1) * t.shape[1])
+
+
 class _HipConvFn(torch.autograd.Function):
 
     @staticmethod
@@ -296,11 +306,18 @@ class _HipConvFn(torch.autograd.Function):
         layer = ctx.layer
         plan = layer.plan
         xp, y = ctx.saved_tensors
-        dy = dy.contiguous()
         if layer.relu:
-            dym = torch.empty_like(dy)
+            # the relu mask kernel accepts a channel slice of a concat
+            # buffer directly (uniform row stride), so the inception
+            # backward's split needs no .contiguous() copy
+            if not (dy.is_contiguous() or _row_strided(dy)):
+                dy = dy.contiguous()
+            dym = torch.empty(dy.shape, dtype=dy.dtype,
+                              device=dy.device)
             C.relu_bwd(dy, y, dym)
             dy = dym
+        else:
+            dy = dy.contiguous()
         # bias grad.  The arena slices (_dw_buf/_db_buf) are pre-zeroed
         # by zero_grad; they are only safe to hand to autograd while
         # .grad is None (otherwise AccumulateGrad would add the slice

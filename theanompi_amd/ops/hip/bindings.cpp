@@ -439,10 +439,30 @@ static void relu_fwd(torch::Tensor x, torch::Tensor y) {
                             cur_stream()));
 }
 
+hipError_t launch_relu_bwd_s(const void*, const void*, void*, long, int,
+                             long, hipStream_t);
+
+// Accepts a contiguous dy OR a channel slice of a concat buffer
+// (last-dim contiguous, uniform row stride) — the inception backward
+// hands slices straight in, no .contiguous() copy.
 static void relu_bwd(torch::Tensor dy, torch::Tensor y, torch::Tensor dx) {
-  check_bf16(dy, "dy");
-  CHECK_HIP(launch_relu_bwd(dy.data_ptr(), y.data_ptr(), dx.data_ptr(),
-                            dy.numel(), cur_stream()));
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16);
+  if (dy.is_contiguous()) {
+    CHECK_HIP(launch_relu_bwd(dy.data_ptr(), y.data_ptr(), dx.data_ptr(),
+                              dx.numel(), cur_stream()));
+    return;
+  }
+  long C = dy.size(-1);
+  long ld = dy.stride(-2);
+  bool rowform = dy.stride(-1) == 1 && ld >= C;
+  long rows = dx.numel() / C, expect = ld;
+  for (int d = (int)dy.dim() - 3; d >= 0 && rowform; --d) {
+    expect *= dy.size(d + 1);
+    rowform = dy.stride(d) == expect;
+  }
+  TORCH_CHECK(rowform, "relu_bwd: unsupported dy layout");
+  CHECK_HIP(launch_relu_bwd_s(dy.data_ptr(), y.data_ptr(), dx.data_ptr(),
+                              rows * C, (int)C, ld, cur_stream()));
 }
 
 static void add_relu(torch::Tensor a, torch::Tensor b, torch::Tensor y,

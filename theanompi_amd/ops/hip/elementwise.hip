@@ -64,6 +64,27 @@ __global__ void relu_bwd_k(const __bf16* __restrict__ dy,
   }
 }
 
+// Strided-source variant: dy rows of length C sit ldSrc elements
+// apart — a channel slice of a concat buffer, as produced by the
+// inception backward's split.  The masked copy this kernel performs
+// anyway subsumes the .contiguous() the slice would otherwise need.
+__global__ void relu_bwd_s_k(const __bf16* __restrict__ dy,
+                             const __bf16* __restrict__ y,
+                             __bf16* __restrict__ dx, long n4, int C4,
+                             long ldSrc) {
+  const __bf16 z = (__bf16)0.f;
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += (long)gridDim.x * blockDim.x) {
+    long row = i / C4;
+    long c = (i - row * C4) * 4;
+    bf16x4 g = *(const bf16x4*)(dy + row * ldSrc + c);
+    bf16x4 v = *(const bf16x4*)(y + i * 4);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) g[j] = v[j] > z ? g[j] : z;
+    *(bf16x4*)(dx + i * 4) = g;
+  }
+}
+
 __global__ void add_relu_k(const __bf16* __restrict__ p,
                            const __bf16* __restrict__ q,
                            __bf16* __restrict__ y, long n4, int relu) {
@@ -275,6 +296,15 @@ hipError_t launch_relu_bwd(const void* dy, const void* y, void* dx, long n,
   hipLaunchKernelGGL(relu_bwd_k, dim3(ew_grid(n / 4)), dim3(256), 0,
                      stream, (const __bf16*)dy, (const __bf16*)y,
                      (__bf16*)dx, n / 4);
+  return hipGetLastError();
+}
+
+hipError_t launch_relu_bwd_s(const void* dy, const void* y, void* dx,
+                             long n, int C, long ldSrc,
+                             hipStream_t stream) {
+  hipLaunchKernelGGL(relu_bwd_s_k, dim3(ew_grid(n / 4)), dim3(256), 0,
+                     stream, (const __bf16*)dy, (const __bf16*)y,
+                     (__bf16*)dx, n / 4, C / 4, ldSrc);
   return hipGetLastError();
 }
 
