@@ -55,6 +55,8 @@ hipError_t launch_bias_wgrad(const void*, float*, long, int, hipStream_t);
 hipError_t launch_halo_pad(const void*, void*, int, int, int, int, int, int,
                            int, int, float, hipStream_t);
 hipError_t launch_relu_fwd(const void*, void*, long, hipStream_t);
+hipError_t launch_relu_bwd_s(const void*, const void*, void*, long, int,
+                             long, long, hipStream_t);
 hipError_t launch_relu_bwd(const void*, const void*, void*, long,
                            hipStream_t);
 hipError_t launch_add_relu(const void*, const void*, void*, long, int,
@@ -254,6 +256,8 @@ static torch::Tensor halo_pad_sym(torch::Tensor x, long ph, long pw,
   return halo_pad(x, ph, ph, pw, pw, fill);
 }
 
+static long row_ld(const torch::Tensor& t);
+
 // Split-K fp32 workspace, cached per device and re-zeroed by the
 // finalize kernel as it drains it, so the steady state needs no fill
 // launch at all (fills were ~14% of the GoogLeNet step).
@@ -271,7 +275,11 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
                      bool relu) {
   check_bf16(xp, "xp");
   check_bf16(w, "w");
-  check_bf16(y, "y");
+  // y may be a channel slice of a concat buffer (concat-writing conv:
+  // last dim contiguous, uniform row stride) — strided epilogue writes.
+  TORCH_CHECK(y.scalar_type() == torch::kBFloat16 && y.is_cuda());
+  const bool yContig = y.is_contiguous();
+  TORCH_CHECK(yContig || row_ld(y) > 0, "conv_fwd: unsupported y layout");
   if (bias.numel()) check_f32(bias, "bias");
   TORCH_CHECK(table.scalar_type() == torch::kInt32 && table.is_cuda());
   TORCH_CHECK(w.dim() == 3 && xp.dim() == 4 && y.dim() == 4);
@@ -291,9 +299,9 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
   a.M = (int)(N * P * Q);
   a.P = (int)P;
   a.Q = (int)Q;
-  a.sN = P * Q * Kphys;
-  a.sP = Q * Kphys;
-  a.sQ = Kphys;
+  a.sN = yContig ? P * Q * Kphys : y.stride(0);
+  a.sP = yContig ? Q * Kphys : y.stride(1);
+  a.sQ = yContig ? Kphys : y.stride(2);
   a.outOff = 0;
   a.inNS = Hp * Wp * Cphys;
   a.inPS = sh * rowStride;
@@ -305,7 +313,9 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
   a.xTotal = xp.numel();
   long BN = Kgp > 64 ? 128 : 64;
   long baseBlocks = ((a.M + 127) / 128) * ((Kgp + BN - 1) / BN) * G;
-  a.splitK = pick_splitk(baseBlocks, Kdf / 64);
+  // split-K indexes its fp32 workspace with the output strides, so it
+  // requires a contiguous y
+  a.splitK = yContig ? pick_splitk(baseBlocks, Kdf / 64) : 1;
   a.Y32 = nullptr;
   if (a.splitK > 1)
     a.Y32 = splitk_ws(y.numel(), y.options());
@@ -439,30 +449,36 @@ static void relu_fwd(torch::Tensor x, torch::Tensor y) {
                             cur_stream()));
 }
 
-hipError_t launch_relu_bwd_s(const void*, const void*, void*, long, int,
-                             long, hipStream_t);
+// Row stride of a 4D view whose last dim is contiguous and whose rows
+// sit a uniform stride apart (contiguous tensors return C; anything
+// else returns -1).
+static long row_ld(const torch::Tensor& t) {
+  if (t.is_contiguous()) return t.size(-1);
+  if (t.dim() != 4 || t.stride(3) != 1) return -1;
+  long ld = t.stride(2);
+  if (ld < t.size(3)) return -1;
+  if (t.stride(1) != ld * t.size(2)) return -1;
+  if (t.stride(0) != t.stride(1) * t.size(1)) return -1;
+  return ld;
+}
 
-// Accepts a contiguous dy OR a channel slice of a concat buffer
+// Accepts contiguous dy/y OR channel slices of a concat buffer
 // (last-dim contiguous, uniform row stride) — the inception backward
 // hands slices straight in, no .contiguous() copy.
 static void relu_bwd(torch::Tensor dy, torch::Tensor y, torch::Tensor dx) {
   TORCH_CHECK(dy.scalar_type() == torch::kBFloat16);
-  if (dy.is_contiguous()) {
+  if (dy.is_contiguous() && y.is_contiguous()) {
     CHECK_HIP(launch_relu_bwd(dy.data_ptr(), y.data_ptr(), dx.data_ptr(),
                               dx.numel(), cur_stream()));
     return;
   }
   long C = dy.size(-1);
-  long ld = dy.stride(-2);
-  bool rowform = dy.stride(-1) == 1 && ld >= C;
-  long rows = dx.numel() / C, expect = ld;
-  for (int d = (int)dy.dim() - 3; d >= 0 && rowform; --d) {
-    expect *= dy.size(d + 1);
-    rowform = dy.stride(d) == expect;
-  }
-  TORCH_CHECK(rowform, "relu_bwd: unsupported dy layout");
+  long ldDy = row_ld(dy), ldY = row_ld(y);
+  TORCH_CHECK(ldDy > 0 && ldY > 0, "relu_bwd: unsupported layout");
+  long rows = dx.numel() / C;
   CHECK_HIP(launch_relu_bwd_s(dy.data_ptr(), y.data_ptr(), dx.data_ptr(),
-                              rows * C, (int)C, ld, cur_stream()));
+                              rows * C, (int)C, ldDy, ldY,
+                              cur_stream()));
 }
 
 static void add_relu(torch::Tensor a, torch::Tensor b, torch::Tensor y,

@@ -64,21 +64,22 @@ __global__ void relu_bwd_k(const __bf16* __restrict__ dy,
   }
 }
 
-// Strided-source variant: dy rows of length C sit ldSrc elements
-// apart — a channel slice of a concat buffer, as produced by the
-// inception backward's split.  The masked copy this kernel performs
-// anyway subsumes the .contiguous() the slice would otherwise need.
+// Strided-source variant: dy and/or y rows of length C sit ldDy/ldY
+// elements apart — channel slices of a concat buffer (the inception
+// split in backward; the concat-written y in forward).  The masked
+// copy this kernel performs anyway subsumes the .contiguous() the
+// slices would otherwise need.
 __global__ void relu_bwd_s_k(const __bf16* __restrict__ dy,
                              const __bf16* __restrict__ y,
                              __bf16* __restrict__ dx, long n4, int C4,
-                             long ldSrc) {
+                             long ldDy, long ldY) {
   const __bf16 z = (__bf16)0.f;
   for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
        i += (long)gridDim.x * blockDim.x) {
     long row = i / C4;
     long c = (i - row * C4) * 4;
-    bf16x4 g = *(const bf16x4*)(dy + row * ldSrc + c);
-    bf16x4 v = *(const bf16x4*)(y + i * 4);
+    bf16x4 g = *(const bf16x4*)(dy + row * ldDy + c);
+    bf16x4 v = *(const bf16x4*)(y + row * ldY + c);
 #pragma unroll
     for (int j = 0; j < 4; ++j) g[j] = v[j] > z ? g[j] : z;
     *(bf16x4*)(dx + i * 4) = g;
@@ -300,11 +301,11 @@ hipError_t launch_relu_bwd(const void* dy, const void* y, void* dx, long n,
 }
 
 hipError_t launch_relu_bwd_s(const void* dy, const void* y, void* dx,
-                             long n, int C, long ldSrc,
+                             long n, int C, long ldDy, long ldY,
                              hipStream_t stream) {
   hipLaunchKernelGGL(relu_bwd_s_k, dim3(ew_grid(n / 4)), dim3(256), 0,
                      stream, (const __bf16*)dy, (const __bf16*)y,
-                     (__bf16*)dx, n / 4, C / 4, ldSrc);
+                     (__bf16*)dx, n / 4, C / 4, ldDy, ldY);
   return hipGetLastError();
 }
 
