@@ -459,3 +459,37 @@ def test_relu_bwd_sliced_dy():
     C.relu_bwd(sl, y, out)
     ref = torch.where(y.float() > 0, sl.float(), torch.zeros(()).to(d))
     assert rel_err(out, ref) == 0.0 or rel_err(out, ref) < 1e-6
+
+
+@pytest.mark.gpu
+def test_inception_concat_write_equivalence():
+    """Concat-writing inception (branches write channel slices of the
+    concat buffer in place, zero-copy split in backward) must match the
+    torch.cat composition exactly — forward, input grad, param grads."""
+    import torch as T
+    from theanompi_amd.models.googlenet import _Inception
+    T.manual_seed(5)
+    d = dev()
+    m = _Inception(8, 32, 16, 16, 32, 16, 16, 16, str(d), 4)
+    x = q(T.randn(4, 8, 8, 32)).to(d).to(T.bfloat16)
+
+    def run(fn):
+        xi = x.clone().requires_grad_()
+        y = fn(xi)
+        loss = (y.float() ** 2).mean()
+        loss.backward()
+        grads = [p.grad.detach().clone() for p in m.parameters()]
+        for p in m.parameters():
+            p.grad = None
+        return y.detach().float(), xi.grad.detach().float(), grads
+
+    def old_path(xi):
+        return T.cat([m.b1(xi), m.b3b(m.b3a(xi)), m.b5b(m.b5a(xi)),
+                      m.bp(m.pool(xi))], dim=-1)
+
+    y_new, dx_new, g_new = run(m)               # concat-write path
+    y_old, dx_old, g_old = run(old_path)
+    assert T.equal(y_new, y_old)
+    assert rel_err(dx_new, dx_old.cpu()) < 1e-6
+    for a, b in zip(g_new, g_old):
+        assert rel_err(a, b.cpu()) < 2e-2       # wgrad atomics reorder

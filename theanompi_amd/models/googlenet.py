@@ -36,6 +36,21 @@ class _Inception(nn.Module):
                       + self.b5b.out_shape[2] + self.bp.out_shape[2])
 
     def forward(self, x):
+        if self.b1.gpu and x.is_cuda:
+            # concat-writing branches: every branch's final conv writes
+            # its channel slice of the concat buffer in place
+            N = x.shape[0]
+            P = self.b1.out_shape[0]
+            buf = torch.empty((N, P, P, self.out_c),
+                              dtype=torch.bfloat16, device=x.device)
+            ws = [self.b1.out_shape[2], self.b3b.out_shape[2],
+                  self.b5b.out_shape[2], self.bp.out_shape[2]]
+            o = [0, ws[0], ws[0] + ws[1], ws[0] + ws[1] + ws[2]]
+            y1 = self.b1(x, out=buf[..., o[0]:o[1]])
+            y3 = self.b3b(self.b3a(x), out=buf[..., o[1]:o[2]])
+            y5 = self.b5b(self.b5a(x), out=buf[..., o[2]:o[3]])
+            yp = self.bp(self.pool(x), out=buf[..., o[3]:])
+            return L.join_views(buf, y1, y3, y5, yp)
         y1 = self.b1(x)
         y3 = self.b3b(self.b3a(x))
         y5 = self.b5b(self.b5a(x))

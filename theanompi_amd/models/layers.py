@@ -265,6 +265,32 @@ class _EmptyF32Cache:
 _EMPTY_F32 = _EmptyF32Cache()
 
 
+class _JoinViews(torch.autograd.Function):
+    """Declare that ``buf`` is composed of the given channel slices
+    (already written in place by concat-writing convs).  Forward is a
+    no-op; backward splits dy into channel-slice VIEWS, which the
+    branch convs' relu backward consumes directly — the whole
+    concat/split round-trip costs zero copies."""
+
+    @staticmethod
+    def forward(ctx, buf, *slices):
+        ctx.widths = [sl.shape[-1] for sl in slices]
+        return buf
+
+    @staticmethod
+    def backward(ctx, dy):
+        outs = []
+        c0 = 0
+        for w in ctx.widths:
+            outs.append(dy[..., c0:c0 + w])
+            c0 += w
+        return (None, *outs)
+
+
+def join_views(buf, *slices):
+    return _JoinViews.apply(buf, *slices)
+
+
 def _row_strided(t):
     """Last-dim-contiguous view whose rows sit a uniform stride apart
     (e.g. a channel slice of a concat buffer)."""
@@ -285,8 +311,20 @@ class _HipConvFn(torch.autograd.Function):
         xp = C.halo_pad(x, ph, pw, 0.0) if (ph or pw) else x.contiguous()
         N = x.shape[0]
         M = N * plan.P * plan.Q
-        y = torch.empty((N, plan.P, plan.Q, plan.Kphys),
-                        dtype=torch.bfloat16, device=x.device)
+        out = getattr(layer, "_out_slice", None)
+        if out is not None:
+            # concat-writing conv: the epilogue writes a channel slice
+            # of the caller's concat buffer directly (uniform row
+            # stride) — torch.cat disappears.  Restricted to fused-relu
+            # convs: the backward's strided-y support lives in the relu
+            # mask kernel.
+            layer._out_slice = None
+            assert layer.relu and out.shape == (N, plan.P, plan.Q,
+                                                plan.Kphys)
+            y = out
+        else:
+            y = torch.empty((N, plan.P, plan.Q, plan.Kphys),
+                            dtype=torch.bfloat16, device=x.device)
         C.conv_fwd(xp, layer.w_bf16,
                    layer.b_f32 if layer.b_f32 is not None else
                    _EMPTY_F32.get(x.device),
@@ -442,8 +480,10 @@ class Conv2d(_Base):
             self.out_shape = (P, Q, K)
             self.out_logical_c = K
 
-    def forward(self, x):
+    def forward(self, x, out=None):
         if self.gpu:
+            if out is not None:
+                self._out_slice = out
             return _HipConvFn.apply(x, self.weight, self.bias, self)
         c_log = self.Cg_logical * self.groups
         if x.shape[-1] != c_log:
