@@ -265,6 +265,13 @@ class _EmptyF32Cache:
 _EMPTY_F32 = _EmptyF32Cache()
 
 
+def _arena_alias(v):
+    """Non-view tensor aliasing the same storage as view ``v``."""
+    t = torch.empty(0, dtype=v.dtype, device=v.device)
+    t.set_(v.untyped_storage(), v.storage_offset(), v.shape)
+    return t
+
+
 class _JoinViews(torch.autograd.Function):
     """Declare that ``buf`` is composed of the given channel slices
     (already written in place by concat-writing convs).  Forward is a
@@ -360,18 +367,23 @@ class _HipConvFn(torch.autograd.Function):
         # by zero_grad; they are only safe to hand to autograd while
         # .grad is None (otherwise AccumulateGrad would add the slice
         # into itself), so manual grad accumulation falls back to a
-        # fresh zeroed buffer.
+        # fresh zeroed buffer.  _arena_alias re-wraps the slice as a
+        # fresh NON-view alias: AccumulateGrad deep-copies any returned
+        # grad that is a view (one copyBuffer per parameter per step),
+        # but steals a single-ref non-view tensor.
         db = None
         if layer.b_f32 is not None:
             db = getattr(layer, "_db_buf", None)
-            if db is None or layer.bias.grad is not None:
-                db = torch.zeros_like(layer.b_f32)
+            db = _arena_alias(db) if (db is not None
+                                      and layer.bias.grad is None) else \
+                torch.zeros_like(layer.b_f32)
             C.bias_wgrad(dy, db)
         # weight grad (fp32, flat layout)
         dw = getattr(layer, "_dw_buf", None)
-        if dw is None or layer.weight.grad is not None:
-            dw = torch.zeros(plan.flat_shape(), dtype=torch.float32,
-                             device=dy.device)
+        dw = _arena_alias(dw) if (dw is not None
+                                  and layer.weight.grad is None) else \
+            torch.zeros(plan.flat_shape(), dtype=torch.float32,
+                        device=dy.device)
         C.conv_wgrad(xp, dy, plan.table, dw, plan.kdf_mask,
                      plan.Wp * plan.Cphys, plan.Cphys,
                      plan.stride[0], plan.stride[1],
