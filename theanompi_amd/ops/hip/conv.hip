@@ -722,13 +722,18 @@ __global__ void bias_wgrad_k(const __bf16* __restrict__ dy,
 #pragma unroll
   for (int j = 0; j < 8; ++j) red[rl][cl][j] = s[j];
   __syncthreads();
-  if (rl == 0 && c8 < c8n) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
+  // final reduce spread over (channel-group, j) pairs: CLW*8 threads
+  // each sum RLN values — the rl==0-only form ran 8 threads x 256
+  // serial adds at CLW=8
+  int t = threadIdx.x;
+  if (t < CLW * 8) {
+    int cl2 = t / 8, j = t & 7;
+    int c82 = blockIdx.x * CLW + cl2;
+    if (c82 < c8n) {
       float a = 0.f;
 #pragma unroll
-      for (int t = 0; t < RLN; ++t) a += red[t][cl][j];
-      atomicAdd(&db[c8 * 8 + j], a);
+      for (int r = 0; r < RLN; ++r) a += red[r][cl2][j];
+      atomicAdd(&db[c82 * 8 + j], a);
     }
   }
 }
