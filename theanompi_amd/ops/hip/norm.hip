@@ -182,17 +182,19 @@ __global__ void bn_sums_k(const __bf16* __restrict__ x,
     red[1][rl][cl][j] = b[j];
   }
   __syncthreads();
-  if (rl == 0 && c8 < c8n) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
+  int t = threadIdx.x;          // reduce spread over (channel, j) pairs
+  if (t < CLW * 8) {
+    int cl2 = t / 8, j = t & 7;
+    int c82 = blockIdx.x * CLW + cl2;
+    if (c82 < c8n) {
       float sa = 0.f, sb = 0.f;
 #pragma unroll
-      for (int t = 0; t < RLN; ++t) {
-        sa += red[0][t][cl][j];
-        sb += red[1][t][cl][j];
+      for (int r = 0; r < RLN; ++r) {
+        sa += red[0][r][cl2][j];
+        sb += red[1][r][cl2][j];
       }
-      atomicAdd(&s1[c8 * 8 + j], sa);
-      atomicAdd(&s2[c8 * 8 + j], sb);
+      atomicAdd(&s1[c82 * 8 + j], sa);
+      atomicAdd(&s2[c82 * 8 + j], sb);
     }
   }
 }
@@ -339,17 +341,19 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
     red[1][rl][cl][j] = b[j];
   }
   __syncthreads();
-  if (rl == 0 && c8 < c8n) {
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
+  int t = threadIdx.x;          // reduce spread over (channel, j) pairs
+  if (t < CLW * 8) {
+    int cl2 = t / 8, j = t & 7;
+    int c82 = blockIdx.x * CLW + cl2;
+    if (c82 < c8n) {
       float sa = 0.f, sb = 0.f;
 #pragma unroll
-      for (int t = 0; t < RLN; ++t) {
-        sa += red[0][t][cl][j];
-        sb += red[1][t][cl][j];
+      for (int r = 0; r < RLN; ++r) {
+        sa += red[0][r][cl2][j];
+        sb += red[1][r][cl2][j];
       }
-      atomicAdd(&sdy[c8 * 8 + j], sa);
-      atomicAdd(&sdyx[c8 * 8 + j], sb);
+      atomicAdd(&sdy[c82 * 8 + j], sa);
+      atomicAdd(&sdyx[c82 * 8 + j], sb);
     }
   }
 }
