@@ -373,3 +373,41 @@ def test_imagenet_mean_image(tmp_path):
     finally:
         d.close()
         d2.close()
+
+
+def test_row_strided_detector():
+    """_row_strided: channel slices of a concat buffer qualify; other
+    non-contiguous layouts do not."""
+    from theanompi_amd.models.layers import _row_strided
+    t = torch.zeros(2, 4, 4, 32)
+    assert _row_strided(t)                # contiguous is ld == C
+    sl = t[..., 8:24]
+    assert _row_strided(sl)
+    assert not _row_strided(t.permute(0, 3, 1, 2))
+    assert not _row_strided(t[:, ::2])    # strided H: rows not uniform
+
+
+def test_arena_alias_not_view():
+    """_arena_alias shares storage but carries no autograd view
+    metadata (so grad accumulation can steal it without a clone)."""
+    from theanompi_amd.models.layers import _arena_alias
+    arena = torch.zeros(100)
+    v = arena[10:30].view(4, 5)
+    a = _arena_alias(v)
+    assert not a._is_view()
+    a[0, 0] = 7.0
+    assert arena[10].item() == 7.0
+    assert a.shape == (4, 5)
+
+
+def test_grad_arena_wiring_cpu():
+    """On CPU there is no arena (GPU-only fast path) and models still
+    train; the opt tolerates grad_arena=None."""
+    from theanompi_amd.models.test_model import TestModel
+    m = TestModel({"rank": 0, "size": 1, "device": "cpu",
+                   "verbose": False, "batch_size": 8, "n_iters": 2,
+                   "val_iters": 1})
+    assert m.grad_arena is None
+    m.compile_iter_fns("grad")
+    m.train_iter(0)
+    m.apply_update()
