@@ -167,3 +167,33 @@ def test_strategy_equivalence(tmp_path):
     for k in outs["ar"].files:
         np.testing.assert_allclose(outs["ar"][k], outs["asa"][k],
                                    atol=2e-5)
+
+
+def test_heartbeat_aborts_wedged_job(tmp_path):
+    """End-to-end failure detection: one BSP rank wedges mid-epoch
+    (process and sockets alive, heartbeat silent — the hung-collective
+    analog the transport cannot detect); the watchdog on the OTHER
+    rank aborts with exit code 13 instead of hanging forever
+    (SURVEY §5.3)."""
+    import time
+    from theanompi_amd import BSP
+    from theanompi_amd.lib.heartbeat import Heartbeat
+    rule = BSP()
+    rule.init(devices=["cpu", "cpu"],
+              modelfile="theanompi_amd.models.test_model",
+              modelclass="TestModel", epochs=50, sync_type="cdd",
+              exchange_strategy="ar", seed=1, heartbeat=4,
+              die_at=3, die_rank=1, die_mode="wedge",
+              n_iters=4, val_iters=1)
+    t0 = time.time()
+    deadline = t0 + 120
+    codes = [None, None]
+    while time.time() < deadline:
+        codes = [p.poll() for p in rule.procs]
+        if codes[0] is not None:
+            break
+        time.sleep(0.5)
+    # rank 0 must have aborted via the watchdog, well before any
+    # distributed timeout would fire
+    assert codes[0] == Heartbeat.EXIT_CODE, codes
+    rule.terminate()

@@ -53,12 +53,27 @@ def run():
                                 else None)})
     proc.barrier()
 
+    # test hooks: simulate a failing rank.  die_mode "exit" dies
+    # abruptly (transport-detectable); "wedge" keeps the process and
+    # its sockets alive but silences the heartbeat — the analog of a
+    # rank wedged in a hung collective, which only the watchdog catches
+    die_at = int(_env_opt("die_at", -1))
+    die_rank = int(_env_opt("die_rank", 0))
+    die_mode = _env_opt("die_mode", "exit")
     for epoch in range(n_epochs):
         rec.start_epoch()
         if epoch > 0:
             model.reset_iter("train")   # per-epoch shard reshuffle
         n_batches = model.data.n_batch_train
         for i in range(n_batches):
+            if die_at >= 0 and rec.uidx >= die_at \
+                    and proc.rank == die_rank:
+                if die_mode == "wedge":
+                    if hb:
+                        hb.stop()
+                    import time as _t
+                    _t.sleep(3600)
+                os._exit(1)
             rec.start()
             model.train_iter(i, rec)
             with _rtx("exchange"):
