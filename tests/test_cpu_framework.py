@@ -411,3 +411,24 @@ def test_grad_arena_wiring_cpu():
     m.compile_iter_fns("grad")
     m.train_iter(0)
     m.apply_update()
+
+
+def test_alexnet_on_shard_pipeline(tmp_path):
+    """AlexNet consumes the real shard pipeline when the session passes
+    train_shards (the reference's ImageNet workflow)."""
+    from theanompi_amd.models.alex_net import AlexNet
+    from theanompi_amd.models.data.imagenet import (ImageNetData,
+                                                    write_synthetic_shards)
+    paths = write_synthetic_shards(str(tmp_path), n_shards=3,
+                                   images_per_shard=4, hw=232)
+    m = AlexNet({"rank": 0, "size": 1, "device": "cpu", "verbose": False,
+                 "batch_size": 4, "train_shards": ",".join(paths),
+                 "val_shards": paths[0]})
+    try:
+        assert isinstance(m.data, ImageNetData)
+        m.compile_iter_fns("grad")
+        loss, err = m.train_iter(0)
+        m.apply_update()
+        assert 0 <= float(err) <= 1
+    finally:
+        m.data.close()

@@ -304,6 +304,26 @@ class BaseModel:
         helper_funcs.load_momentums(self.vels, directory, epoch)
         self._refresh_layers()
 
+    def imagenet_or_synthetic(self, crop, synthetic_factory):
+        """Real shard pipeline when the session passed
+        ``train_shards``/``val_shards`` (comma-separated .npy shard
+        paths, reference workflow SURVEY §2.5); synthetic otherwise."""
+        shards = self.config.get("train_shards", "")
+        if not shards:
+            return synthetic_factory()
+        from theanompi_amd.models.data.imagenet import ImageNetData
+        train = [p for p in str(shards).split(",") if p]
+        val = [p for p in
+               str(self.config.get("val_shards", "")).split(",") if p] \
+            or train[:1]
+        mean = self.config.get("mean_path", "") or None
+        return ImageNetData(train, val, batch_size=self.batch_size,
+                            crop=crop, n_classes=self.n_classes,
+                            device=self.device, rank=self.rank,
+                            size=self.size,
+                            c_phys=4 if self.gpu else 3,
+                            mean_path=mean)
+
     def reset_iter(self, mode):
         if hasattr(self.data, "reset"):
             self.data.reset(mode)

@@ -126,6 +126,8 @@ class ImageNetData:
             self.copy_events = [torch.cuda.Event() for _ in range(depth)]
             self.done_events = [torch.cuda.Event() for _ in range(depth)]
             self._slot_used = [False] * depth
+            self._debug = os.environ.get("THEANOMPI_AMD_DEBUG", "0") \
+                not in ("0", "")
 
     def _partition(self):
         shards = list(self.train_shards)
@@ -165,6 +167,8 @@ class ImageNetData:
                     # don't overwrite a device slot the compute stream
                     # may still be reading
                     self.copy_stream.wait_event(self.done_events[slot])
+                    if self._debug:       # §5.2 race-detection mode
+                        self.done_events[slot].synchronize()
                 self.dev_bufs[slot].copy_(x_u8, non_blocking=True)
                 self.dev_lbufs[slot].copy_(y, non_blocking=True)
                 self.copy_events[slot].record(self.copy_stream)

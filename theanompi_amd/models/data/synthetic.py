@@ -15,6 +15,7 @@ the compute stream.  BASELINE benchmarks use this path with random data
 and random labels (no network, no datasets in the environment).
 """
 
+import os
 import threading
 
 import torch
@@ -124,6 +125,13 @@ class _PinnedRingLoader:
         self.slot = 0
         self._last_slot = None
         self._lock = threading.Lock()
+        # SURVEY §5.2 race-detection mode (THEANOMPI_AMD_DEBUG=1):
+        # host-synchronize the consumer fence before every slot reuse
+        # and keep a per-slot generation counter, so an overwrite-in-
+        # flight would fail here instead of corrupting a batch.
+        self.debug = os.environ.get("THEANOMPI_AMD_DEBUG", "0") \
+            not in ("0", "")
+        self.gen = [0] * depth
         for k in range(depth):  # prime the whole ring
             self._stage(k)
 
@@ -144,6 +152,11 @@ class _PinnedRingLoader:
                 prev = self._last_slot
                 self.done_events[prev].record(compute)
                 self.copy_stream.wait_event(self.done_events[prev])
+                if self.debug:
+                    self.done_events[prev].synchronize()
+                    assert self.copy_events[prev].query(), \
+                        f"slot {prev} refilled before copy completed"
+                self.gen[prev] += 1
                 self._stage(prev)
             slot = self.slot
             compute.wait_event(self.copy_events[slot])

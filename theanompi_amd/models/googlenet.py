@@ -176,6 +176,10 @@ class GoogLeNet(BaseModel):
         return loss.detach(), err
 
     def build_data(self):
+        return self.imagenet_or_synthetic(
+            self.image_size, lambda: self._synthetic_data())
+
+    def _synthetic_data(self):
         return SyntheticData(
             batch_size=self.batch_size, image_hw=self.image_size,
             n_classes=self.n_classes, device=self.device,

@@ -101,6 +101,10 @@ class ResNet50(BaseModel):
                             self.n_classes, self.gpu)
 
     def build_data(self):
+        return self.imagenet_or_synthetic(
+            self.image_size, lambda: self._synthetic_data())
+
+    def _synthetic_data(self):
         return SyntheticData(
             batch_size=self.batch_size, image_hw=self.image_size,
             n_classes=self.n_classes, device=self.device,

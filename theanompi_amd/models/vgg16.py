@@ -54,6 +54,10 @@ class VGG16(BaseModel):
         return m
 
     def build_data(self):
+        return self.imagenet_or_synthetic(
+            self.image_size, lambda: self._synthetic_data())
+
+    def _synthetic_data(self):
         return SyntheticData(
             batch_size=self.batch_size, image_hw=self.image_size,
             n_classes=self.n_classes, device=self.device,
