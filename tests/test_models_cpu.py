@@ -134,3 +134,29 @@ def test_zoo_alias_paths():
     assert VGG16.name == "vgg16"
     assert ResNet50.name == "resnet50"
     assert WResNet.name == "wresnet"
+
+
+@pytest.mark.gpu
+def test_alexnet_shard_pipeline_gpu(tmp_path):
+    """AlexNet trains from the real shard pipeline on GPU: loader
+    child -> pinned ring -> u8_finish -> HIP conv stack."""
+    from theanompi_amd.models.data.imagenet import write_synthetic_shards
+    paths = write_synthetic_shards(str(tmp_path), n_shards=3,
+                                   images_per_shard=8, hw=232)
+    import importlib
+    cls = getattr(importlib.import_module("theanompi_amd.models.alex_net"),
+                  "AlexNet")
+    m = cls({"rank": 0, "size": 1, "device": "cuda:0", "verbose": False,
+             "batch_size": 8, "train_shards": ",".join(paths),
+             "val_shards": paths[0]})
+    try:
+        m.compile_iter_fns("grad")
+        l0, _ = m.train_iter(0)
+        m.apply_update()
+        l1, _ = m.train_iter(1)
+        m.apply_update()
+        assert np.isfinite(float(l0)) and np.isfinite(float(l1))
+        c, t1, t5 = m.val_iter(0)
+        assert np.isfinite(c)
+    finally:
+        m.data.close()
