@@ -12,7 +12,6 @@ matches the reference's scaling tables.
 """
 
 import json
-import os
 import pickle
 import time
 

@@ -4,7 +4,6 @@ bottleneck residual blocks with BatchNorm (N7 kernels), batch 256/GPU
 per BASELINE config #5 (288 GB HBM sizing).
 """
 
-import torch
 import torch.nn as nn
 
 from theanompi_amd.models import layers as L

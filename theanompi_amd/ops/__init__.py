@@ -9,8 +9,6 @@ reference implementations in ``functional.py`` run instead — they define
 the numerics the HIP kernels are tested against.
 """
 
-import os
-
 import torch
 
 _C = None

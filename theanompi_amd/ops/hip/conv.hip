@@ -845,7 +845,12 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
                              hipStream_t stream) {
   WgradArgs a = *args;
   if (a.Kdf > 2048 * 16) return hipErrorInvalidValue;
+  static const int bnf_env = [] {
+    const char* e = getenv("THEANOMPI_WGRAD_BNF");
+    return e ? atoi(e) : 0;
+  }();
   int BNF = (a.Kdf % 128 == 0) ? 128 : 64;  // 128-tile halves occupancy; only pays on dense-Kdf layers
+  if (bnf_env == 64) BNF = 64;
   dim3 grid(cdiv(a.Kgp, 64), cdiv(a.Kdf, BNF), G * a.splitK);
   dim3 block(256);
   if (BNF == 128) {
