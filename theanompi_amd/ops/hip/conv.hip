@@ -849,8 +849,11 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
     const char* e = getenv("THEANOMPI_WGRAD_BNF");
     return e ? atoi(e) : 0;
   }();
-  int BNF = (a.Kdf % 128 == 0) ? 128 : 64;  // 128-tile halves occupancy; only pays on dense-Kdf layers
-  if (bnf_env == 64) BNF = 64;
+  // 64-column tiles (occupancy 4) measured faster than the old
+  // dense-Kdf 128-tile heuristic (occupancy 2) once the reduction
+  // epilogues were parallelized; THEANOMPI_WGRAD_BNF=128 restores it.
+  int BNF = 64;
+  if (bnf_env == 128 && a.Kdf % 128 == 0) BNF = 128;
   dim3 grid(cdiv(a.Kgp, 64), cdiv(a.Kdf, BNF), G * a.splitK);
   dim3 block(256);
   if (BNF == 128) {
