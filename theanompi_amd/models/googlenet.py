@@ -17,18 +17,6 @@ from theanompi_amd.models.data.synthetic import SyntheticData
 from theanompi_amd.ops import functional as OF
 
 
-_MS = os.environ.get("THEANOMPI_MULTISTREAM", "0") not in ("0", "")
-_STREAM_POOL = {}
-
-
-def _side_streams(device, n=3):
-    key = (device.index, n)
-    if key not in _STREAM_POOL:
-        _STREAM_POOL[key] = [torch.cuda.Stream(device=device)
-                             for _ in range(n)]
-    return _STREAM_POOL[key]
-
-
 class _Inception(nn.Module):
     """branches: 1x1 | 1x1->3x3 | 1x1->5x5 | pool->1x1, concat on C."""
 
@@ -60,31 +48,7 @@ class _Inception(nn.Module):
             ws = [self.b1.out_shape[2], self.b3b.out_shape[2],
                   self.b5b.out_shape[2], self.bp.out_shape[2]]
             o = [0, ws[0], ws[0] + ws[1], ws[0] + ws[1] + ws[2]]
-            if _MS:
-                # THEANOMPI_MULTISTREAM=1: the four branches run on
-                # separate HIP streams (batch-32 branch GEMMs underfill
-                # 256 CUs individually).  Backward replays each op on
-                # its forward stream; torch syncs the ambient stream on
-                # backward() return.
-                main = torch.cuda.current_stream()
-                ev_x = torch.cuda.Event()
-                ev_x.record(main)
-                sides = _side_streams(x.device)
-                branches = [
-                    lambda sl: self.b3b(self.b3a(x), out=sl),
-                    lambda sl: self.b5b(self.b5a(x), out=sl),
-                    lambda sl: self.bp(self.pool(x), out=sl),
-                ]
-                ys = []
-                for st, fn, (c0, c1) in zip(
-                        sides, branches,
-                        [(o[1], o[2]), (o[2], o[3]), (o[3], self.out_c)]):
-                    st.wait_event(ev_x)
-                    x.record_stream(st)
-                    buf.record_stream(st)
-                    with torch.cuda.stream(st):
-                        ys.append(fn(buf[..., c0:c1]))
-                y1 = self.b1(x, out=buf[..., o[0]:o[1]])
+            y1 = self.b1(x, out=buf[..., o[0]:o[1]])
                 for st in sides:
                     ev = torch.cuda.Event()
                     ev.record(st)
