@@ -49,12 +49,6 @@ class _Inception(nn.Module):
                   self.b5b.out_shape[2], self.bp.out_shape[2]]
             o = [0, ws[0], ws[0] + ws[1], ws[0] + ws[1] + ws[2]]
             y1 = self.b1(x, out=buf[..., o[0]:o[1]])
-                for st in sides:
-                    ev = torch.cuda.Event()
-                    ev.record(st)
-                    main.wait_event(ev)
-                return L.join_views(buf, y1, *ys)
-            y1 = self.b1(x, out=buf[..., o[0]:o[1]])
             y3 = self.b3b(self.b3a(x), out=buf[..., o[1]:o[2]])
             y5 = self.b5b(self.b5a(x), out=buf[..., o[2]:o[3]])
             yp = self.bp(self.pool(x), out=buf[..., o[3]:])
