@@ -6,7 +6,6 @@ stress exchanger bucketing, and the 1x1/3x3/5x5 mix stresses the
 implicit-GEMM tiler.
 """
 
-import os
 
 import torch
 import torch.nn as nn
