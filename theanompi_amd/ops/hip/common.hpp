@@ -103,8 +103,15 @@ DEV bf16x8 frag_load_tr_swz(const __bf16* tile, int ks, int ch0,
   const int pix1 = ks + 8 * (lane >> 4) + ((lane & 15) >> 2);
   const int pix2 = pix1 + 4;
   const int co = ch0 + 4 * (lane & 3);
-  const __bf16* a1 = tile + pix1 * LDW + (co ^ ((pix1 & 7) * 8));
-  const __bf16* a2 = tile + pix2 * LDW + (co ^ ((pix2 & 7) * 8));
+  // swizzle: (pix&7) term decorrelates rows; the (pix>>3) term
+  // decorrelates the four 16-lane groups, whose pixel octaves share a
+  // bank phase (without it the tr-reads ran 4-8 lanes/bank — measured
+  // as SQ_LDS_BANK_CONFLICT = 86% of wgrad's SQ_BUSY; with it the
+  // model hits the ideal 2).  Must match the wgrad staging map.
+  const int s1 = ((pix1 & 7) * 8) ^ (((pix1 >> 3) & 3) * (LDW / 4));
+  const int s2 = ((pix2 & 7) * 8) ^ (((pix2 >> 3) & 3) * (LDW / 4));
+  const __bf16* a1 = tile + pix1 * LDW + (co ^ s1);
+  const __bf16* a2 = tile + pix2 * LDW + (co ^ s2);
   bf16x4 lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf16x4*)a1);
   bf16x4 hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf16x4*)a2);
   bf16x8 r;

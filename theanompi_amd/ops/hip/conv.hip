@@ -573,7 +573,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
       int piece = (wave * XI + i) * 64 + lane;
       int row = piece / (BNF / 8);
       int k8i = piece % (BNF / 8);
-      int k8 = ((k8i ^ row) & 7) * 8 + (k8i & ~7) * 8;   // swizzle low 3
+      // slot k8i*8 holds logical k = slot ^ s(row); s matches
+      // frag_load_tr_swz (row-octave XOR + group-phase XOR)
+      int k8 = (k8i * 8) ^ ((row & 7) * 8) ^ (((row >> 3) & 3)
+                                              * (BNF / 4));
       int pa = pix0 + row;
       int pc = pa < a.M ? pa : a.M - 1;
       int q = pc % a.Q;
@@ -595,7 +598,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
     for (int i = 0; i < DI; ++i) {
       int piece = (wave * DI + i) * 64 + lane;
       int row = piece >> 3;
-      int k8 = ((piece & 7) ^ (row & 7)) * 8;
+      int k8 = (((piece & 7) * 8) ^ ((row & 7) * 8)
+                ^ (((row >> 3) & 3) * 16));
       int pa = pix0 + row;
       int pc = pa < a.M ? pa : a.M - 1;
       long dyaddr = (long)pc * a.Kphys + g * a.Kgp + kout0 + k8;
