@@ -432,3 +432,27 @@ def test_alexnet_on_shard_pipeline(tmp_path):
         assert 0 <= float(err) <= 1
     finally:
         m.data.close()
+
+
+def test_recorder_jsonl(tmp_path):
+    """dump_jsonl emits one machine-readable row per train record with
+    the calc/comm/wait split (the scaling-curve harness input)."""
+    import json
+    from theanompi_amd.lib.recorder import Recorder
+    r = Recorder({"rank": 2, "verbose": False})
+    for i in range(3):
+        r.start()
+        r.end("calc")
+        r.start()
+        r.end("comm")
+        r.train_error(i, 2.0 - i, 0.5)
+        r.print_train_info(i, force=True)
+    p = tmp_path / "rec.jsonl"
+    r.dump_jsonl(str(p))
+    rows = [json.loads(line) for line in open(p)]
+    assert len(rows) == 3
+    for row in rows:
+        assert set(row) == {"epoch", "iter", "cost", "error", "calc_s",
+                            "comm_s", "wait_s", "rank"}
+        assert row["rank"] == 2
+        assert row["calc_s"] >= 0 and row["comm_s"] >= 0
