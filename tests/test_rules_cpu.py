@@ -197,3 +197,20 @@ def test_heartbeat_aborts_wedged_job(tmp_path):
     # distributed timeout would fire
     assert codes[0] == Heartbeat.EXIT_CODE, codes
     rule.terminate()
+
+
+def test_gosgd_checkpoint(tmp_path):
+    """GOSGD rank 0 writes the reference-format checkpoint at the end
+    of training (W_<i>_<epoch>.npy per param)."""
+    import os
+    from theanompi_amd import GOSGD
+    rule = GOSGD()
+    rule.init(devices=["cpu", "cpu"],
+              modelfile="theanompi_amd.models.test_model",
+              modelclass="TestModel",
+              epochs=1, p_push=0.5, seed=21,
+              save_dir=str(tmp_path))
+    assert rule.wait() == [0, 0]
+    files = sorted(os.listdir(tmp_path))
+    assert any(f.startswith("W_0_") and f.endswith(".npy")
+               for f in files), files
