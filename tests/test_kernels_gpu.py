@@ -493,3 +493,74 @@ def test_inception_concat_write_equivalence():
     assert rel_err(dx_new, dx_old.cpu()) < 1e-6
     for a, b in zip(g_new, g_old):
         assert rel_err(a, b.cpu()) < 2e-2       # wgrad atomics reorder
+
+
+@pytest.mark.gpu
+def test_conv_fused_bn_stats():
+    """The conv epilogue's bucketed sum/sum-sq must equal torch sums
+    over the produced bf16 output (fused BN statistics)."""
+    import torch as T
+    from theanompi_amd.models import layers as L
+    T.manual_seed(12)
+    d = dev()
+    conv = L.Conv2d((14, 14, 64), 96, 3, pad=1, relu=False,
+                    device=str(d), batch=8, collect_stats=True,
+                    has_bias=False)
+    conv.train()
+    x = q(T.randn(8, 14, 14, 64)).to(d).to(T.bfloat16)
+    y = conv(x)
+    st = conv.pop_stats()
+    assert st is not None
+    C = y.shape[-1]
+    got = st.view(32, 2, C).sum(dim=0)
+    yf = y.detach().float().reshape(-1, C)
+    ref1 = yf.sum(dim=0)
+    ref2 = (yf * yf).sum(dim=0)
+    assert rel_err(got[0], ref1.cpu()) < 1e-4
+    assert rel_err(got[1], ref2.cpu()) < 1e-4
+    # buckets are consumed by the BN side; a second forward refills
+    y2 = conv(x)
+    st2 = conv.pop_stats()
+    assert st2 is not None
+
+
+@pytest.mark.gpu
+def test_convbn_fused_stats_equivalence():
+    """_ConvBN with fused conv-epilogue statistics matches the separate
+    bn_sums path (forward + grads)."""
+    import torch as T
+    from theanompi_amd.models import resnet50 as R
+    T.manual_seed(13)
+    d = dev()
+
+    def build(fuse):
+        old = R._BN_FUSE
+        R._BN_FUSE = fuse
+        try:
+            T.manual_seed(13)
+            m = R._ConvBN(10, 32, 48, 3, pad=1, relu=True, dev=str(d),
+                          B=4)
+        finally:
+            R._BN_FUSE = old
+        m.train()
+        return m
+
+    ma, mb = build(True), build(False)
+    mb.load_state_dict(ma.state_dict())
+    x = q(T.randn(4, 10, 10, 32)).to(d).to(T.bfloat16)
+
+    def run(m):
+        xi = x.clone().requires_grad_()
+        y = m(xi)
+        (y.float() ** 2).mean().backward()
+        gs = [p.grad.detach().clone() for p in m.parameters()]
+        for p in m.parameters():
+            p.grad = None
+        return y.detach().float(), xi.grad.detach().float(), gs
+
+    ya, dxa, ga = run(ma)
+    yb, dxb, gb = run(mb)
+    assert rel_err(ya, yb.cpu()) < 1e-3
+    assert rel_err(dxa, dxb.cpu()) < 1e-2
+    for a, b in zip(ga, gb):
+        assert rel_err(a, b.cpu()) < 2e-2

@@ -318,6 +318,13 @@ class _HipConvFn(torch.autograd.Function):
         xp = C.halo_pad(x, ph, pw, 0.0) if (ph or pw) else x.contiguous()
         N = x.shape[0]
         M = N * plan.P * plan.Q
+        stats = None
+        if layer._collect_stats and layer.training:
+            if layer._stats_buf is None:
+                layer._stats_buf = torch.zeros(
+                    32 * 2 * layer.plan.Kphys, dtype=torch.float32,
+                    device=x.device)
+            stats = layer._stats_buf
         out = getattr(layer, "_out_slice", None)
         if out is not None:
             # concat-writing conv: the epilogue writes a channel slice
@@ -332,15 +339,16 @@ class _HipConvFn(torch.autograd.Function):
         else:
             y = torch.empty((N, plan.P, plan.Q, plan.Kphys),
                             dtype=torch.bfloat16, device=x.device)
-        C.conv_fwd(xp, layer.w_bf16,
-                   layer.b_f32 if layer.b_f32 is not None else
-                   _EMPTY_F32.get(x.device),
-                   plan.table, y,
-                   plan.Wp * plan.Cphys,            # pixel row stride (h)
-                   plan.Cphys,                      # pixel col stride (w)
-                   plan.stride[0], plan.stride[1],
-                   plan.Cg if plan.G > 1 else 0,    # group in-chan offset
-                   bool(layer.relu))
+        layer._stats_valid = C.conv_fwd(
+            xp, layer.w_bf16,
+            layer.b_f32 if layer.b_f32 is not None else
+            _EMPTY_F32.get(x.device),
+            plan.table, y,
+            plan.Wp * plan.Cphys,                # pixel row stride (h)
+            plan.Cphys,                          # pixel col stride (w)
+            plan.stride[0], plan.stride[1],
+            plan.Cg if plan.G > 1 else 0,        # group in-chan offset
+            bool(layer.relu), stats)
         ctx.save_for_backward(xp, y if layer.relu else torch.empty(0))
         ctx.layer = layer
         return y
@@ -436,8 +444,15 @@ class Conv2d(_Base):
 
     def __init__(self, in_shape, K, R, S=None, stride=1, pad=0, groups=1,
                  relu=False, init=("he",), bias_init=0.0, device="cpu",
-                 batch=None, needs_dgrad=True, has_bias=True):
+                 batch=None, needs_dgrad=True, has_bias=True,
+                 collect_stats=False):
         super().__init__()
+        # collect_stats: the epilogue also produces bucketed per-channel
+        # sum/sum-sq of the output for a following BatchNorm (fused
+        # stats; consumer fetches via pop_stats())
+        self._collect_stats = collect_stats
+        self._stats_buf = None
+        self._stats_valid = False
         self.needs_dgrad = needs_dgrad
         self.has_bias = has_bias
         S = S if S is not None else R
@@ -502,6 +517,15 @@ class Conv2d(_Base):
             x = x[..., :c_log]
         return OF.conv2d_ref(x, self.weight, self.bias, self.stride,
                              self.pad, self.groups, self.relu)
+
+    def pop_stats(self):
+        """Bucketed output statistics from the last forward (fused BN),
+        or None if the epilogue could not collect them (split-K path /
+        alternate tiles) — the consumer then falls back to bn_sums."""
+        if not self._stats_valid:
+            return None
+        self._stats_valid = False
+        return self._stats_buf
 
     def _dgrad(self, dy):
         C = _ops.ext()
@@ -722,10 +746,10 @@ class BatchNorm2d(_Base):
                              torch.ones(c_phys, device=dev))
         self.eps, self.momentum, self.relu = eps, momentum, relu
 
-    def forward(self, x):
+    def forward(self, x, prestats=None):
         return OF.batchnorm(x, self.gamma, self.beta, self.running_mean,
                             self.running_var, self.training, self.momentum,
-                            self.eps, self.relu)
+                            self.eps, self.relu, prestats=prestats)
 
     def logical_params(self):
         return [("gamma", lambda: self.gamma.data,

@@ -4,6 +4,8 @@ bottleneck residual blocks with BatchNorm (N7 kernels), batch 256/GPU
 per BASELINE config #5 (288 GB HBM sizing).
 """
 
+import os
+
 import torch.nn as nn
 
 from theanompi_amd.models import layers as L
@@ -12,21 +14,31 @@ from theanompi_amd.models.data.synthetic import SyntheticData
 from theanompi_amd.ops import functional as OF
 
 
+# fused conv-epilogue BN statistics (the conv's LDS-staged epilogue
+# also emits per-channel sum/sum-sq, replacing the separate bn_sums
+# pass); measured on/off via this gate
+_BN_FUSE = os.environ.get("THEANOMPI_BN_FUSE", "1") not in ("0", "")
+
+
 class _ConvBN(nn.Module):
     def __init__(self, hw, cin, k, r, stride=1, pad=0, relu=True,
                  dev="cpu", B=1, needs_dgrad=True):
         super().__init__()
+        gpu = str(dev).startswith("cuda")
         self.conv = L.Conv2d((hw, hw, cin), k, r, stride=stride, pad=pad,
                              relu=False, device=dev, batch=B,
                              needs_dgrad=needs_dgrad, init=("he",),
-                             has_bias=False)    # BN supplies the shift
+                             has_bias=False,    # BN supplies the shift
+                             collect_stats=_BN_FUSE and gpu)
         self.bn = L.BatchNorm2d(self.conv.out_shape[2], relu=relu,
                                 device=dev)
         self.out_hw = self.conv.out_shape[0]
         self.out_c = self.conv.out_shape[2]
 
     def forward(self, x):
-        return self.bn(self.conv(x))
+        y = self.conv(x)
+        pre = self.conv.pop_stats() if self.conv.gpu else None
+        return self.bn(y, prestats=pre)
 
 
 class _Bottleneck(nn.Module):

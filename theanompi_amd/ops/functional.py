@@ -196,7 +196,7 @@ class _LRNFn(torch.autograd.Function):
 class _BatchNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var,
-                training, momentum, eps, relu):
+                training, momentum, eps, relu, prestats=None):
         C = _ops.ext()
         y = torch.empty_like(x)
         c = x.shape[-1]
@@ -204,7 +204,7 @@ class _BatchNormFn(torch.autograd.Function):
         save_invstd = torch.empty(c, dtype=torch.float32, device=x.device)
         C.bn_fwd(x, gamma, beta, y, save_mean, save_invstd,
                  running_mean, running_var, float(momentum), float(eps),
-                 bool(training), bool(relu))
+                 bool(training), bool(relu), prestats)
         ctx.save_for_backward(x, gamma, beta, save_mean, save_invstd, y)
         ctx.relu = relu
         ctx.training = training
@@ -219,7 +219,8 @@ class _BatchNormFn(torch.autograd.Function):
         dbeta = torch.empty_like(gamma)
         C.bn_bwd(dy.contiguous(), x, gamma, beta, save_mean,
                  save_invstd, y, dx, dgamma, dbeta, bool(ctx.relu))
-        return dx, dgamma, dbeta, None, None, None, None, None, None
+        return (dx, dgamma, dbeta, None, None, None, None, None, None,
+                None)
 
 
 class _ReluFn(torch.autograd.Function):
@@ -348,10 +349,11 @@ def lrn(x, size=5, alpha=1e-4, beta=0.75, k=1.0):
 
 
 def batchnorm(x, gamma, beta, running_mean, running_var, training,
-              momentum=0.1, eps=1e-5, relu=False):
+              momentum=0.1, eps=1e-5, relu=False, prestats=None):
     if x.is_cuda:
-        return _BatchNormFn.apply(x, gamma, beta, running_mean, running_var,
-                                  training, momentum, eps, relu)
+        return _BatchNormFn.apply(x, gamma, beta, running_mean,
+                                  running_var, training, momentum, eps,
+                                  relu, prestats)
     return batchnorm_ref(x, gamma.float(), beta.float(), running_mean,
                          running_var, training, momentum, eps, relu)
 

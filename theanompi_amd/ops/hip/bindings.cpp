@@ -30,6 +30,8 @@ struct GatherGemmArgs {
   int splitK;
   float* Y32;
   long xTotal;
+  float* stats;      // fused BN statistics buckets (or nullptr)
+  int statsC;
 };
 
 struct WgradArgs {
@@ -47,7 +49,7 @@ struct WgradArgs {
 };
 
 extern "C" {
-hipError_t launch_conv_gather_gemm(const GatherGemmArgs*, int G, int layout,
+hipError_t launch_conv_gather_gemm(GatherGemmArgs*, int G, int layout,
                                    hipStream_t);
 hipError_t launch_conv_wgrad(const WgradArgs*, int G, int layout,
                              hipStream_t);
@@ -91,9 +93,10 @@ hipError_t launch_lrn_fwd(const void*, void*, void*, long, int, int, float,
 hipError_t launch_lrn_bwd(const void*, const void*, const void*, void*,
                           long, int, int, float, float, float,
                           hipStream_t);
-hipError_t launch_bn_fwd_train(const void*, void*, float*, float*, float*,
-                               float*, float*, float*, const float*,
-                               const float*, long, int, float, float, int,
+hipError_t launch_bn_fwd_train(const void*, void*, float*, float*,
+                               float*, float*, float*, float*,
+                               const float*, const float*, float*, int,
+                               long, int, float, float, int,
                                hipStream_t);
 hipError_t launch_bn_fwd_eval(const void*, void*, float*, float*,
                               const float*, const float*, const float*,
@@ -261,6 +264,8 @@ static torch::Tensor halo_pad_sym(torch::Tensor x, long ph, long pw,
 
 static long row_ld(const torch::Tensor& t);
 
+constexpr int STATS_NB = 32;   // must match conv.hip
+
 // Split-K fp32 workspace, cached per device and re-zeroed by the
 // finalize kernel as it drains it, so the steady state needs no fill
 // launch at all (fills were ~14% of the GoogLeNet step).
@@ -272,10 +277,11 @@ static float* splitk_ws(long numel, const torch::TensorOptions& opts) {
   return ws[d].data_ptr<float>();
 }
 
-static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
+static bool conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
                      torch::Tensor table, torch::Tensor y, long rowStride,
                      long colStride, long sh, long sw, long groupInOff,
-                     bool relu) {
+                     bool relu,
+                     c10::optional<torch::Tensor> stats = c10::nullopt) {
   check_bf16(xp, "xp");
   check_bf16(w, "w");
   // y may be a channel slice of a concat buffer (concat-writing conv:
@@ -322,6 +328,16 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
   a.Y32 = nullptr;
   if (a.splitK > 1)
     a.Y32 = splitk_ws(y.numel(), y.options());
+  a.stats = nullptr;
+  a.statsC = (int)(G * Kgp);
+  if (stats.has_value()) {
+    auto& st = stats.value();
+    TORCH_CHECK(st.scalar_type() == torch::kFloat32 && st.is_cuda()
+                && st.is_contiguous()
+                && st.numel() == (long)STATS_NB * 2 * G * Kgp,
+                "stats must be fp32 [", STATS_NB, ", 2, C]");
+    a.stats = st.data_ptr<float>();
+  }
   CHECK_HIP(launch_conv_gather_gemm(&a, (int)G, detect_layout(),
                                     cur_stream()));
   if (a.splitK > 1) {
@@ -329,6 +345,7 @@ static void conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
                                      y.numel(), (int)(G * Kgp),
                                      a.relu, cur_stream()));
   }
+  return a.stats != nullptr;   // whether fused stats were collected
 }
 
 static void conv_dgrad(torch::Tensor dyp, torch::Tensor wdg,
@@ -370,6 +387,8 @@ static void conv_dgrad(torch::Tensor dyp, torch::Tensor wdg,
   a.xTotal = dyp.numel();
   a.splitK = 1;
   a.Y32 = nullptr;
+  a.stats = nullptr;
+  a.statsC = 0;
   if (sh == 1 && sw == 1) {      // one class = dense dx: split-K legal
     long BN = BNtot > 64 ? 128 : 64;
     long baseBlocks = ((a.M + 127) / 128) * ((BNtot + BN - 1) / BN) * G;
@@ -643,7 +662,8 @@ static void bn_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
                    torch::Tensor y, torch::Tensor save_mean,
                    torch::Tensor save_invstd, torch::Tensor rmean,
                    torch::Tensor rvar, double momentum, double eps,
-                   bool training, bool relu) {
+                   bool training, bool relu,
+                   c10::optional<torch::Tensor> prestats = c10::nullopt) {
   check_bf16(x, "x");
   check_f32(gamma, "gamma");
   long C = x.size(-1);
@@ -651,12 +671,20 @@ static void bn_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
   if (training) {
     auto s1 = torch::zeros_like(save_mean);
     auto s2 = torch::zeros_like(save_mean);
+    float* pre = nullptr;
+    if (prestats.has_value()) {
+      auto& st = prestats.value();
+      TORCH_CHECK(st.numel() == (long)STATS_NB * 2 * C,
+                  "prestats must be [", STATS_NB, ", 2, C]");
+      pre = st.data_ptr<float>();
+    }
     CHECK_HIP(launch_bn_fwd_train(
         x.data_ptr(), y.data_ptr(), s1.data_ptr<float>(),
         s2.data_ptr<float>(), save_mean.data_ptr<float>(),
         save_invstd.data_ptr<float>(), rmean.data_ptr<float>(),
         rvar.data_ptr<float>(), gamma.data_ptr<float>(),
-        beta.data_ptr<float>(), rows, (int)C, (float)momentum, (float)eps,
+        beta.data_ptr<float>(), pre, STATS_NB, rows, (int)C,
+        (float)momentum, (float)eps,
         relu ? 1 : 0, cur_stream()));
   } else {
     CHECK_HIP(launch_bn_fwd_eval(
@@ -852,7 +880,11 @@ static void gather_multi_run(torch::Tensor desc, long nseg, long max_n) {
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("halo_pad", &halo_pad_sym);
   m.def("halo_pad_lrtb", &halo_pad);
-  m.def("conv_fwd", &conv_fwd);
+  m.def("conv_fwd", &conv_fwd, py::arg("xp"), py::arg("w"),
+        py::arg("bias"), py::arg("table"), py::arg("y"),
+        py::arg("rowStride"), py::arg("colStride"), py::arg("sh"),
+        py::arg("sw"), py::arg("groupInOff"), py::arg("relu"),
+        py::arg("stats") = py::none());
   m.def("conv_dgrad", &conv_dgrad);
   m.def("conv_wgrad", &conv_wgrad);
   m.def("bias_wgrad", &bias_wgrad);
@@ -873,7 +905,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("avgpool_bwd", &avgpool_bwd);
   m.def("lrn_fwd", &lrn_fwd);
   m.def("lrn_bwd", &lrn_bwd);
-  m.def("bn_fwd", &bn_fwd);
+  m.def("bn_fwd", &bn_fwd, py::arg("x"), py::arg("gamma"),
+        py::arg("beta"), py::arg("y"), py::arg("save_mean"),
+        py::arg("save_invstd"), py::arg("rmean"), py::arg("rvar"),
+        py::arg("momentum"), py::arg("eps"), py::arg("training"),
+        py::arg("relu"), py::arg("prestats") = py::none());
   m.def("bn_bwd", &bn_bwd);
   m.def("softmax_xent_fwd", &softmax_xent_fwd);
   m.def("softmax_xent_bwd", &softmax_xent_bwd);

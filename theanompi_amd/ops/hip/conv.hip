@@ -42,7 +42,14 @@ struct GatherGemmArgs {
   int splitK;                       // >1: K-slices atomic-add into Y32
   float* __restrict__ Y32;          // fp32 workspace for split-K
   long xTotal;                      // X elements (gather tail clamp)
+  // optional fused BN statistics: bucketed per-channel sum / sum-sq of
+  // the bf16 output, [STATS_NB][2][statsC] fp32 (atomics per bucket).
+  // Collected only on the LDS-staged epilogue with splitK == 1.
+  float* __restrict__ stats;
+  int statsC;
 };
+
+constexpr int STATS_NB = 32;      // stat atomic buckets (contention)
 
 template <int BM, int BN, int LAYOUT, int DEPTH, int BK = 64>
 __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
@@ -283,10 +290,16 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
     }
   }
   __syncthreads();
-  // each lane streams 16-B pieces of consecutive rows out to global
+  // each lane streams 16-B pieces of consecutive rows out to global.
+  // With the piece stride (THREADS is a multiple of BN/8) every thread
+  // keeps ONE fixed 8-channel group, so fused BN statistics ride along
+  // as per-thread partials.
   constexpr int PIECES = (BM * BN) / 8;    // 8 bf16 per piece
   constexpr int THREADS = BM * 2;
+  static_assert(THREADS % (BN / 8) == 0, "fixed c8 per thread");
   const int nvalid = a.BNtot - bn0;        // columns that exist (< BN ok)
+  float s1l[8] = {0, 0, 0, 0, 0, 0, 0, 0};
+  float s2l[8] = {0, 0, 0, 0, 0, 0, 0, 0};
   for (int pc = tid; pc < PIECES; pc += THREADS) {
     int rowl = pc / (BN / 8);
     int c8 = (pc % (BN / 8)) * 8;
@@ -298,11 +311,50 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
     int n = tmp / a.P;
     long obase = a.outOff + (long)n * a.sN + (long)p * a.sP
                  + (long)q * a.sQ + cOut0 + c8;
-    if (c8 + 8 <= nvalid) {
+    int ne = nvalid - c8 < 8 ? nvalid - c8 : 8;
+    if (ne == 8) {
       *(pack16*)(a.Y + obase) = *(const pack16*)(Yt + rowl * BN + c8);
     } else {
-      for (int e = 0; e < nvalid - c8; ++e)
+      for (int e = 0; e < ne; ++e)
         a.Y[obase + e] = Yt[rowl * BN + c8 + e];
+    }
+    if (a.stats) {
+      for (int e = 0; e < ne; ++e) {
+        float v = (float)Yt[rowl * BN + c8 + e];
+        s1l[e] += v;
+        s2l[e] += v * v;
+      }
+    }
+  }
+  if (a.stats) {
+    // cross-thread reduce (threads sharing a c8 group) through Bs,
+    // which is idle after the MFMA loop; exact fit at BN=64.
+    __syncthreads();
+    float* S = (float*)&Bs[0][0][0];      // [THREADS][8] x2
+#pragma unroll
+    for (int e = 0; e < 8; ++e) {
+      S[tid * 8 + e] = s1l[e];
+      S[(THREADS + tid) * 8 + e] = s2l[e];
+    }
+    __syncthreads();
+    constexpr int GROUPS = BN / 8;
+    constexpr int PER = THREADS / GROUPS;
+    if (tid < GROUPS * 8) {
+      int c8i = tid >> 3, e = tid & 7;
+      int c = c8i * 8 + e;
+      if (c < nvalid) {
+        float t1 = 0.f, t2 = 0.f;
+#pragma unroll
+        for (int r = 0; r < PER; ++r) {
+          int src = c8i + r * GROUPS;
+          t1 += S[src * 8 + e];
+          t2 += S[(THREADS + src) * 8 + e];
+        }
+        int bucket = blockIdx.x & (STATS_NB - 1);
+        long gc = cOut0 + c;
+        atomicAdd(&a.stats[((long)bucket * 2) * a.statsC + gc], t1);
+        atomicAdd(&a.stats[((long)bucket * 2 + 1) * a.statsC + gc], t2);
+      }
     }
   }
 }
@@ -759,10 +811,14 @@ hipError_t launch_splitk_finalize(float* y32, const float* bias,
   return hipGetLastError();
 }
 
-hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
+hipError_t launch_conv_gather_gemm(GatherGemmArgs* args, int G,
                                    int layout, hipStream_t stream) {
   GatherGemmArgs a = *args;
   if (a.Kdf > 2048 * 16) return hipErrorInvalidValue;  // LDS table cap
+  // fused BN stats only exist on the default LDS-staged epilogue with
+  // splitK == 1; clear the field (and tell the caller) on every other
+  // route so the host falls back to the separate bn_sums kernel
+  if (a.splitK > 1) { a.stats = nullptr; args->stats = nullptr; }
   // big-tile path: 256x128, 8 waves -- 2x MFMA work per staged byte,
   // for large-M dense layers
   static int bigTile = -1;
@@ -772,6 +828,8 @@ hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
   }
   if (bigTile && a.splitK == 1 && a.M >= 4096 && a.BNtot >= 128
       && a.Kdf >= 128) {
+    a.stats = nullptr;
+    args->stats = nullptr;
     dim3 grid(cdiv(a.M, 256), cdiv(a.BNtot, 128), G);
     dim3 block(512);
     if (bigTile == 2 && a.Kdf >= 3 * 64) {
@@ -799,6 +857,8 @@ hipError_t launch_conv_gather_gemm(const GatherGemmArgs* args, int G,
     bk32 = e ? atoi(e) : 0;
   }
   if (bk32 && a.Kdf % 64 == 0) {
+    a.stats = nullptr;
+    args->stats = nullptr;
     if (BN == 128) {
       if (layout == FRAG_CONTIG8)
         hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_CONTIG8, 2,

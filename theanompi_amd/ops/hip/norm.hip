@@ -199,6 +199,28 @@ __global__ void bn_sums_k(const __bf16* __restrict__ x,
   }
 }
 
+// Reduce the conv epilogue's bucketed statistics ([NB][2][C]) into
+// s1/s2 and ZERO the buckets behind itself (steady state needs no
+// fill launch, same trick as the split-K workspace).
+__global__ void bn_bucket_reduce_k(float* __restrict__ buckets,
+                                   float* __restrict__ s1,
+                                   float* __restrict__ s2, int C,
+                                   int NB) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float a = 0.f, b = 0.f;
+  for (int r = 0; r < NB; ++r) {
+    long i1 = ((long)r * 2) * C + c;
+    long i2 = ((long)r * 2 + 1) * C + c;
+    a += buckets[i1];
+    b += buckets[i2];
+    buckets[i1] = 0.f;
+    buckets[i2] = 0.f;
+  }
+  s1[c] = a;
+  s2[c] = b;
+}
+
 __global__ void bn_finalize_k(const float* __restrict__ s1,
                               const float* __restrict__ s2,
                               float* __restrict__ mean,
@@ -473,9 +495,13 @@ hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
                                float* s2, float* mean, float* invstd,
                                float* rmean, float* rvar,
                                const float* gamma, const float* beta,
+                               float* prestats, int nbuckets,
                                long rows, int C, float momentum,
                                float eps, int relu, hipStream_t stream) {
-  {
+  if (prestats) {
+    hipLaunchKernelGGL(bn_bucket_reduce_k, dim3(cdiv(C, 256)), dim3(256),
+                       0, stream, prestats, s1, s2, C, nbuckets);
+  } else {
     int c8n = C / 8;
     int clw = c8n >= 32 ? 32 : (c8n >= 16 ? 16 : 8);
     int gx = cdiv(c8n, clw);
