@@ -319,7 +319,7 @@ class _HipConvFn(torch.autograd.Function):
         N = x.shape[0]
         M = N * plan.P * plan.Q
         stats = None
-        if layer._collect_stats and layer.training:
+        if getattr(layer, "_collect_stats", False) and layer.training:
             if layer._stats_buf is None:
                 layer._stats_buf = torch.zeros(
                     32 * 2 * layer.plan.Kphys, dtype=torch.float32,
@@ -517,6 +517,10 @@ class Conv2d(_Base):
             x = x[..., :c_log]
         return OF.conv2d_ref(x, self.weight, self.bias, self.stride,
                              self.pad, self.groups, self.relu)
+
+    _collect_stats = False          # class defaults: Linear shares the
+    _stats_buf = None               # conv path but not __init__
+    _stats_valid = False
 
     def pop_stats(self):
         """Bucketed output statistics from the last forward (fused BN),
