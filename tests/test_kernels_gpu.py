@@ -503,11 +503,13 @@ def test_conv_fused_bn_stats():
     from theanompi_amd.models import layers as L
     T.manual_seed(12)
     d = dev()
-    conv = L.Conv2d((14, 14, 64), 96, 3, pad=1, relu=False,
+    # shape must stay off the split-K path (which correctly declines
+    # to collect): M = 8*56*56 = 25088 -> ~196 base blocks
+    conv = L.Conv2d((56, 56, 64), 96, 3, pad=1, relu=False,
                     device=str(d), batch=8, collect_stats=True,
                     has_bias=False)
     conv.train()
-    x = q(T.randn(8, 14, 14, 64)).to(d).to(T.bfloat16)
+    x = q(T.randn(8, 56, 56, 64)).to(d).to(T.bfloat16)
     y = conv(x)
     st = conv.pop_stats()
     assert st is not None
@@ -538,8 +540,8 @@ def test_convbn_fused_stats_equivalence():
         R._BN_FUSE = fuse
         try:
             T.manual_seed(13)
-            m = R._ConvBN(10, 32, 48, 3, pad=1, relu=True, dev=str(d),
-                          B=4)
+            m = R._ConvBN(56, 32, 48, 3, pad=1, relu=True, dev=str(d),
+                          B=8)      # off the split-K path
         finally:
             R._BN_FUSE = old
         m.train()
@@ -547,7 +549,9 @@ def test_convbn_fused_stats_equivalence():
 
     ma, mb = build(True), build(False)
     mb.load_state_dict(ma.state_dict())
-    x = q(T.randn(4, 10, 10, 32)).to(d).to(T.bfloat16)
+    # the fused path must actually engage
+    assert ma.conv._collect_stats
+    x = q(T.randn(8, 56, 56, 32)).to(d).to(T.bfloat16)
 
     def run(m):
         xi = x.clone().requires_grad_()
