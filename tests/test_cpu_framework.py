@@ -456,3 +456,27 @@ def test_recorder_jsonl(tmp_path):
                             "comm_s", "wait_s", "rank"}
         assert row["rank"] == 2
         assert row["calc_s"] >= 0 and row["comm_s"] >= 0
+
+
+def test_heartbeat_disabled_paths():
+    """maybe_start returns None when disabled (no timeout / world 1);
+    Heartbeat.start is a no-op for size < 2."""
+    from theanompi_amd.lib.heartbeat import Heartbeat
+    hb = Heartbeat(_DictStore(), 0, 1, timeout=5).start()
+    assert hb._thread is None          # size 1: nothing to watch
+    hb2 = Heartbeat(_DictStore(), 0, 4, timeout=0).start()
+    assert hb2._thread is None         # timeout 0: disabled
+    hb.stop()
+    hb2.stop()
+
+
+def test_conv_stats_cpu_noop():
+    """collect_stats is a GPU-epilogue feature; on CPU the layer still
+    works and pop_stats reports nothing."""
+    from theanompi_amd.models import layers as L
+    conv = L.Conv2d((8, 8, 3), 16, 3, pad=1, device="cpu", batch=2,
+                    collect_stats=True)
+    x = torch.randn(2, 8, 8, 3)
+    y = conv(x)
+    assert y.shape == (2, 8, 8, 16)
+    assert conv.pop_stats() is None
