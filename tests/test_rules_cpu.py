@@ -185,18 +185,10 @@ def test_heartbeat_aborts_wedged_job(tmp_path):
               exchange_strategy="ar", seed=1, heartbeat=4,
               die_at=3, die_rank=1, die_mode="wedge",
               n_iters=4, val_iters=1)
-    t0 = time.time()
-    deadline = t0 + 120
-    codes = [None, None]
-    while time.time() < deadline:
-        codes = [p.poll() for p in rule.procs]
-        if codes[0] is not None:
-            break
-        time.sleep(0.5)
-    # rank 0 must have aborted via the watchdog, well before any
-    # distributed timeout would fire
+    # rule.wait() itself must not hang on the wedged rank: rank 0
+    # aborts via the watchdog and wait() cleans up the straggler
+    codes = rule.wait(timeout=120)
     assert codes[0] == Heartbeat.EXIT_CODE, codes
-    rule.terminate()
 
 
 def test_gosgd_checkpoint(tmp_path):

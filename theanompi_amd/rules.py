@@ -94,17 +94,31 @@ class _Rule:
         self._launched = True
 
     def wait(self, timeout=None):
-        """Join all workers; return list of exit codes (reference rule.wait())."""
-        codes = []
+        """Join all workers; return list of exit codes (reference
+        rule.wait()).  A rank that dies while others are wedged cannot
+        hang the join: once any rank exits non-zero the rest get a
+        grace window (transport errors / the heartbeat watchdog
+        normally end them) and are then terminated."""
+        deadline = time.time() + timeout if timeout else None
         try:
-            for p in self.procs:
-                codes.append(p.wait(timeout=timeout))
+            while True:
+                codes = [p.poll() for p in self.procs]
+                if all(c is not None for c in codes):
+                    return codes
+                if any(c is not None and c != 0 for c in codes):
+                    grace = time.time() + 15
+                    while time.time() < grace and any(
+                            p.poll() is None for p in self.procs):
+                        time.sleep(0.2)
+                    self.terminate()
+                    return [p.poll() for p in self.procs]
+                if deadline is not None and time.time() > deadline:
+                    raise subprocess.TimeoutExpired(
+                        "worker", timeout)
+                time.sleep(0.2)
         except Exception:
             self.terminate()
             raise
-        if any(c != 0 for c in codes):
-            self.terminate()  # kill stragglers if one rank died
-        return codes
 
     def terminate(self):
         for p in self.procs:
