@@ -480,3 +480,26 @@ def test_conv_stats_cpu_noop():
     y = conv(x)
     assert y.shape == (2, 8, 8, 16)
     assert conv.pop_stats() is None
+
+
+def test_bench_contract_cpu(tmp_path):
+    """bench.py (the driver contract) runs on CPU and prints one valid
+    JSON line with the required fields."""
+    import json
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "bench.py", "--model", "cifar10", "--batch",
+         "8", "--steps", "2", "--warmup", "1"],
+        capture_output=True, text=True, timeout=300,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    d = json.loads(line)
+    for k in ("metric", "value", "unit", "n_gpus", "steps", "warmup",
+              "ms_per_step", "higher_is_better", "scaling",
+              "vs_baseline", "dtype", "data", "config"):
+        assert k in d, k
+    assert d["n_gpus"] == 1 and d["steps"] == 2
+    assert d["data"] == "synthetic"
+    assert d["config"]["global_batch"] == 8
