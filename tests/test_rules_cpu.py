@@ -206,3 +206,28 @@ def test_gosgd_checkpoint(tmp_path):
     files = sorted(os.listdir(tmp_path))
     assert any(f.startswith("W_0_") and f.endswith(".npy")
                for f in files), files
+
+
+def test_bench_multirank_cpu(tmp_path):
+    """The bench contract under torch.distributed.run at world size 2
+    (the driver's multi-GPU invocation shape), CPU/gloo."""
+    import json
+    import subprocess
+    import sys
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29771", "bench.py", "--gpus", "2",
+         "--model", "cifar10", "--batch", "8", "--steps", "2",
+         "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, cwd=repo)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [ln for ln in out.stdout.splitlines()
+             if ln.startswith("{")]
+    assert len(lines) == 1, lines          # ONE line, from rank 0
+    d = json.loads(lines[0])
+    assert d["n_gpus"] == 2
+    assert d["config"]["parallelism"] == "dp2"
+    assert d["config"]["global_batch"] == 16
