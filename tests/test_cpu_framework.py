@@ -246,6 +246,72 @@ def test_imagenet_shard_loader(tmp_path):
         data.close()
 
 
+def test_imagenet_sub_batches_per_shard(tmp_path):
+    """Reference file_batch_size semantics: a shard
+    bigger than batch_size feeds multiple iterations — no images are
+    silently dropped, and labels line up with their sub-batch."""
+    from theanompi_amd.models.data.imagenet import (ImageNetData,
+                                                    write_synthetic_shards)
+    paths = write_synthetic_shards(str(tmp_path), n_shards=2,
+                                   images_per_shard=16, hw=40)
+    d = ImageNetData(paths, paths[:1], batch_size=4, crop=40,
+                     n_classes=1000, device="cpu", shuffle=False)
+    try:
+        assert d.sub_batches == 4
+        assert d.n_batch_train == 8          # 2 shards x 4 sub-batches
+        assert d.n_batch_val == 4
+        seen = []
+        for i in range(d.n_batch_train):
+            x, y = d.get_train_batch(i)
+            assert x.shape == (4, 40, 40, 3)
+            seen.append(y.clone())
+        labels = torch.cat(seen)
+        # every training image's label was served exactly once
+        import numpy as np
+        want = np.concatenate(
+            [np.load(p.replace(".npy", "_labels.npy"))
+             for p in d.my_shards])
+        assert sorted(labels.tolist()) == sorted(want.tolist())
+        xv, yv = d.get_val_batch(3)          # last val sub-batch
+        assert xv.shape == (4, 40, 40, 3)
+    finally:
+        d.close()
+
+
+def test_imagenet_underfilled_shard_tail_zeroed(tmp_path):
+    """A shard smaller than the slot leaves no stale images/labels."""
+    import numpy as np
+    from theanompi_amd.models.data.imagenet import (ImageNetData,
+                                                    write_synthetic_shards)
+    paths = write_synthetic_shards(str(tmp_path), n_shards=1,
+                                   images_per_shard=8, hw=40)
+    # a second shard with only 5 images (tail of 3 must be zeroed)
+    small = str(tmp_path / "small.npy")
+    np.save(small, np.random.RandomState(7).randint(
+        0, 256, (5, 40, 40, 3), dtype=np.uint8))
+    np.save(small.replace(".npy", "_labels.npy"),
+            np.arange(5) + 100)
+    d = ImageNetData([paths[0], small], [paths[0]], batch_size=8,
+                     crop=32, n_classes=1000, device="cpu", shuffle=False)
+    try:
+        x0, y0 = d.get_train_batch(0)
+        x1, y1 = d.get_train_batch(1)        # the underfilled shard
+        assert (y1[:5] >= 100).all()
+        assert (y1[5:] == 0).all()           # not shard 0's labels
+    finally:
+        d.close()
+
+
+def test_avg_mode_defaults_to_fp32_wire():
+    """'avg' ships fp32 master weights — bf16 wire would
+    erase small updates.  The worker must default avg to rccl32."""
+    from theanompi_amd.bsp_worker import default_strategy
+    assert default_strategy("avg", True) == "rccl32"
+    assert default_strategy("cdd", True) == "rccl_bf16"
+    assert default_strategy("grad", True) == "rccl_bf16"
+    assert default_strategy("avg", False) == "ar"
+
+
 def test_cifar10_data_object():
     from theanompi_amd.models.data.cifar10_data import Cifar10Data
     d = Cifar10Data(batch_size=16, device="cpu", synthetic_train=64,

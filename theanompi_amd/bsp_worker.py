@@ -19,6 +19,16 @@ from theanompi_amd.lib.recorder import Recorder
 from theanompi_amd.models.base_model import _rtx
 
 
+def default_strategy(sync_type, on_gpu):
+    """bf16 wire is safe for deltas (cdd velocities / grads) but not for
+    'avg', which ships the fp32 master weights themselves: quantizing
+    those to bf16 every iteration erases updates below the weights' bf16
+    ulp.  Weight averaging therefore defaults to an fp32 wire."""
+    if not on_gpu:
+        return "ar"
+    return "rccl32" if sync_type == "avg" else "rccl_bf16"
+
+
 def run():
     proc = GPUProcess()
     device = proc.init_device()
@@ -28,7 +38,7 @@ def run():
 
     sync_type = _env_opt("sync_type", "cdd")
     strategy = _env_opt("exchange_strategy",
-                        "rccl_bf16" if proc.on_gpu else "ar")
+                        default_strategy(sync_type, proc.on_gpu))
     n_epochs = int(_env_opt("epochs", model.n_epochs))
     save_dir = _env_opt("save_dir", "")
     save_freq = int(_env_opt("save_freq", 0))

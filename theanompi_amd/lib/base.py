@@ -61,12 +61,21 @@ class GPUProcess:
         if not dist.is_initialized():
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "23456")
-            dist.init_process_group(
-                backend=backend,
-                rank=self.rank,
-                world_size=self.size,
-                timeout=datetime.timedelta(seconds=timeout_s),
-            )
+            try:
+                dist.init_process_group(
+                    backend=backend,
+                    rank=self.rank,
+                    world_size=self.size,
+                    timeout=datetime.timedelta(seconds=timeout_s),
+                )
+            except Exception as e:  # rendezvous failed (e.g. the port
+                # picked by the rule launcher was grabbed by a concurrent
+                # launch between probe and bind).  Exit code 17 tells
+                # rule.wait() to relaunch the job on a fresh port.
+                import sys
+                print(f"[theanompi_amd rank {self.rank}] rendezvous "
+                      f"failed: {e}", file=sys.stderr)
+                os._exit(17)
         self.comm = dist.group.WORLD
         # control plane: gloo works for CPU tensors regardless of data plane
         if backend != "gloo":

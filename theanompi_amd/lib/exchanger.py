@@ -184,7 +184,11 @@ class OverlappedGradExchanger:
         for r in st["recs"]:
             off = st["offs"][id(r.master)]
             n = r.master.numel()
-            wire[off:off + n].copy_(r.master.grad.view(-1))
+            g = r.master.grad
+            if g is None:       # param got no grad this step (unused head)
+                wire[off:off + n].zero_()
+            else:
+                wire[off:off + n].copy_(g.view(-1))
         st["work"] = dist.all_reduce(wire, group=self.group,
                                      async_op=True)
 
@@ -206,6 +210,9 @@ class OverlappedGradExchanger:
             for r in st["recs"]:
                 off = st["offs"][id(r.master)]
                 n = r.master.numel()
+                if r.master.grad is None:
+                    # other ranks may still have contributed a gradient
+                    r.master.grad = torch.zeros_like(r.master)
                 r.master.grad.view(-1).copy_(st["wire"][off:off + n])
             st["ready"] = 0
             st["work"] = None

@@ -33,7 +33,15 @@ import torch.multiprocessing as tmp
 
 def _loader_child(task_q, slot_q, done_q, slots, lslots, crop, train):
     """Child process: load shard -> augment -> write into a pinned slot.
-    Mirrors proc_load_mpi.py's recv(filename) / Send(batch) loop."""
+    Mirrors proc_load_mpi.py's recv(filename) / Send(batch) loop.
+
+    The whole usable shard (slot capacity = sub_batches*batch_size
+    images) is written per task; the parent slices per-iteration
+    sub-batches out of it (reference file_batch_size semantics: a
+    256-image file feeds 2 iterations at batch 128).
+    Shards smaller than the slot get their slot tail zeroed so no
+    stale images/labels from a previous shard survive.
+    """
     rng = np.random.RandomState(os.getpid())
     while True:
         task = task_q.get()
@@ -58,6 +66,9 @@ def _loader_child(task_q, slot_q, done_q, slots, lslots, crop, train):
         m = min(n, dst.shape[0])
         dst[:m, :, :, :3] = torch.from_numpy(np.ascontiguousarray(view[:m]))
         dst[:m, :, :, 3:] = 0
+        if m < dst.shape[0]:                     # underfilled shard
+            dst[m:] = 0
+            lslots[slot][m:] = 0
         lslots[slot][:m] = torch.from_numpy(labels[:m].astype(np.int64))
         done_q.put(slot)
 
@@ -79,6 +90,15 @@ class ImageNetData:
         self.shuffle = shuffle
         self.c_phys = c_phys
         self.epoch = 0
+        # reference file_batch_size semantics: each
+        # shard feeds images_per_shard//batch_size training iterations;
+        # a 256-image file is 2 sub-batches at batch 128.  The tail
+        # (n % batch_size images) is dropped, as the reference's
+        # file-granular loader does.
+        n0 = int(np.load(self.train_shards[0], mmap_mode="r").shape[0]) \
+            if self.train_shards else batch_size
+        self.sub_batches = max(1, n0 // batch_size)
+        self.slot_images = self.sub_batches * batch_size
         # optional dataset mean image ([crop, crop, 3] fp32 .npy, the
         # reference's img_mean): subtracted on-GPU inside u8_finish so
         # the host path still ships uint8.
@@ -94,11 +114,11 @@ class ImageNetData:
                 self.mean_dev = self.mean_host.to(self.device)
         self._partition()
 
-        # pinned ring shared with the child process
+        # pinned ring shared with the child process (slot = whole shard)
         ctx = tmp.get_context("spawn")
-        self.slots = [torch.empty((batch_size, crop, crop, c_phys),
+        self.slots = [torch.empty((self.slot_images, crop, crop, c_phys),
                                   dtype=torch.uint8) for _ in range(depth)]
-        self.lslots = [torch.empty((batch_size,), dtype=torch.int64)
+        self.lslots = [torch.empty((self.slot_images,), dtype=torch.int64)
                        for _ in range(depth)]
         # shared (not pinned): pinned host memory cannot be shared across
         # processes portably; the h2d copy below still overlaps compute
@@ -135,71 +155,94 @@ class ImageNetData:
             rng = np.random.RandomState(1000 + self.epoch)
             rng.shuffle(shards)
         self.my_shards = shards[self.rank::self.size]
-        self.n_batch_train = len(self.my_shards)
-        self.n_batch_val = len(self.val_shards)
+        # n_batch_* count SUB-batches (training iterations), not shards
+        self.n_batch_train = len(self.my_shards) * self.sub_batches
+        self.n_batch_val = len(self.val_shards) * self.sub_batches
 
     # -- contract -------------------------------------------------------------
 
     def reset(self, mode):
         if mode == "train":
             self.epoch += 1
+            # drain shards prefetched for the previous epoch so slots
+            # are free and ordering restarts cleanly
+            while self._inflight:
+                self.done_q.get()
+                self._inflight.pop()
+            self._cur_slot = None
             self._partition()
 
     def _request(self, path, slot):
         self.task_q.put((path, slot))
         self._inflight.append(slot)
 
+    _cur_slot = None          # slot currently being consumed sub-batch-wise
+    _cur_x = None             # device-finished full-shard fp tensor (GPU)
+    _cur_y = None
+
     def get_train_batch(self, i):
-        # prime the pipeline depth
-        while len(self._inflight) < self.depth and i + len(
-                self._inflight) < self.n_batch_train:
-            s = self._next_slot
-            self._next_slot = (s + 1) % self.depth
-            self._request(self.my_shards[(i + len(self._inflight))
-                                         % self.n_batch_train], s)
-        slot = self.done_q.get()                 # child finished a shard
-        self._inflight.remove(slot)
-        x_u8, y = self.slots[slot], self.lslots[slot]
-        if self.gpu:
-            compute = torch.cuda.current_stream()
-            with torch.cuda.stream(self.copy_stream):
-                if self._slot_used[slot]:
-                    # don't overwrite a device slot the compute stream
-                    # may still be reading
-                    self.copy_stream.wait_event(self.done_events[slot])
-                    if self._debug:       # §5.2 race-detection mode
-                        self.done_events[slot].synchronize()
-                self.dev_bufs[slot].copy_(x_u8, non_blocking=True)
-                self.dev_lbufs[slot].copy_(y, non_blocking=True)
-                self.copy_events[slot].record(self.copy_stream)
-            compute.wait_event(self.copy_events[slot])
-            from theanompi_amd import ops as _ops
-            x = _ops.ext().u8_finish(self.dev_bufs[slot], 3,
-                                     self.mean_dev)
-            yd = self.dev_lbufs[slot].clone()
-            self.done_events[slot].record(compute)
-            self._slot_used[slot] = True
-            return x, yd
-        if self.mean_host is not None:
-            x = (x_u8.float() - self.mean_host) * (1 / 128.0)
-        else:
-            x = x_u8.float() * (1 / 128.0) - 1.0
-        x[..., 3:] = 0
-        return x[..., :3] if self.c_phys == 4 and not self.gpu else x, \
-            y.clone()
+        shard_i, sub = divmod(i, self.sub_batches)
+        bs = self.batch_size
+        if sub == 0 or self._cur_slot is None:
+            # need the next shard: prime the pipeline depth ahead
+            n_shards = len(self.my_shards)
+            while len(self._inflight) < self.depth and shard_i + len(
+                    self._inflight) < n_shards:
+                s = self._next_slot
+                self._next_slot = (s + 1) % self.depth
+                self._request(self.my_shards[(shard_i + len(self._inflight))
+                                             % n_shards], s)
+            slot = self.done_q.get()             # child finished a shard
+            self._inflight.remove(slot)
+            self._cur_slot = slot
+            x_u8, y = self.slots[slot], self.lslots[slot]
+            if self.gpu:
+                compute = torch.cuda.current_stream()
+                with torch.cuda.stream(self.copy_stream):
+                    if self._slot_used[slot]:
+                        # don't overwrite a device slot the compute
+                        # stream may still be reading
+                        self.copy_stream.wait_event(self.done_events[slot])
+                        if self._debug:   # §5.2 race-detection mode
+                            self.done_events[slot].synchronize()
+                    self.dev_bufs[slot].copy_(x_u8, non_blocking=True)
+                    self.dev_lbufs[slot].copy_(y, non_blocking=True)
+                    self.copy_events[slot].record(self.copy_stream)
+                compute.wait_event(self.copy_events[slot])
+                from theanompi_amd import ops as _ops
+                self._cur_x = _ops.ext().u8_finish(self.dev_bufs[slot], 3,
+                                                   self.mean_dev)
+                self._cur_y = self.dev_lbufs[slot].clone()
+                self.done_events[slot].record(compute)
+                self._slot_used[slot] = True
+            else:
+                if self.mean_host is not None:
+                    xf = (x_u8.float() - self.mean_host) * (1 / 128.0)
+                else:
+                    xf = x_u8.float() * (1 / 128.0) - 1.0
+                xf[..., 3:] = 0
+                self._cur_x = xf[..., :3] if self.c_phys == 4 else xf
+                self._cur_y = y.clone()
+        return self._cur_x[sub * bs:(sub + 1) * bs], \
+            self._cur_y[sub * bs:(sub + 1) * bs]
 
     def get_val_batch(self, i):
-        path = self.val_shards[i % len(self.val_shards)]
+        shard_i, sub = divmod(i, self.sub_batches)
+        bs = self.batch_size
+        path = self.val_shards[shard_i % len(self.val_shards)]
         arr = np.load(path)
         labels = np.load(path.replace(".npy", "_labels.npy"))
-        H = arr.shape[1]
+        n, H = arr.shape[0], arr.shape[1]
         o = (H - self.crop) // 2
+        lo, hi = sub * bs, min((sub + 1) * bs, n)
         view = np.ascontiguousarray(
-            arr[:self.batch_size, o:o + self.crop, o:o + self.crop])
-        x = torch.from_numpy(view)
-        pad = torch.zeros((*x.shape[:3], self.c_phys), dtype=torch.uint8)
-        pad[..., :3] = x
-        y = torch.from_numpy(labels[:self.batch_size].astype(np.int64))
+            arr[lo:hi, o:o + self.crop, o:o + self.crop])
+        m = view.shape[0]
+        pad = torch.zeros((bs, self.crop, self.crop, self.c_phys),
+                          dtype=torch.uint8)
+        pad[:m, ..., :3] = torch.from_numpy(view)
+        y = torch.zeros((bs,), dtype=torch.int64)
+        y[:m] = torch.from_numpy(labels[lo:hi].astype(np.int64))
         if self.gpu:
             from theanompi_amd import ops as _ops
             return _ops.ext().u8_finish(pad.to(self.device), 3,

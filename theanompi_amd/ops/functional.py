@@ -306,6 +306,10 @@ class _SoftmaxXentFn(torch.autograd.Function):
         C.softmax_xent_fwd(logits, labels, probs, loss, int(n_classes))
         ctx.save_for_backward(probs, labels)
         ctx.n_classes = n_classes
+        # probs is an inference-only output: backward below only
+        # accounts for dloss, so autograd must not route gradients
+        # through probs silently
+        ctx.mark_non_differentiable(probs)
         return loss.mean(), probs
 
     @staticmethod

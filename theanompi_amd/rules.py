@@ -51,6 +51,8 @@ class _Rule:
     def __init__(self):
         self.procs = []
         self._launched = False
+        self._relaunch = None
+        self._retries_left = 0
 
     # -- reference-parity API -------------------------------------------------
 
@@ -69,8 +71,20 @@ class _Rule:
         if n < 1:
             raise ValueError("need at least one device")
         parsed = [_parse_device(d) for d in devices]
-        port = kwargs.pop("master_port", None) or _free_port()
+        fixed_port = kwargs.pop("master_port", None)
+        port = fixed_port or _free_port()
+        # _free_port is inherently TOCTOU (the socket is closed before
+        # rank 0 binds the rendezvous store); a worker losing the race
+        # exits 17 and wait() relaunches everything on a fresh port.
+        self._relaunch = (None if fixed_port
+                          else (parsed, modelfile, modelclass, dict(kwargs)))
+        self._retries_left = 3
 
+        self._spawn(parsed, modelfile, modelclass, kwargs, port)
+        self._launched = True
+
+    def _spawn(self, parsed, modelfile, modelclass, kwargs, port):
+        n = len(parsed)
         env_common = dict(os.environ)
         env_common.update({
             "MASTER_ADDR": "127.0.0.1",
@@ -83,6 +97,7 @@ class _Rule:
         for k, v in kwargs.items():
             env_common[f"THEANOMPI_AMD_OPT_{k.upper()}"] = str(v)
 
+        self.procs = []
         for rank, (kind, idx) in enumerate(parsed):
             env = dict(env_common)
             env["RANK"] = str(rank)
@@ -91,7 +106,18 @@ class _Rule:
             mod = self._module_for_rank(rank, n)
             cmd = [sys.executable, "-u", "-m", mod]
             self.procs.append(subprocess.Popen(cmd, env=env))
-        self._launched = True
+
+    def _maybe_relaunch(self, codes):
+        """Exit code 17 = lost the rendezvous-port race: relaunch the
+        whole job on a fresh port (bounded retries)."""
+        if 17 not in codes or self._relaunch is None \
+                or self._retries_left <= 0 or any(c == 0 for c in codes):
+            return False
+        self._retries_left -= 1
+        self.terminate()
+        parsed, modelfile, modelclass, kwargs = self._relaunch
+        self._spawn(parsed, modelfile, modelclass, kwargs, _free_port())
+        return True
 
     def wait(self, timeout=None):
         """Join all workers; return list of exit codes (reference
@@ -104,6 +130,8 @@ class _Rule:
             while True:
                 codes = [p.poll() for p in self.procs]
                 if all(c is not None for c in codes):
+                    if self._maybe_relaunch(codes):
+                        continue
                     return codes
                 if any(c is not None and c != 0 for c in codes):
                     grace = time.time() + 15
@@ -111,7 +139,10 @@ class _Rule:
                             p.poll() is None for p in self.procs):
                         time.sleep(0.2)
                     self.terminate()
-                    return [p.poll() for p in self.procs]
+                    codes = [p.poll() for p in self.procs]
+                    if self._maybe_relaunch(codes):
+                        continue
+                    return codes
                 if deadline is not None and time.time() > deadline:
                     raise subprocess.TimeoutExpired(
                         "worker", timeout)
