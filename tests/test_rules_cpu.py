@@ -208,6 +208,45 @@ def test_gosgd_checkpoint(tmp_path):
                for f in files), files
 
 
+def test_resume_bit_equal(tmp_path):
+    """resume is true SGD-state resume.  Train 4 epochs
+    uninterrupted vs 2 epochs + checkpoint + resume(2) + 2 epochs: the
+    final parameters must be bit-equal (weights + momenta + LR schedule
+    all restored)."""
+    full = tmp_path / "full"
+    ck = tmp_path / "ck"
+    resumed = tmp_path / "resumed"
+    for d in (full, ck, resumed):
+        d.mkdir()
+    common = ("modelfile='theanompi_amd.models.test_model', "
+              "modelclass='TestModel', sync_type='cdd', "
+              "exchange_strategy='ar', seed=99")
+    body = f"""
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import BSP
+    # run A: 4 uninterrupted epochs
+    r = BSP(); r.init(devices=['cpu', 'cpu'], {common},
+                      epochs=4, dump_dir={str(full)!r})
+    assert r.wait() == [0, 0]
+    # run B1: 2 epochs, checkpoint at epoch 2
+    r = BSP(); r.init(devices=['cpu', 'cpu'], {common},
+                      epochs=2, save_dir={str(ck)!r}, save_freq=2)
+    assert r.wait() == [0, 0]
+    # run B2: resume from epoch 2, train to 4
+    r = BSP(); r.init(devices=['cpu', 'cpu'], {common},
+                      epochs=4, resume_dir={str(ck)!r}, resume_epoch=2,
+                      dump_dir={str(resumed)!r})
+    assert r.wait() == [0, 0]
+    """
+    _run_session(tmp_path, body, timeout=480)
+    a = np.load(os.path.join(str(full), "params_rank0.npz"))
+    b = np.load(os.path.join(str(resumed), "params_rank0.npz"))
+    assert set(a.files) == set(b.files)
+    for k in a.files:
+        np.testing.assert_array_equal(a[k], b[k])
+
+
 def test_bench_multirank_cpu(tmp_path):
     """The bench contract under torch.distributed.run at world size 2
     (the driver's multi-GPU invocation shape), CPU/gloo."""

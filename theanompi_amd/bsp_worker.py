@@ -12,7 +12,7 @@ import torch
 import torch.distributed as dist
 
 from theanompi_amd.lib.heartbeat import maybe_start
-from theanompi_amd.lib.base import GPUProcess, _env_opt
+from theanompi_amd.lib.base import GPUProcess, _env_opt, maybe_resume
 from theanompi_amd.lib.exchanger import (BSP_Exchanger,
                                          OverlappedGradExchanger)
 from theanompi_amd.lib.recorder import Recorder
@@ -61,6 +61,7 @@ def run():
                     "print_freq": int(_env_opt("print_freq", 40)),
                     "sync_fn": (torch.cuda.synchronize if proc.on_gpu
                                 else None)})
+    start_epoch = maybe_resume(model, rec)
     proc.barrier()
 
     # test hooks: simulate a failing rank.  die_mode "exit" dies
@@ -70,9 +71,9 @@ def run():
     die_at = int(_env_opt("die_at", -1))
     die_rank = int(_env_opt("die_rank", 0))
     die_mode = _env_opt("die_mode", "exit")
-    for epoch in range(n_epochs):
+    for epoch in range(start_epoch, n_epochs):
         rec.start_epoch()
-        if epoch > 0:
+        if epoch > start_epoch:
             model.reset_iter("train")   # per-epoch shard reshuffle
         n_batches = model.data.n_batch_train
         for i in range(n_batches):

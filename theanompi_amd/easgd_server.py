@@ -14,7 +14,7 @@ import time
 import torch
 
 from theanompi_amd.lib.heartbeat import maybe_start
-from theanompi_amd.lib.base import GPUProcess, _env_opt
+from theanompi_amd.lib.base import GPUProcess, _env_opt, maybe_resume
 from theanompi_amd.lib.exchanger import EASGD_Exchanger
 from theanompi_amd.lib.recorder import Recorder
 
@@ -34,10 +34,17 @@ def run():
     val_freq = int(_env_opt("val_freq", model.data.n_batch_train))
     save_dir = _env_opt("save_dir", "")
 
+    # resume BEFORE the broadcast so workers start from the restored
+    # center variables
+    start_epoch = maybe_resume(model)
     proc.broadcast_params(model)
     exch = EASGD_Exchanger({"alpha": alpha}, model.params, role="server")
     rec = Recorder({"rank": proc.rank, "size": proc.size,
                     "verbose": proc.verbose})
+    if start_epoch:
+        ip = os.path.join(_env_opt("resume_dir", ""), "inforec.pkl")
+        if os.path.exists(ip):
+            rec.load(ip)
     store = proc.control_store()
     proc.barrier()
 
@@ -46,7 +53,7 @@ def run():
 
     uidx = 0
     last_val = 0
-    epoch = 0
+    epoch = start_epoch
     t0 = time.time()
     while workers:
         progressed = False

@@ -18,7 +18,7 @@ import torch
 import torch.distributed as dist
 
 from theanompi_amd.lib.heartbeat import maybe_start
-from theanompi_amd.lib.base import GPUProcess, _env_opt
+from theanompi_amd.lib.base import GPUProcess, _env_opt, maybe_resume
 from theanompi_amd.lib.exchanger import GOSGD_Exchanger
 from theanompi_amd.lib.recorder import Recorder
 
@@ -48,6 +48,7 @@ def run():
     send_seq = {dst: 0 for dst in peers}    # pushes I sent to dst
     recv_seq = {src: 0 for src in peers}    # pushes I consumed from src
     me = proc.rank
+    start_epoch = maybe_resume(model, rec)
     proc.barrier()
 
     def drain():
@@ -64,7 +65,7 @@ def run():
             model.refresh_all()
 
     uidx = 0
-    for epoch in range(n_epochs):
+    for epoch in range(start_epoch, n_epochs):
         rec.start_epoch()
         for i in range(model.data.n_batch_train):
             rec.start()

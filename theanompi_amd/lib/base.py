@@ -23,6 +23,31 @@ def _env_opt(name, default=None):
     return os.environ.get(f"THEANOMPI_AMD_OPT_{name.upper()}", default)
 
 
+def maybe_resume(model, rec=None):
+    """True SGD-state resume: ``rule.init(...,
+    resume_dir=..., resume_epoch=N)`` restores per-param
+    ``W_<i>_<N>.npy`` weights + ``V_<i>_<N>.npy`` momenta, the pickled
+    recorder history, and the LR schedule (by re-running
+    ``adjust_hyperp(N)``).  Returns the epoch to start from (0 when not
+    resuming)."""
+    d = _env_opt("resume_dir", "")
+    if not d:
+        return 0
+    epoch = int(_env_opt("resume_epoch", "0"))
+    model.load(d, epoch)
+    model.adjust_hyperp(epoch)
+    ip = os.path.join(d, "inforec.pkl")
+    if rec is not None and os.path.exists(ip):
+        rec.load(ip)
+    # fast-forward per-epoch data state (shard reshuffle counters) so a
+    # resumed run sees the same per-epoch shard order as an
+    # uninterrupted one
+    if hasattr(model, "reset_iter"):
+        for _ in range(epoch):
+            model.reset_iter("train")
+    return epoch
+
+
 class GPUProcess:
     """Rank/device binding + communicators for one worker process."""
 
