@@ -208,6 +208,46 @@ def test_gosgd_checkpoint(tmp_path):
                for f in files), files
 
 
+def test_foreign_module_under_bsp(tmp_path):
+    """Reference parity (SURVEY §2.4 Wide-ResNet row): the contract is
+    framework-agnostic.  A CNN written with stock torch.nn — no
+    BaseModel inheritance, no framework layers — trains under BSP via
+    the ContractAdapter, and ranks end bit-identical."""
+    out = str(tmp_path)
+    body = f"""
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import BSP
+    rule = BSP()
+    rule.init(devices=['cpu', 'cpu'],
+              modelfile='theanompi_amd.models.module_adapter',
+              modelclass='PlainTorchCNN',
+              epochs=1, sync_type='cdd', exchange_strategy='ar',
+              seed=31, dump_dir={out!r})
+    assert rule.wait() == [0, 0]
+    """
+    _run_session(tmp_path, body)
+    a = np.load(os.path.join(out, "params_rank0.npz"))
+    b = np.load(os.path.join(out, "params_rank1.npz"))
+    for k in a.files:
+        np.testing.assert_allclose(a[k], b[k], atol=1e-6)
+
+
+def test_foreign_module_under_gosgd(tmp_path):
+    body = """
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import GOSGD
+    rule = GOSGD()
+    rule.init(devices=['cpu', 'cpu'],
+              modelfile='theanompi_amd.models.module_adapter',
+              modelclass='PlainTorchCNN',
+              epochs=1, p_push=0.5, seed=32)
+    assert rule.wait() == [0, 0]
+    """
+    _run_session(tmp_path, body)
+
+
 def test_resume_bit_equal(tmp_path):
     """resume is true SGD-state resume.  Train 4 epochs
     uninterrupted vs 2 epochs + checkpoint + resume(2) + 2 epochs: the
