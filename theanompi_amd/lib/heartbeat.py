@@ -111,7 +111,12 @@ def maybe_start(proc):
         if hb: hb.stop()
     """
     from theanompi_amd.lib.base import _env_opt
-    timeout = float(_env_opt("heartbeat", 0) or 0)
+    # default ON (120 s): a wedged rank aborts the job instead of
+    # hanging it forever, with no kwarg needed (SURVEY §5.3).  Disable
+    # with heartbeat=0; THEANOMPI_HEARTBEAT_DEFAULT overrides the
+    # default for tests.
+    default = os.environ.get("THEANOMPI_HEARTBEAT_DEFAULT", "120")
+    timeout = float(_env_opt("heartbeat", default) or 0)
     if timeout <= 0 or proc.size < 2:
         return None
     store = proc.control_store()
