@@ -335,7 +335,8 @@ def test_exchanger_per_tensor_mode():
 def test_strategy_aliases():
     from theanompi_amd.lib.exchanger_strategy import make_strategy
     for name in ("ar", "nccl32", "rccl32", "nccl16", "rccl_bf16",
-                 "asa", "asa32", "asa16", "asa_bf16"):
+                 "asa", "asa32", "asa16", "asa_bf16", "asa_xgmi",
+                 "asa_xgmi_bf16"):
         assert make_strategy(name) is not None
     import pytest as _pt
     with _pt.raises(ValueError):

@@ -60,6 +60,25 @@ def test_relu_add_pack():
     assert torch.equal(back, dst.float())
 
 
+def test_segsum():
+    """N14 segment-sum (asa_xgmi phase 2) vs torch fp32 reference."""
+    from theanompi_amd import ops as O
+    C = O.ext()
+    for w in (2, 7, 8):
+        seg = 4 * 1000
+        x = torch.randn(w * seg, device=dev())
+        out = torch.empty(seg, device=dev())
+        C.segsum(x, out, w)
+        ref = x.view(w, seg).sum(0)
+        assert torch.allclose(out, ref, atol=1e-5), (w, seg)
+        xb = x.to(torch.bfloat16)
+        ob = torch.empty(seg, dtype=torch.bfloat16, device=dev())
+        C.segsum(xb, ob, w)
+        # fp32 accumulation: must match the fp32 sum of the bf16 inputs
+        refb = xb.float().view(w, seg).sum(0).to(torch.bfloat16)
+        assert torch.equal(ob, refb), w
+
+
 def test_halo_pad():
     from theanompi_amd import ops as O
     C = O.ext()

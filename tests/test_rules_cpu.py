@@ -56,7 +56,7 @@ def test_bsp_two_workers_cpu(tmp_path):
         np.testing.assert_allclose(a[k], b[k], atol=1e-6)
 
 
-@pytest.mark.parametrize("strategy", ["ar", "asa"])
+@pytest.mark.parametrize("strategy", ["ar", "asa", "asa_xgmi"])
 def test_bsp_strategies_cpu(tmp_path, strategy):
     body = f"""
     import os
@@ -147,7 +147,7 @@ def test_strategy_equivalence(tmp_path):
     """'ar' and 'asa' produce identical results (same seeded init/data),
     2 CPU workers."""
     outs = {}
-    for strat in ("ar", "asa"):
+    for strat in ("ar", "asa", "asa_xgmi"):
         d = tmp_path / strat
         d.mkdir()
         body = f"""
@@ -166,6 +166,8 @@ def test_strategy_equivalence(tmp_path):
         outs[strat] = np.load(os.path.join(str(d), "params_rank0.npz"))
     for k in outs["ar"].files:
         np.testing.assert_allclose(outs["ar"][k], outs["asa"][k],
+                                   atol=2e-5)
+        np.testing.assert_allclose(outs["ar"][k], outs["asa_xgmi"][k],
                                    atol=2e-5)
 
 
@@ -264,6 +266,24 @@ def test_foreign_module_under_gosgd(tmp_path):
     assert rule.wait() == [0, 0]
     """
     _run_session(tmp_path, body)
+
+
+def test_gosgd_zoo_scale_vgg16(tmp_path):
+    """GOSGD at zoo scale: VGG-16's full 138 M
+    params travel per push; 2 ranks, every iteration pushes."""
+    body = """
+    import os
+    os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
+    from theanompi_amd import GOSGD
+    rule = GOSGD()
+    rule.init(devices=['cpu', 'cpu'],
+              modelfile='theanompi_amd.models.vgg16',
+              modelclass='VGG16',
+              epochs=1, p_push=1.0, seed=17,
+              batch_size=2, n_iters=2, val_iters=1)
+    assert rule.wait() == [0, 0]
+    """
+    _run_session(tmp_path, body, timeout=600)
 
 
 def test_resume_bit_equal(tmp_path):

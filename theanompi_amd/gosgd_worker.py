@@ -36,7 +36,18 @@ def run():
     rng = random.Random(1234 + proc.rank)
 
     proc.broadcast_params(model)
-    exch = GOSGD_Exchanger({}, model.params, group=proc.ctrl)
+    # GPU: one directed RCCL communicator per ordered rank pair so a
+    # blocked A->B send can never head-of-line block the B->A direction
+    # (new_group is collective: every rank iterates the same order)
+    pair_groups = None
+    if proc.on_gpu and proc.size > 1:
+        pair_groups = {}
+        for s in range(proc.size):
+            for d in range(proc.size):
+                if s != d:
+                    pair_groups[(s, d)] = dist.new_group([s, d])
+    exch = GOSGD_Exchanger({}, model.params, group=proc.ctrl,
+                           pair_groups=pair_groups, rank=proc.rank)
     rec = Recorder({"rank": proc.rank, "size": proc.size,
                     "verbose": proc.verbose,
                     "print_freq": int(_env_opt("print_freq", 40)),
@@ -44,7 +55,6 @@ def run():
                                 else None)})
     store = proc.control_store()
     peers = [r for r in range(proc.size) if r != proc.rank]
-    rbufs = {src: exch.recv_buffer() for src in peers}
     send_seq = {dst: 0 for dst in peers}    # pushes I sent to dst
     recv_seq = {src: 0 for src in peers}    # pushes I consumed from src
     me = proc.rank
@@ -55,10 +65,11 @@ def run():
         """Receive + merge every announced, not-yet-consumed push."""
         merged = False
         for src in peers:
-            while store.check([f"go/push/{me}/{src}/{recv_seq[src]}"]):
-                dist.recv(rbufs[src], src=src, group=proc.ctrl,
-                          tag=400 + src)
-                exch.try_merge_buf(rbufs[src])
+            while True:
+                key = f"go/push/{me}/{src}/{recv_seq[src]}"
+                if not store.check([key]):
+                    break
+                exch.recv_merge(src, store.get(key), tag=400 + src)
                 recv_seq[src] += 1
                 merged = True
         if merged:
@@ -76,8 +87,9 @@ def run():
             drain()
             if peers and rng.random() < p_push:
                 dst = rng.choice(peers)
-                exch.push(dst, tag=400 + me)       # isend (non-blocking)
-                store.set(f"go/push/{dst}/{me}/{send_seq[dst]}", "1")
+                a = exch.push(dst, tag=400 + me)   # isend (non-blocking)
+                # alpha rides the announcement (GPU wire = params only)
+                store.set(f"go/push/{dst}/{me}/{send_seq[dst]}", repr(a))
                 send_seq[dst] += 1
             rec.end("comm")
             uidx += 1
@@ -98,6 +110,10 @@ def run():
             raise RuntimeError("gosgd shutdown drain timed out "
                                "(peer likely died)")
     drain()                                   # announced-but-unreceived
+    if proc.on_gpu:
+        # pair-group comm streams are not covered by the nccl barrier;
+        # make sure every p2p kernel retired before teardown
+        torch.cuda.synchronize()
     proc.barrier()
 
     stats = torch.zeros(3)

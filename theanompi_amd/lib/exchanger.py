@@ -275,23 +275,52 @@ class EASGD_Exchanger:
 class GOSGD_Exchanger:
     """Gossip pushes (reference gosgd_worker.py [pub], C5).
 
-    Wire format per push: one flat fp32 tensor [alpha, w0, w1, ...] sent
-    over gloo (host memory).  Sender halves its own alpha and sends
-    (w_i scaled by alpha_i/2 implicitly via the merge rule); receiver
-    merges  w <- (a*w + a_i*w_i) / (a + a_i),  a <- a + a_i.
+    GPU path (reference C5 semantics: device-buffer Isend): the payload
+    is one flat fp32 device tensor riding RCCL p2p over the direct xGMI
+    link; alpha travels in the store announcement, so the merge
+    w <- (a*w + a_j*w_j)/(a+a_j) runs entirely on-device (a fused lerp
+    per parameter) with no host staging and no device<->host sync.
+    Each ordered rank pair gets its OWN communicator (directed pair
+    group): an RCCL send blocks its communicator's stream until the
+    matching recv is posted, so A->B and B->A pushes sharing one
+    communicator could head-of-line deadlock; per-direction comms make
+    the matching order independent.
+
+    CPU path: one flat fp32 tensor [alpha, w0, w1, ...] over gloo from
+    pinned host staging.  Sender halves its own alpha and sends; the
+    receiver merges  w <- (a*w + a_j*w_j) / (a + a_j),  a <- a + a_j.
     """
 
-    def __init__(self, config, param_list, group=None):
+    def __init__(self, config, param_list, group=None, pair_groups=None,
+                 rank=0):
         self.group = group
         self.params = list(param_list)
         self._flat, self._views = flatten_params(self.params)
         self.n = sum(f.numel() for f in self._flat)
         self.alpha = 1.0
+        self.rank = rank
         dev = self.params[0].device
+        self.on_gpu = dev.type == "cuda" and pair_groups is not None
+        self.pair_groups = pair_groups or {}
+        self._pending_send = None
+        if self.on_gpu:
+            self._dev_out = torch.empty(self.n, dtype=torch.float32,
+                                        device=dev)
+            self._dev_in = torch.empty_like(self._dev_out)
+            self._out_views, self._in_views = [], []
+            off = 0
+            for p in self.params:
+                n = p.numel()
+                self._out_views.append(
+                    self._dev_out[off:off + n].view(p.shape))
+                self._in_views.append(
+                    self._dev_in[off:off + n].view(p.shape))
+                off += n
+            return
         self._host_out = torch.empty(self.n + 1, dtype=torch.float32,
                                      pin_memory=dev.type == "cuda")
         self._host_in = torch.empty_like(self._host_out)
-        self._pending_send = None
+        self._rbufs = {}
 
     def _gather_flat(self, out):
         off = 1
@@ -309,21 +338,54 @@ class GOSGD_Exchanger:
             off += n
 
     def push(self, dst, tag=0):
-        """Send (alpha_i/2, w_i) to dst; halve own alpha."""
+        """Send (alpha_i/2, w_i) to dst; halve own alpha.  Returns the
+        alpha that rides the announcement (GPU wire carries params
+        only)."""
         if self._pending_send is not None:
+            # stream-level on RCCL: also orders the repack below after
+            # the previous send's kernel on the comm stream
             self._pending_send.wait()
         self.alpha *= 0.5
+        if self.on_gpu:
+            with torch.no_grad():
+                for v, p in zip(self._out_views, self.params):
+                    v.copy_(p.detach().to(torch.float32))
+            self._pending_send = dist.isend(
+                self._dev_out, dst=dst,
+                group=self.pair_groups[(self.rank, dst)])
+            return self.alpha
         self._host_out[0] = self.alpha
         self._gather_flat(self._host_out)
         self._pending_send = dist.isend(self._host_out, dst=dst,
                                         group=self.group, tag=tag)
+        return self.alpha
+
+    def recv_merge(self, src, announced, tag=0):
+        """Receive one announced push from src and merge it.
+        ``announced`` is the store value the sender published (its
+        alpha_j on GPU; unused on CPU where alpha rides the wire)."""
+        if self.on_gpu:
+            a_j = float(announced)
+            dist.recv(self._dev_in, src=src,
+                      group=self.pair_groups[(src, self.rank)])
+            t = a_j / (self.alpha + a_j)
+            with torch.no_grad():
+                for v, p in zip(self._in_views, self.params):
+                    # w <- (1-t)*w + t*w_j  ==  (a*w + a_j*w_j)/(a+a_j)
+                    p.data.lerp_(v.to(p.dtype), t)
+            self.alpha += a_j
+            return
+        buf = self._rbufs.setdefault(
+            src, torch.empty_like(self._host_in))
+        dist.recv(buf, src=src, group=self.group, tag=tag)
+        self.try_merge_buf(buf)
 
     def recv_buffer(self):
         """Allocate one host recv buffer (one per peer in the worker)."""
         return torch.empty_like(self._host_in)
 
     def try_merge_buf(self, buf):
-        """Merge a fully-received push buffer."""
+        """Merge a fully-received push buffer (CPU/gloo wire format)."""
         a_j = float(buf[0])
         a = self.alpha
         denom = a + a_j

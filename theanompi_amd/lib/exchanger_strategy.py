@@ -194,6 +194,84 @@ class ExchAsaBf16(ExchAsa):
     wire_dtype = torch.bfloat16
 
 
+class ExchAsaXgmi(_Strategy):
+    """'asa_xgmi': the SURVEY §5.8 design, fully explicit — the
+    reference paper's 3-phase ASA exchange with OUR reduction kernel:
+
+      1. all-to-all of 1/world bucket shards.  RCCL implements this as
+         grouped ncclSend/ncclRecv — on an 8-GPU xGMI node every rank
+         sends a distinct 1/8 shard to each of its 7 peers over that
+         peer's dedicated point-to-point link simultaneously (~7x a
+         ring's per-link bound).
+      2. local segment-sum across the world received copies — the
+         hand-written HIP kernel N14 (fp32 accumulation even on a bf16
+         wire, unlike ncclReduceScatter's wire-dtype math).
+      3. all-gather of the reduced shards.
+
+    Compared to 'asa' (fused ncclReduceScatter+AllGather) this trades
+    one extra HBM round-trip for explicit control of the reduction
+    dtype and per-link traffic; A/B both on the 8-GPU node.
+    """
+    name = "asa_xgmi"
+    wire_dtype = None
+
+    def __init__(self, process=None, group=None):
+        super().__init__(process, group)
+        self._scratch = {}
+
+    def _buffers(self, key, pn, dtype, device):
+        s = self._scratch.get(key)
+        if s is None or s[0].numel() != pn or s[0].dtype != dtype:
+            work = torch.zeros(pn, dtype=dtype, device=device)
+            recv = torch.empty(pn, dtype=dtype, device=device)
+            shard = torch.empty(pn // dist.get_world_size(self.group),
+                                dtype=dtype, device=device)
+            s = (work, recv, shard)
+            self._scratch[key] = s
+        return s
+
+    def exchange(self, flat_buffers, scale=None):
+        world = dist.get_world_size(self.group)
+        rank = dist.get_rank(self.group)
+        for buf in flat_buffers:
+            src = buf
+            if self.wire_dtype is not None and buf.dtype != self.wire_dtype:
+                w = self._scratch.get(("wire", id(buf)))
+                if w is None or w.numel() != buf.numel():
+                    w = torch.empty_like(buf, dtype=self.wire_dtype)
+                    self._scratch[("wire", id(buf))] = w
+                w.copy_(buf)
+                src = w
+            n = src.numel()
+            # pad to a multiple of 4*world: equal shards, 4-elem
+            # vectorized segsum
+            q = 4 * world
+            pn = ((n + q - 1) // q) * q
+            work, recv, shard = self._buffers(id(buf), pn, src.dtype,
+                                              src.device)
+            work[:n].copy_(src.view(-1))
+            if pn > n:
+                work[n:].zero_()
+            # phase 1: all-to-all (grouped RCCL send/recv, all links)
+            dist.all_to_all_single(recv, work, group=self.group)
+            # phase 2: local segment-sum (N14)
+            if work.is_cuda:
+                from theanompi_amd import ops as _ops
+                _ops.ext().segsum(recv, shard, world)
+            else:
+                torch.sum(recv.view(world, -1), dim=0, out=shard)
+            # phase 3: all-gather the reduced shards
+            dist.all_gather_into_tensor(work, shard, group=self.group)
+            buf.view(-1).copy_(work[:n])
+            if scale is not None:
+                buf.mul_(scale)
+
+
+class ExchAsaXgmiBf16(ExchAsaXgmi):
+    name = "asa_xgmi_bf16"
+    wire_dtype = torch.bfloat16
+
+
 _STRATEGIES = {
     "ar": ExchAllreduceHost,
     "rccl32": ExchRccl32,
@@ -205,6 +283,8 @@ _STRATEGIES = {
     "asa32": ExchAsa,
     "asa_bf16": ExchAsaBf16,
     "asa16": ExchAsaBf16,
+    "asa_xgmi": ExchAsaXgmi,
+    "asa_xgmi_bf16": ExchAsaXgmiBf16,
 }
 
 

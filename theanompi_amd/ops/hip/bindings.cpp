@@ -77,6 +77,7 @@ hipError_t launch_u8_finish(const unsigned char*, void*, const float*,
                             hipStream_t);
 hipError_t launch_pack_bf16(const float*, void*, long, hipStream_t);
 hipError_t launch_unpack_bf16(const void*, float*, long, hipStream_t);
+hipError_t launch_segsum(const void*, void*, long, int, int, hipStream_t);
 hipError_t launch_maxpool_fwd(const void*, void*, unsigned char*, int, int,
                               int, int, int, int, int, int, int, int,
                               hipStream_t);
@@ -588,6 +589,19 @@ static void unpack_bf16(torch::Tensor src, torch::Tensor dst) {
                                src.numel(), cur_stream()));
 }
 
+static void segsum(torch::Tensor in, torch::Tensor out, long w) {
+  // in: [w*seg] contiguous, out: [seg]; fp32 accumulation (N14)
+  TORCH_CHECK(in.dtype() == out.dtype(), "segsum: dtype mismatch");
+  TORCH_CHECK(in.is_contiguous() && out.is_contiguous());
+  long seg = out.numel();
+  TORCH_CHECK(in.numel() == w * seg, "segsum: in != w*out");
+  TORCH_CHECK(seg % 4 == 0, "segsum: seg must be a multiple of 4");
+  int bf16 = in.dtype() == torch::kBFloat16;
+  TORCH_CHECK(bf16 || in.dtype() == torch::kFloat32);
+  CHECK_HIP(launch_segsum(in.data_ptr(), out.data_ptr(), seg, (int)w,
+                          bf16, cur_stream()));
+}
+
 // ---- pooling ----------------------------------------------------------------
 
 static void maxpool_fwd(torch::Tensor xp, torch::Tensor y,
@@ -899,6 +913,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("meanimg") = py::none());
   m.def("pack_bf16", &pack_bf16);
   m.def("unpack_bf16", &unpack_bf16);
+  m.def("segsum", &segsum);
   m.def("maxpool_fwd", &maxpool_fwd);
   m.def("maxpool_bwd", &maxpool_bwd);
   m.def("avgpool_fwd", &avgpool_fwd);

@@ -264,6 +264,31 @@ __global__ void unpack_bf16_k(const __bf16* __restrict__ src,
   }
 }
 
+// ---- N14: segment sum for the asa_xgmi exchange ---------------------------
+// Rebuild of the reference's hand-written ASA segment-sum CUDA kernel
+// (exchanger_strategy.py [pub]): after the all-to-all
+// phase each rank holds `w` remote copies of its 1/world bucket shard;
+// out[i] = sum_j in[j*seg + i], accumulated in fp32 even on a bf16 wire.
+// seg is a multiple of 4 (the strategy pads buckets to 4*world elems).
+
+template <typename T, typename V4>
+__global__ void segsum_k(const T* __restrict__ in, T* __restrict__ out,
+                         long seg4, int w) {
+  for (long i = blockIdx.x * blockDim.x + threadIdx.x; i < seg4;
+       i += (long)gridDim.x * blockDim.x) {
+    float acc[4] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < w; ++j) {
+      V4 v = *(const V4*)(in + (j * seg4 + i) * 4);
+#pragma unroll
+      for (int k = 0; k < 4; ++k) acc[k] += (float)v[k];
+    }
+    V4 o;
+#pragma unroll
+    for (int k = 0; k < 4; ++k) o[k] = (T)acc[k];
+    *(V4*)(out + i * 4) = o;
+  }
+}
+
 // ---------------------------------------------------------------------------
 
 static inline int ew_grid(long work, int block = 256) {
@@ -366,6 +391,21 @@ hipError_t launch_pack_bf16(const float* src, void* dst, long n,
                             hipStream_t stream) {
   hipLaunchKernelGGL(pack_bf16_k, dim3(ew_grid(n / 4)), dim3(256), 0,
                      stream, src, (__bf16*)dst, n / 4);
+  return hipGetLastError();
+}
+
+hipError_t launch_segsum(const void* in, void* out, long seg, int w,
+                         int is_bf16, hipStream_t stream) {
+  long seg4 = seg / 4;
+  if (is_bf16) {
+    hipLaunchKernelGGL((segsum_k<__bf16, bf16x4>), dim3(ew_grid(seg4)),
+                       dim3(256), 0, stream, (const __bf16*)in,
+                       (__bf16*)out, seg4, w);
+  } else {
+    hipLaunchKernelGGL((segsum_k<float, f32x4>), dim3(ew_grid(seg4)),
+                       dim3(256), 0, stream, (const float*)in,
+                       (float*)out, seg4, w);
+  }
   return hipGetLastError();
 }
 
