@@ -339,6 +339,14 @@ class _HipConvFn(torch.autograd.Function):
         else:
             y = torch.empty((N, plan.P, plan.Q, plan.Kphys),
                             dtype=torch.bfloat16, device=x.device)
+        vf = layer._v_fwd
+        if vf is None or layer._v_M != M:
+            from theanompi_amd.ops import tuning
+            vf = tuning.variant_for("fwd", M, plan.Kgp, plan.Kdf, plan.G)
+            layer._v_fwd = vf
+            layer._v_M = M
+            layer._v_wgrad = tuning.wgrad_for(M, plan.Kgp, plan.Kdf,
+                                              plan.G)
         layer._stats_valid = C.conv_fwd(
             xp, layer.w_bf16,
             layer.b_f32 if layer.b_f32 is not None else
@@ -348,7 +356,7 @@ class _HipConvFn(torch.autograd.Function):
             plan.Cphys,                          # pixel col stride (w)
             plan.stride[0], plan.stride[1],
             plan.Cg if plan.G > 1 else 0,        # group in-chan offset
-            bool(layer.relu), stats)
+            bool(layer.relu), stats, vf)
         ctx.save_for_backward(xp, y if layer.relu else torch.empty(0))
         ctx.layer = layer
         return y
@@ -392,10 +400,11 @@ class _HipConvFn(torch.autograd.Function):
                                   and layer.weight.grad is None) else \
             torch.zeros(plan.flat_shape(), dtype=torch.float32,
                         device=dy.device)
+        wv, wt = getattr(layer, "_v_wgrad", (0, 0)) or (0, 0)
         C.conv_wgrad(xp, dy, plan.table, dw, plan.kdf_mask,
                      plan.Wp * plan.Cphys, plan.Cphys,
                      plan.stride[0], plan.stride[1],
-                     plan.Cg if plan.G > 1 else 0)
+                     plan.Cg if plan.G > 1 else 0, wv, wt)
         dx = None
         if ctx.needs_input_grad[0]:
             dx = layer._dgrad(dy)
@@ -521,6 +530,9 @@ class Conv2d(_Base):
     _collect_stats = False          # class defaults: Linear shares the
     _stats_buf = None               # conv path but not __init__
     _stats_valid = False
+    _v_fwd = None                   # tuned kernel variants (ops.tuning),
+    _v_M = -1                       # cached per observed GEMM-M
+    _v_wgrad = (0, 0)
 
     def pop_stats(self):
         """Bucketed output statistics from the last forward (fused BN),
@@ -547,12 +559,20 @@ class Conv2d(_Base):
                 if (ht or hb or hl or hr) else dy
             i, j = plan2["cls"]
             Qp = plan.Q + hl + hr
+            vd = plan2.get("variant")
+            if vd is None or plan2.get("vN") != N:
+                from theanompi_amd.ops import tuning
+                vd = tuning.variant_for(
+                    "dgrad", N * plan2["Hi"] * plan2["Wj"], plan.Cg,
+                    plan2["Kdf2"], plan.G)
+                plan2["variant"] = vd
+                plan2["vN"] = N
             C.conv_dgrad(dyp, wdg, plan2["table"], dx,
                          Qp * plan.Kphys, plan.Kphys,
                          plan2["Hi"], plan2["Wj"],
                          i, j, plan.stride[0], plan.stride[1],
                          plan.Kgp if plan.G > 1 else 0,
-                         plan.Cg)
+                         plan.Cg, vd)
         return dx
 
     def _alloc_wdg(self, maps):

@@ -50,9 +50,9 @@ struct WgradArgs {
 
 extern "C" {
 hipError_t launch_conv_gather_gemm(GatherGemmArgs*, int G, int layout,
-                                   hipStream_t);
+                                   hipStream_t, int variant);
 hipError_t launch_conv_wgrad(const WgradArgs*, int G, int layout,
-                             hipStream_t);
+                             hipStream_t, int variant);
 hipError_t launch_bias_wgrad(const void*, float*, long, int, hipStream_t);
 hipError_t launch_halo_pad(const void*, void*, int, int, int, int, int, int,
                            int, int, float, hipStream_t);
@@ -282,7 +282,8 @@ static bool conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
                      torch::Tensor table, torch::Tensor y, long rowStride,
                      long colStride, long sh, long sw, long groupInOff,
                      bool relu,
-                     c10::optional<torch::Tensor> stats = c10::nullopt) {
+                     c10::optional<torch::Tensor> stats = c10::nullopt,
+                     long variant = 0) {
   check_bf16(xp, "xp");
   check_bf16(w, "w");
   // y may be a channel slice of a concat buffer (concat-writing conv:
@@ -340,7 +341,7 @@ static bool conv_fwd(torch::Tensor xp, torch::Tensor w, torch::Tensor bias,
     a.stats = st.data_ptr<float>();
   }
   CHECK_HIP(launch_conv_gather_gemm(&a, (int)G, detect_layout(),
-                                    cur_stream()));
+                                    cur_stream(), (int)variant));
   if (a.splitK > 1) {
     CHECK_HIP(launch_splitk_finalize(a.Y32, a.bias, y.data_ptr(),
                                      y.numel(), (int)(G * Kgp),
@@ -353,7 +354,7 @@ static void conv_dgrad(torch::Tensor dyp, torch::Tensor wdg,
                        torch::Tensor table, torch::Tensor dx,
                        long rowStride, long colStride, long Hi, long Wj,
                        long clsI, long clsJ, long sh, long sw,
-                       long groupInOff, long Cgp) {
+                       long groupInOff, long Cgp, long variant = 0) {
   check_bf16(dyp, "dyp");
   check_bf16(wdg, "wdg");
   check_bf16(dx, "dx");
@@ -398,7 +399,7 @@ static void conv_dgrad(torch::Tensor dyp, torch::Tensor wdg,
       a.Y32 = splitk_ws(dx.numel(), dx.options());
   }
   CHECK_HIP(launch_conv_gather_gemm(&a, (int)G, detect_layout(),
-                                    cur_stream()));
+                                    cur_stream(), (int)variant));
   if (a.splitK > 1) {
     CHECK_HIP(launch_splitk_finalize(a.Y32, nullptr, dx.data_ptr(),
                                      dx.numel(), (int)Cgp, 0,
@@ -410,7 +411,8 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
                        torch::Tensor table, torch::Tensor dw,
                        torch::Tensor kdfMask,
                        long rowStride, long colStride, long sh, long sw,
-                       long groupInOff) {
+                       long groupInOff, long variant = 0,
+                       long split_target = 0) {
   check_bf16(xp, "xp");
   check_bf16(dy, "dy");
   check_f32(dw, "dw");
@@ -445,13 +447,16 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
   // grouped convs shrink the per-group GEMM (few base blocks) and
   // measured ~1.5% faster end-to-end with a 2048-block target;
   // ungrouped layers measured best at 1024 (swept on hardware).
-  long wgrad_target = wgrad_env ? wgrad_env : (G > 1 ? 2048 : 1024);
+  long wgrad_target = wgrad_env ? wgrad_env
+                      : split_target ? split_target
+                      : (G > 1 ? 2048 : 1024);
   long splitK = wgrad_target / (base_blocks ? base_blocks : 1);
   long maxSplit = (a.M + 63) / 64;
   if (splitK < 1) splitK = 1;
   if (splitK > maxSplit) splitK = maxSplit;
   a.splitK = (int)splitK;
-  CHECK_HIP(launch_conv_wgrad(&a, (int)G, detect_layout(), cur_stream()));
+  CHECK_HIP(launch_conv_wgrad(&a, (int)G, detect_layout(), cur_stream(),
+                              (int)variant));
 }
 
 static void bias_wgrad(torch::Tensor dy, torch::Tensor db) {
@@ -898,9 +903,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("bias"), py::arg("table"), py::arg("y"),
         py::arg("rowStride"), py::arg("colStride"), py::arg("sh"),
         py::arg("sw"), py::arg("groupInOff"), py::arg("relu"),
-        py::arg("stats") = py::none());
-  m.def("conv_dgrad", &conv_dgrad);
-  m.def("conv_wgrad", &conv_wgrad);
+        py::arg("stats") = py::none(), py::arg("variant") = 0);
+  m.def("conv_dgrad", &conv_dgrad, py::arg("dyp"), py::arg("wdg"),
+        py::arg("table"), py::arg("dx"), py::arg("rowStride"),
+        py::arg("colStride"), py::arg("Hi"), py::arg("Wj"),
+        py::arg("clsI"), py::arg("clsJ"), py::arg("sh"), py::arg("sw"),
+        py::arg("groupInOff"), py::arg("Cgp"), py::arg("variant") = 0);
+  m.def("conv_wgrad", &conv_wgrad, py::arg("xp"), py::arg("dy"),
+        py::arg("table"), py::arg("dw"), py::arg("kdfMask"),
+        py::arg("rowStride"), py::arg("colStride"), py::arg("sh"),
+        py::arg("sw"), py::arg("groupInOff"), py::arg("variant") = 0,
+        py::arg("split_target") = 0);
   m.def("bias_wgrad", &bias_wgrad);
   m.def("relu_fwd", &relu_fwd);
   m.def("relu_bwd", &relu_bwd);

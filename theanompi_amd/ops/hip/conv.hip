@@ -70,8 +70,11 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
   __shared__ __align__(16) __bf16 Bs[DEPTH][BN][BK];
   // gather table staged to LDS once: its reads then ride the lgkm
   // counter, so they never force a vmcnt drain of in-flight glds
-  // (cdna guide §5 "three .s-level traps" (b)).
-  __shared__ int Tab[2048];
+  // (cdna guide §5 "three .s-level traps" (b)).  Dynamically sized
+  // (launcher passes Kdf/16 ints): most layers need <1 KB, and the old
+  // fixed 8 KB was the difference between 2 and 3 blocks/CU on the
+  // slimmer tile variants.
+  extern __shared__ int Tab[];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -376,7 +379,7 @@ __global__ __launch_bounds__(512) void conv_gather_gemm_8p(
 
   __shared__ __align__(16) __bf16 As[3][BM][BK];
   __shared__ __align__(16) __bf16 Bs[3][BN][BK];
-  __shared__ int Tab[2048];
+  extern __shared__ int Tab[];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -595,7 +598,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   
   __shared__ __align__(16) __bf16 DYt[2][BKP][64];        // [pix][kout]
   __shared__ __align__(16) __bf16 XT[2][BKP][BNF];        // [pix][fd]
-  __shared__ int Tab[2048];
+  extern __shared__ int Tab[];
 
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -811,104 +814,124 @@ hipError_t launch_splitk_finalize(float* y32, const float* bias,
   return hipGetLastError();
 }
 
+// Variant table for the per-shape autotuner (tools/autotune_conv.py;
+// table shipped as theanompi_amd/ops/conv_tuned.json, threaded through
+// the bindings as the `variant` argument):
+//   0  auto (BN by width heuristic + env overrides — round-1 behavior)
+//   1  128x128 D2 BK64          5  128x128 D2 BK32
+//   2  128x 64 D2 BK64          6  128x 64 D2 BK32
+//   3  256x128 D2 BK64 (8 wv)   7  128x128 D3 BK32
+//   4  256x128 8-phase          8  128x 64 D3 BK32
+//                               9  128x 64 D3 BK64
+//                              10  128x128 D3 BK64
+// Slimmer tiles (BN=64 / BK=32 / dynamic gather table) trade MFMA work
+// per staged byte for occupancy — the round-1 PMC data shows the
+// default tile L2-LATENCY bound (WAIT_ANY ~50%) on mid-size shapes, so
+// more blocks/CU is the lever, measured per shape, not guessed.
 hipError_t launch_conv_gather_gemm(GatherGemmArgs* args, int G,
-                                   int layout, hipStream_t stream) {
+                                   int layout, hipStream_t stream,
+                                   int variant) {
   GatherGemmArgs a = *args;
   if (a.Kdf > 2048 * 16) return hipErrorInvalidValue;  // LDS table cap
+  const int tabBytes = ((a.Kdf >> 4) * 4 + 15) & ~15;
   // fused BN stats only exist on the default LDS-staged epilogue with
   // splitK == 1; clear the field (and tell the caller) on every other
   // route so the host falls back to the separate bn_sums kernel
   if (a.splitK > 1) { a.stats = nullptr; args->stats = nullptr; }
-  // big-tile path: 256x128, 8 waves -- 2x MFMA work per staged byte,
-  // for large-M dense layers
+
   static int bigTile = -1;
   if (bigTile < 0) {
     const char* e = getenv("THEANOMPI_GEMM_BIG");
-    bigTile = e ? atoi(e) : 0;   // needs the 8-phase schedule to pay
+    bigTile = e ? atoi(e) : 0;
   }
-  if (bigTile && a.splitK == 1 && a.M >= 4096 && a.BNtot >= 128
-      && a.Kdf >= 128) {
-    a.stats = nullptr;
-    args->stats = nullptr;
-    dim3 grid(cdiv(a.M, 256), cdiv(a.BNtot, 128), G);
-    dim3 block(512);
-    if (bigTile == 2 && a.Kdf >= 3 * 64) {
-      if (layout == FRAG_CONTIG8)
-        hipLaunchKernelGGL((conv_gather_gemm_8p<FRAG_CONTIG8>), grid,
-                           block, 0, stream, a);
-      else
-        hipLaunchKernelGGL((conv_gather_gemm_8p<FRAG_HALVES>), grid,
-                           block, 0, stream, a);
-    } else if (layout == FRAG_CONTIG8) {
-      hipLaunchKernelGGL((conv_gather_gemm<256, 128, FRAG_CONTIG8, 2>),
-                         grid, block, 0, stream, a);
-    } else {
-      hipLaunchKernelGGL((conv_gather_gemm<256, 128, FRAG_HALVES, 2>),
-                         grid, block, 0, stream, a);
-    }
-    return hipGetLastError();
-  }
-  int BN = a.BNtot > 64 ? 128 : 64;
-  dim3 grid(cdiv(a.M, 128), cdiv(a.BNtot, BN), G * a.splitK);
-  dim3 block(256);
   static int bk32 = -1;
   if (bk32 < 0) {
     const char* e = getenv("THEANOMPI_GEMM_BK32");
     bk32 = e ? atoi(e) : 0;
   }
-  if (bk32 && a.Kdf % 64 == 0) {
-    a.stats = nullptr;
-    args->stats = nullptr;
-    if (BN == 128) {
-      if (layout == FRAG_CONTIG8)
-        hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_CONTIG8, 2,
-                                             32>),
-                           grid, block, 0, stream, a);
-      else
-        hipLaunchKernelGGL((conv_gather_gemm<128, 128, FRAG_HALVES, 2,
-                                             32>),
-                           grid, block, 0, stream, a);
-    } else {
-      if (layout == FRAG_CONTIG8)
-        hipLaunchKernelGGL((conv_gather_gemm<128, 64, FRAG_CONTIG8, 2,
-                                             32>),
-                           grid, block, 0, stream, a);
-      else
-        hipLaunchKernelGGL((conv_gather_gemm<128, 64, FRAG_HALVES, 2,
-                                             32>),
-                           grid, block, 0, stream, a);
-    }
-    return hipGetLastError();
+  static int forceVar = -1;
+  if (forceVar < 0) {
+    const char* e = getenv("THEANOMPI_GEMM_VARIANT");
+    forceVar = e ? atoi(e) : 0;
   }
-  const int depth = 2;   // 3-deep ring measured slower (1 blk/CU)
-#define GG_LAUNCH(BN_, L_, D_)                                             \
-  hipLaunchKernelGGL((conv_gather_gemm<128, BN_, L_, D_>), grid, block,    \
-                     0, stream, a)
-  if (BN == 128) {
-    if (layout == FRAG_CONTIG8) {
-      if (depth == 3) GG_LAUNCH(128, FRAG_CONTIG8, 3);
-      else GG_LAUNCH(128, FRAG_CONTIG8, 2);
-    } else {
-      if (depth == 3) GG_LAUNCH(128, FRAG_HALVES, 3);
-      else GG_LAUNCH(128, FRAG_HALVES, 2);
-    }
-  } else {
-    if (layout == FRAG_CONTIG8) {
-      if (depth == 3) GG_LAUNCH(64, FRAG_CONTIG8, 3);
-      else GG_LAUNCH(64, FRAG_CONTIG8, 2);
-    } else {
-      if (depth == 3) GG_LAUNCH(64, FRAG_HALVES, 3);
-      else GG_LAUNCH(64, FRAG_HALVES, 2);
-    }
+  if (forceVar) variant = forceVar;
+
+  // map the legacy env switches onto variants
+  if (variant == 0) {
+    if (bigTile && a.splitK == 1 && a.M >= 4096 && a.BNtot >= 128
+        && a.Kdf >= 128)
+      variant = (bigTile == 2 && a.Kdf >= 3 * 64) ? 4 : 3;
+    else if (bk32)
+      variant = a.BNtot > 64 ? 5 : 6;
+    else
+      variant = a.BNtot > 64 ? 1 : 2;
   }
-#undef GG_LAUNCH
+  // feasibility clamps
+  if (variant == 4 && a.Kdf < 3 * 64) variant = 3;
+  if ((variant == 3 || variant == 4) && a.splitK > 1) {
+    // big tiles carry no split-K path; drop the split (caller's
+    // finalize is skipped because args->splitK is rewritten)
+    a.splitK = 1;
+    args->splitK = 1;
+    a.Y32 = nullptr;
+    args->Y32 = nullptr;
+  }
+  // fused stats ride only the LDS-staged bf16 epilogue (2*BK >= BN)
+  const bool ldsep = (variant == 1 || variant == 3 || variant == 10)
+                     ? true
+                     : (variant == 2 || variant == 6 || variant == 8
+                        || variant == 9);
+  if (!ldsep || variant == 4) { a.stats = nullptr; args->stats = nullptr; }
+
+  const int BM = (variant == 3 || variant == 4) ? 256 : 128;
+  const int BN = (variant == 2 || variant == 6 || variant == 8
+                  || variant == 9) ? 64 : 128;
+  dim3 grid(cdiv(a.M, BM), cdiv(a.BNtot, BN), G * a.splitK);
+  dim3 block(BM * 2);
+
+#define GG(BM_, BN_, D_, BK_)                                              \
+  do {                                                                     \
+    if (layout == FRAG_CONTIG8)                                            \
+      hipLaunchKernelGGL((conv_gather_gemm<BM_, BN_, FRAG_CONTIG8, D_,     \
+                                           BK_>),                          \
+                         grid, block, tabBytes, stream, a);                \
+    else                                                                   \
+      hipLaunchKernelGGL((conv_gather_gemm<BM_, BN_, FRAG_HALVES, D_,      \
+                                           BK_>),                          \
+                         grid, block, tabBytes, stream, a);                \
+  } while (0)
+  switch (variant) {
+    case 3: GG(256, 128, 2, 64); break;
+    case 4:
+      if (layout == FRAG_CONTIG8)
+        hipLaunchKernelGGL((conv_gather_gemm_8p<FRAG_CONTIG8>), grid,
+                           block, tabBytes, stream, a);
+      else
+        hipLaunchKernelGGL((conv_gather_gemm_8p<FRAG_HALVES>), grid,
+                           block, tabBytes, stream, a);
+      break;
+    case 2: GG(128, 64, 2, 64); break;
+    case 5: GG(128, 128, 2, 32); break;
+    case 6: GG(128, 64, 2, 32); break;
+    case 7: GG(128, 128, 3, 32); break;
+    case 8: GG(128, 64, 3, 32); break;
+    case 9: GG(128, 64, 3, 64); break;
+    case 10: GG(128, 128, 3, 64); break;
+    case 1:
+    default: GG(128, 128, 2, 64); break;
+  }
+#undef GG
   return hipGetLastError();
 }
 
+// wgrad variants (autotuner): 0 auto (BNF heuristic + env), 1 BNF=64,
+// 2 BNF=128.  The gather table is dynamically sized here too: at
+// BNF=128 that is the difference between 2 and 3 blocks/CU.
 hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
-                             hipStream_t stream) {
+                             hipStream_t stream, int variant) {
   WgradArgs a = *args;
   if (a.Kdf > 2048 * 16) return hipErrorInvalidValue;
+  const int tabBytes = ((a.Kdf >> 4) * 4 + 15) & ~15;
   static const int bnf_env = [] {
     const char* e = getenv("THEANOMPI_WGRAD_BNF");
     return e ? atoi(e) : 0;
@@ -918,22 +941,24 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
   // epilogues were parallelized; THEANOMPI_WGRAD_BNF=128 restores it.
   int BNF = 64;
   if (bnf_env == 128 && a.Kdf % 128 == 0) BNF = 128;
+  if (variant == 1) BNF = 64;
+  if (variant == 2 && a.Kdf % 128 == 0) BNF = 128;
   dim3 grid(cdiv(a.Kgp, 64), cdiv(a.Kdf, BNF), G * a.splitK);
   dim3 block(256);
   if (BNF == 128) {
     if (layout == FRAG_CONTIG8)
       hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, 128>), grid, block,
-                         0, stream, a);
+                         tabBytes, stream, a);
     else
       hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, 128>), grid, block,
-                         0, stream, a);
+                         tabBytes, stream, a);
   } else {
     if (layout == FRAG_CONTIG8)
       hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, 64>), grid, block,
-                         0, stream, a);
+                         tabBytes, stream, a);
     else
       hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, 64>), grid, block,
-                         0, stream, a);
+                         tabBytes, stream, a);
   }
   return hipGetLastError();
 }
