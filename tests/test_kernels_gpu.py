@@ -371,6 +371,13 @@ def test_softmax_xent():
     loss_r.backward()
     assert rel_err(lg.grad.float().cpu()[:, :CLS], lr_.grad[:, :CLS]) < 3e-2
     assert lg.grad.float()[:, CLS:].abs().sum() == 0
+    # fused loss-mean + top-1 err (kernel atomics) vs torch reductions
+    lg2 = logits.to(dev()).to(torch.bfloat16)
+    loss_f, _p, err_f = OF.softmax_xent_err(lg2, labels.to(dev()), CLS)
+    pred = lg2.float()[:, :CLS].argmax(dim=1)
+    err_ref = (pred != labels.to(dev())).float().mean()
+    assert abs(float(err_f) - float(err_ref)) < 1e-6
+    assert abs(float(loss_f) - float(loss_r)) < 2e-2 * max(1, float(loss_r))
 
 
 def test_sgd_step_kernel():

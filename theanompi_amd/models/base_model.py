@@ -223,11 +223,9 @@ class BaseModel:
     def _fwd_bwd(self, x, y):
         self.opt.zero_grad()
         logits = self.forward(x)
-        loss, _probs = OF.softmax_xent(logits, y, self.n_classes)
+        # loss mean + top-1 err ride the xent kernel (fused reductions)
+        loss, _probs, err = OF.softmax_xent_err(logits, y, self.n_classes)
         loss.backward()
-        with torch.no_grad():
-            pred = logits[:, :self.n_classes].argmax(dim=1)
-            err = (pred != y).float().mean()
         return loss.detach(), err
 
     def apply_update(self):

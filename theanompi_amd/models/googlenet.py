@@ -163,17 +163,14 @@ class GoogLeNet(BaseModel):
         out = self.module(x)
         if isinstance(out, tuple):
             logits, a1, a2 = out
-            loss0, _ = OF.softmax_xent(logits, y, self.n_classes)
+            loss0, _, err = OF.softmax_xent_err(logits, y, self.n_classes)
             loss1, _ = OF.softmax_xent(a1, y, self.n_classes)
             loss2, _ = OF.softmax_xent(a2, y, self.n_classes)
             loss = loss0 + 0.3 * (loss1 + loss2)
         else:
             logits = out
-            loss, _ = OF.softmax_xent(logits, y, self.n_classes)
+            loss, _, err = OF.softmax_xent_err(logits, y, self.n_classes)
         loss.backward()
-        with torch.no_grad():
-            pred = logits[:, :self.n_classes].argmax(dim=1)
-            err = (pred != y).float().mean()
         return loss.detach(), err
 
     def build_data(self):

@@ -418,6 +418,13 @@ class _HipConvFn(torch.autograd.Function):
 class _Base(nn.Module):
     """Common bits: device/gpu flag + checkpoint hooks."""
 
+    # tuned kernel variants (ops.tuning), cached per observed GEMM-M —
+    # shared by every layer that routes through _HipConvFn (Conv2d AND
+    # Linear-as-1x1-conv)
+    _v_fwd = None
+    _v_M = -1
+    _v_wgrad = (0, 0)
+
     def logical_params(self):
         """[(name, getter()->np-compatible tensor, setter(tensor))] in the
         reference's per-layer W-then-b order."""
@@ -530,9 +537,6 @@ class Conv2d(_Base):
     _collect_stats = False          # class defaults: Linear shares the
     _stats_buf = None               # conv path but not __init__
     _stats_valid = False
-    _v_fwd = None                   # tuned kernel variants (ops.tuning),
-    _v_M = -1                       # cached per observed GEMM-M
-    _v_wgrad = (0, 0)
 
     def pop_stats(self):
         """Bucketed output statistics from the last forward (fused BN),

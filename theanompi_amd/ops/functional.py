@@ -303,17 +303,25 @@ class _SoftmaxXentFn(torch.autograd.Function):
         b = logits.shape[0]
         probs = torch.empty_like(logits)
         loss = torch.empty(b, dtype=torch.float32, device=logits.device)
-        C.softmax_xent_fwd(logits, labels, probs, loss, int(n_classes))
+        # fused batch reductions: stats = [sum(loss), count(top1 err)],
+        # accumulated by the kernel's per-row atomics — no torch
+        # mean/argmax glue in the train step
+        stats = torch.zeros(2, dtype=torch.float32, device=logits.device)
+        C.softmax_xent_fwd(logits, labels, probs, loss, int(n_classes),
+                           stats)
         ctx.save_for_backward(probs, labels)
         ctx.n_classes = n_classes
-        # probs is an inference-only output: backward below only
+        sm = stats * (1.0 / b)
+        # probs/err are inference-only outputs: backward below only
         # accounts for dloss, so autograd must not route gradients
-        # through probs silently
+        # through them silently
         ctx.mark_non_differentiable(probs)
-        return loss.mean(), probs
+        err = sm[1]
+        ctx.mark_non_differentiable(err)
+        return sm[0], probs, err
 
     @staticmethod
-    def backward(ctx, dloss, dprobs):
+    def backward(ctx, dloss, dprobs, derr):
         probs, labels = ctx.saved_tensors
         C = _ops.ext()
         dlogits = torch.empty_like(probs)
@@ -387,5 +395,21 @@ def softmax_xent(logits, labels, n_classes=None):
     """Fused softmax + cross-entropy.  Returns (mean_loss, probs)."""
     n_classes = n_classes or logits.shape[1]
     if logits.is_cuda:
-        return _SoftmaxXentFn.apply(logits, labels, n_classes)
+        loss, probs, _err = _SoftmaxXentFn.apply(logits, labels, n_classes)
+        return loss, probs
     return softmax_xent_ref(logits, labels, n_classes)
+
+
+def softmax_xent_err(logits, labels, n_classes=None):
+    """softmax_xent plus the fused top-1 training error: on GPU the
+    loss mean AND the err count ride the xent kernel's atomics, so the
+    train step needs no torch argmax/mean glue.  Returns
+    (mean_loss, probs, err)."""
+    n_classes = n_classes or logits.shape[1]
+    if logits.is_cuda:
+        return _SoftmaxXentFn.apply(logits, labels, n_classes)
+    loss, probs = softmax_xent_ref(logits, labels, n_classes)
+    with torch.no_grad():
+        pred = logits[:, :n_classes].argmax(dim=1)
+        err = (pred != labels).float().mean()
+    return loss, probs, err

@@ -108,7 +108,7 @@ hipError_t launch_bn_bwd(const void*, const void*, const void*, const float*,
                          int*, float*, float*, void*, long, int, int,
                          hipStream_t);
 hipError_t launch_softmax_xent_fwd(const void*, const long*, void*, float*,
-                                   int, int, int, hipStream_t);
+                                   int, int, int, float*, hipStream_t);
 hipError_t launch_softmax_xent_bwd(const void*, const long*, void*, int, int,
                                    int, float, hipStream_t);
 hipError_t launch_sgd_step(float*, const float*, float*, void*, long, float,
@@ -755,15 +755,25 @@ static void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma,
 
 static void softmax_xent_fwd(torch::Tensor logits, torch::Tensor labels,
                              torch::Tensor probs, torch::Tensor loss,
-                             long nCls) {
+                             long nCls,
+                             c10::optional<torch::Tensor> stats
+                             = c10::nullopt) {
   check_bf16(logits, "logits");
   TORCH_CHECK(labels.scalar_type() == torch::kInt64);
   long B = logits.size(0), Cp = logits.size(1);
+  float* sp = nullptr;
+  if (stats.has_value()) {
+    auto& st = stats.value();
+    TORCH_CHECK(st.scalar_type() == torch::kFloat32 && st.is_cuda()
+                && st.numel() == 2, "stats must be fp32 [2]");
+    sp = st.data_ptr<float>();
+  }
   CHECK_HIP(launch_softmax_xent_fwd(logits.data_ptr(),
                                     labels.data_ptr<long>(),
                                     probs.data_ptr(),
                                     loss.data_ptr<float>(), (int)B,
-                                    (int)Cp, (int)nCls, cur_stream()));
+                                    (int)Cp, (int)nCls, sp,
+                                    cur_stream()));
 }
 
 static void softmax_xent_bwd(torch::Tensor probs, torch::Tensor labels,
@@ -939,7 +949,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("momentum"), py::arg("eps"), py::arg("training"),
         py::arg("relu"), py::arg("prestats") = py::none());
   m.def("bn_bwd", &bn_bwd);
-  m.def("softmax_xent_fwd", &softmax_xent_fwd);
+  m.def("softmax_xent_fwd", &softmax_xent_fwd, py::arg("logits"),
+        py::arg("labels"), py::arg("probs"), py::arg("loss"),
+        py::arg("nCls"), py::arg("stats") = py::none());
   m.def("softmax_xent_bwd", &softmax_xent_bwd);
   m.def("sgd_step", &sgd_step);
   m.def("mfma_layout", &mfma_layout);

@@ -13,34 +13,46 @@ __global__ void softmax_xent_fwd_k(const __bf16* __restrict__ logits,
                                    const long* __restrict__ labels,
                                    __bf16* __restrict__ probs,
                                    float* __restrict__ loss, int Cp,
-                                   int nCls) {
+                                   int nCls,
+                                   float* __restrict__ stats) {
+  // stats (optional, 2 fp32): fused batch reductions — stats[0] +=
+  // row loss, stats[1] += (argmax != label).  Replaces the torch
+  // loss.mean()/argmax/ne/mean glue kernels in the train step
+  // (hot path must be hand-written).
   const int b = blockIdx.x;
   const int tid = threadIdx.x;
   const __bf16* row = logits + (long)b * Cp;
   __shared__ float red[8];
+  __shared__ int redi[8];
 
-  // 1) row max
+  // 1) row max (+ argmax for the fused top-1 error)
   float mx = -3.0e38f;
+  int am = 0;
   for (int c = tid; c < nCls; c += blockDim.x) {
     float v = (float)row[c];
-    mx = v > mx ? v : mx;
+    if (v > mx) { mx = v; am = c; }
   }
   for (int off = 32; off > 0; off >>= 1) {
     float o = __shfl_down(mx, off, 64);
-    mx = o > mx ? o : mx;
+    int oi = __shfl_down(am, off, 64);
+    if (o > mx || (o == mx && oi < am)) { mx = o; am = oi; }
   }
-  if ((tid & 63) == 0) red[tid >> 6] = mx;
+  if ((tid & 63) == 0) { red[tid >> 6] = mx; redi[tid >> 6] = am; }
   __syncthreads();
   if (tid < 8) {
-    float v = (tid < (blockDim.x >> 6)) ? red[tid] : -3.0e38f;
+    const bool live = tid < (blockDim.x >> 6);
+    float v = live ? red[tid] : -3.0e38f;
+    int vi = live ? redi[tid] : 0;
     for (int off = 4; off > 0; off >>= 1) {
       float o = __shfl_down(v, off, 64);
-      v = o > v ? o : v;
+      int oi = __shfl_down(vi, off, 64);
+      if (o > v || (o == v && oi < vi)) { v = o; vi = oi; }
     }
-    if (tid == 0) red[0] = v;
+    if (tid == 0) { red[0] = v; redi[0] = vi; }
   }
   __syncthreads();
   mx = red[0];
+  const int amx = redi[0];
   __syncthreads();
 
   // 2) exp-sum
@@ -67,7 +79,12 @@ __global__ void softmax_xent_fwd_k(const __bf16* __restrict__ logits,
                   : (__bf16)0.f;
   if (tid == 0) {
     int lab = (int)labels[b];
-    loss[b] = __logf(sum) - ((float)row[lab] - mx);
+    float l = __logf(sum) - ((float)row[lab] - mx);
+    loss[b] = l;
+    if (stats) {
+      atomicAdd(&stats[0], l);
+      if (amx != lab) atomicAdd(&stats[1], 1.f);
+    }
   }
 }
 
@@ -197,10 +214,11 @@ hipError_t launch_tr_probe(const int* offs, float* out, int fill_mode,
 
 hipError_t launch_softmax_xent_fwd(const void* logits, const long* labels,
                                    void* probs, float* loss, int B,
-                                   int Cp, int nCls, hipStream_t stream) {
+                                   int Cp, int nCls, float* stats,
+                                   hipStream_t stream) {
   hipLaunchKernelGGL(softmax_xent_fwd_k, dim3(B), dim3(256), 0, stream,
                      (const __bf16*)logits, labels, (__bf16*)probs, loss,
-                     Cp, nCls);
+                     Cp, nCls, stats);
   return hipGetLastError();
 }
 
