@@ -98,7 +98,14 @@ def test_halo_pad():
 # ---------------------------------------------------------------------------
 
 def conv_case(B, H, W, Cin, K, R, stride, pad, groups=1, relu=False,
-              atol=0.06):
+              atol=0.02):
+    # Tolerance tier: every compared path here is
+    # bf16 inputs with FP32 ACCUMULATION against a reference computed
+    # from identically bf16-quantized operands, so the only divergence
+    # is accumulation order + one bf16 output rounding (~4e-3 ulp);
+    # 2e-2 on a max-normalized error leaves ~4x headroom (margins
+    # printed below, checked on hardware) while still catching any
+    # systematic bias the old 6e-2 would have hidden.
     from theanompi_amd.models import layers as L
     from theanompi_amd.ops import functional as OF
     torch.manual_seed(0)
@@ -129,7 +136,8 @@ def conv_case(B, H, W, Cin, K, R, stride, pad, groups=1, relu=False,
                        groups, relu)
     got = y.float().cpu()[..., [g * conv.plan.Kgp + k for g in range(groups)
                                 for k in range(conv.plan.Kg)]]
-    assert rel_err(got, yr) < atol, f"fwd rel err {rel_err(got, yr)}"
+    e_fwd = rel_err(got, yr)
+    assert e_fwd < atol, f"fwd rel err {e_fwd}"
 
     # backward
     dy = torch.randn_like(yr)
@@ -151,8 +159,8 @@ def conv_case(B, H, W, Cin, K, R, stride, pad, groups=1, relu=False,
     yr2 = OF.conv2d_ref(xr2, wl2, bl.float(), (stride, stride), (pad, pad),
                         groups, relu)
     yr2.backward(dy)
-    assert rel_err(dw_log, wl2.grad) < atol, \
-        f"wgrad rel err {rel_err(dw_log, wl2.grad)}"
+    e_wg = rel_err(dw_log, wl2.grad)
+    assert e_wg < atol, f"wgrad rel err {e_wg}"
     # bias grad: gather padded layout
     dbg = torch.cat([conv.bias.grad[g * conv.plan.Kgp:
                                     g * conv.plan.Kgp + conv.plan.Kg]
@@ -165,15 +173,16 @@ def conv_case(B, H, W, Cin, K, R, stride, pad, groups=1, relu=False,
     # dx
     if groups == 1:
         dx_got = xg.grad.float().cpu()[..., :Cin]
-        assert rel_err(dx_got, xr.grad) < atol, \
-            f"dgrad rel err {rel_err(dx_got, xr.grad)}"
     else:
         cg_p = cphys // groups
         cg_l = Cin // groups
         parts = [xg.grad.float().cpu()[..., g * cg_p:g * cg_p + cg_l]
                  for g in range(groups)]
         dx_got = torch.cat(parts, dim=-1)
-        assert rel_err(dx_got, xr.grad) < atol
+    e_dx = rel_err(dx_got, xr.grad)
+    assert e_dx < atol, f"dgrad rel err {e_dx}"
+    print(f"[margins atol={atol}] fwd {e_fwd:.4f} wgrad {e_wg:.4f} "
+          f"dgrad {e_dx:.4f}", flush=True)
 
 
 def test_conv_3x3_s1():

@@ -191,9 +191,45 @@ def tune_shape(row, iters, table, report):
     torch.cuda.empty_cache()
 
 
+def model_rows(name):
+    """Instantiate a zoo model at its bench batch size and enumerate
+    EVERY conv/linear layer's exact GEMM shape (deduped) — full
+    coverage instead of hand-picked representatives."""
+    import importlib
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from bench import MODELS  # repo-root bench.py
+    from theanompi_amd.models import layers as L
+    modname, clsname, batch = MODELS[name]
+    cls = getattr(importlib.import_module(modname), clsname)
+    model = cls({"rank": 0, "size": 1, "device": "cuda:0",
+                 "batch_size": batch, "n_iters": 2})
+    rows, seen = [], set()
+    for lname, mod in model.module.named_modules():
+        if not isinstance(mod, (L.Conv2d, L.Linear)):
+            continue
+        p = getattr(mod, "plan", None)
+        if p is None:
+            continue
+        cin = getattr(mod, "Cg_logical", p.Cg) * p.G
+        k = p.Kg * p.G
+        key = (p.NHWC, p.R, p.stride, p.pad, p.G, k)
+        if key in seen:
+            continue
+        seen.add(key)
+        rows.append((f"{name}.{lname}", p.NHWC[0], p.NHWC[1], p.NHWC[2],
+                     cin, k, p.R, p.stride[0], p.pad[0], p.G))
+    del model
+    torch.cuda.empty_cache()
+    return rows
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="all", choices=["all"] + sorted(SHAPES))
+    ap.add_argument("--models", default="",
+                    help="comma list of zoo models: tune every layer "
+                         "shape of each (overrides --model)")
     ap.add_argument("--iters", type=int, default=30)
     ap.add_argument("--out", default="gpurun_out/conv_tuned.json")
     ap.add_argument("--report", default="gpurun_out/autotune_report.md")
@@ -204,7 +240,16 @@ def main():
     table = {}
     report = ["| stage | gemm | winner | default -> best | TF | all |",
               "|---|---|---|---|---|---|"]
-    keys = sorted(SHAPES) if args.model == "all" else [args.model]
+    if args.models:
+        rows = []
+        for name in args.models.split(","):
+            rows.extend(model_rows(name.strip()))
+        for row in rows:
+            tune_shape(row, args.iters, table, report)
+            print(report[-1], flush=True)
+        keys = []
+    else:
+        keys = sorted(SHAPES) if args.model == "all" else [args.model]
     for k in keys:
         for row in SHAPES[k]:
             tune_shape(row, args.iters, table, report)
