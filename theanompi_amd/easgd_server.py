@@ -55,9 +55,15 @@ def run():
     last_val = 0
     epoch = start_epoch
     t0 = time.time()
+    rr = 0          # rotating scan start: fairness under 7 workers
     while workers:
         progressed = False
-        for w in sorted(workers):
+        # rotate the poll order each sweep so a low rank with
+        # back-to-back requests cannot starve the high ranks (the
+        # sorted scan always served rank 1 first)
+        order = sorted(workers)
+        rr = (rr + 1) % max(len(order), 1)
+        for w in order[rr:] + order[:rr]:
             key = f"ea/req/{w}/{seq[w]}"
             if not store.check([key]):
                 continue
