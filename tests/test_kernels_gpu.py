@@ -414,6 +414,78 @@ def test_sgd_step_kernel():
     assert rel_err(w2, wr) < 1e-5
 
 
+def test_conv_variants_numerics():
+    """Every tuned kernel variant computes the same conv as variant 0
+    (fwd within bf16 rounding of each other; wgrad fp32 within
+    accumulation-order noise)."""
+    from theanompi_amd.models import layers as L
+    torch.manual_seed(11)
+    d = dev()
+    B, H, W, Cin, K = 8, 14, 14, 128, 128
+    conv = L.Conv2d((H, W, Cin), K, 3, stride=1, pad=1, relu=False,
+                    device=str(d), batch=B)
+    x = torch.randn(B, H, W, Cin, device=d).to(torch.bfloat16)
+    x.requires_grad_()
+    dy = torch.randn(B, H, W, conv.plan.Kphys, device=d) \
+        .to(torch.bfloat16)
+    M = B * conv.plan.P * conv.plan.Q
+
+    def run(vf, vw):
+        conv._v_fwd, conv._v_M, conv._v_wgrad = vf, M, (vw, 0)
+        for p2 in conv.plan.dgrad_plans:
+            if p2 is not None:
+                p2["variant"], p2["vN"] = vf, B
+        x.grad = None
+        conv.weight.grad = None
+        conv.bias.grad = None
+        y = conv(x)
+        y.backward(dy)
+        return (y.detach().float(), conv.weight.grad.clone(),
+                x.grad.float().clone())
+
+    y0, dw0, dx0 = run(0, 0)
+    for vf in (1, 2, 3, 5, 6, 7, 8, 9, 10):
+        y, _dw, dx = run(vf, 0)
+        assert rel_err(y, y0) < 5e-3, (vf, rel_err(y, y0))
+        assert rel_err(dx, dx0) < 5e-3, (vf, rel_err(dx, dx0))
+    for vw in (1, 2, 3, 4):
+        _y, dw, _dx = run(0, vw)
+        assert rel_err(dw, dw0) < 5e-3, (vw, rel_err(dw, dw0))
+
+
+def test_sgd_cached_descriptor_lr_invalidate():
+    """adjust_hyperp mid-run: the persistent-grads cached descriptor
+    must be rebuilt after opt.invalidate() so the new LR takes effect
+    (VERDICT r1 weak #8)."""
+    from theanompi_amd.lib.opt import ParamRec, SGDMomentum
+    torch.manual_seed(7)
+    d = dev()
+    masters = [torch.randn(257, device=d, requires_grad=True),
+               torch.randn(64, 3, device=d, requires_grad=True)]
+    recs = [ParamRec(p, f"p{i}") for i, p in enumerate(masters)]
+    opt = SGDMomentum(recs, lr=0.1, momentum=0.9, weight_decay=0.0)
+    opt.make_persistent()
+    ref_w = [p.detach().clone() for p in masters]
+    ref_v = [torch.zeros_like(p) for p in masters]
+
+    def ref_step(lr):
+        for i in range(len(ref_w)):
+            ref_v[i] = 0.9 * ref_v[i] - lr * grads[i]
+            ref_w[i] += ref_v[i]
+
+    grads = [torch.randn_like(p) for p in masters]
+    for r, g in zip(recs, grads):
+        r.master.grad.copy_(g)
+    opt.step()
+    ref_step(0.1)
+    opt.lr = 0.01            # adjust_hyperp path
+    opt.invalidate()
+    opt.step()
+    ref_step(0.01)
+    for r, w in zip(recs, ref_w):
+        assert rel_err(r.master.data, w) < 1e-5
+
+
 def test_gather_bf16():
     from theanompi_amd import ops as O
     C = O.ext()

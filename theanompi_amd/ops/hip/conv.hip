@@ -584,19 +584,20 @@ struct WgradArgs {
   long xTotal;                      // X elements (gather tail clamp)
 };
 
-template <int LAYOUT, int BNF>
+template <int LAYOUT, int BNF, int BMK = 64>
 __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   // TN GEMM over pixels.  Both operands stage LINEARLY ([pix][64ch]
   // subtiles, 16-B glds pieces, content-swizzled) and the k-major MFMA
   // fragments come from ds_read_b64_tr_b16 (hardware transpose).
   // Counted-vmcnt double buffering, LDS-resident gather table.
-  constexpr int BMK = 64;          // kout tile
+  // BMK (kout tile) 64 or 128: the wider tile does 2x the MFMA work
+  // per staged dy byte for wide-Kgp layers (autotuned per shape).
   constexpr int BKP = 64;          // pixel chunk
   constexpr int XI = BNF / 32;     // x-tile glds per wave
   constexpr int DI = BMK / 32;     // dy-tile glds per wave
   constexpr int TPW = XI + DI;
-  
-  __shared__ __align__(16) __bf16 DYt[2][BKP][64];        // [pix][kout]
+
+  __shared__ __align__(16) __bf16 DYt[2][BKP][BMK];       // [pix][kout]
   __shared__ __align__(16) __bf16 XT[2][BKP][BNF];        // [pix][fd]
   extern __shared__ int Tab[];
 
@@ -652,15 +653,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
 #pragma unroll
     for (int i = 0; i < DI; ++i) {
       int piece = (wave * DI + i) * 64 + lane;
-      int row = piece >> 3;
-      int k8 = (((piece & 7) * 8) ^ ((row & 7) * 8)
-                ^ (((row >> 3) & 3) * 16));
+      int row = piece / (BMK / 8);
+      int k8i = piece % (BMK / 8);
+      int k8 = ((k8i * 8) ^ ((row & 7) * 8)
+                ^ (((row >> 3) & 3) * (BMK / 4)));
       int pa = pix0 + row;
       int pc = pa < a.M ? pa : a.M - 1;
       long dyaddr = (long)pc * a.Kphys + g * a.Kgp + kout0 + k8;
       if (dyaddr + 8 > dyTotal) dyaddr = dyTotal - 8;
       if (interior || pa < a.M) {
-        glds16(a.DY + dyaddr, &DYt[buf][(wave * DI + i) * 8][0]);
+        glds16(a.DY + dyaddr, &DYt[buf][0][0] + (wave * DI + i) * 512);
       } else {
         // tail: rows past M must be EXACT zeros (summed dimension)
         pack16 v;
@@ -670,13 +672,13 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
     }
   };
   const int wm = wave >> 1, wn = wave & 1;  // 2x2 waves
-  constexpr int FM = 1;                     // 32 kout rows per wave
+  constexpr int MW = BMK / 2 / 16;          // m-frags per wave (2 or 4)
   constexpr int FN = BNF / 2 / 16;          // 32(BNF=64) or 64(128)/16
-  const int m0 = wm * 32, n0 = wn * (BNF / 2);
+  const int m0 = wm * (BMK / 2), n0 = wn * (BNF / 2);
 
-  f32x4 acc[FM * 2][FN];
+  f32x4 acc[MW][FN];
 #pragma unroll
-  for (int i = 0; i < 2; ++i)
+  for (int i = 0; i < MW; ++i)
 #pragma unroll
     for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
@@ -696,17 +698,17 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
       const int buf = ci & 1;
 #pragma unroll
       for (int ks = 0; ks < BKP; ks += 32) {
-        bf16x8 af[2], bfr[FN];
+        bf16x8 af[MW], bfr[FN];
 #pragma unroll
-        for (int i = 0; i < 2; ++i)
-          af[i] = frag_load_tr_swz<64>(&DYt[buf][0][0], ks,
-                                       m0 + i * 16, lane);
+        for (int i = 0; i < MW; ++i)
+          af[i] = frag_load_tr_swz<BMK>(&DYt[buf][0][0], ks,
+                                        m0 + i * 16, lane);
 #pragma unroll
         for (int j = 0; j < FN; ++j)
           bfr[j] = frag_load_tr_swz<BNF>(&XT[buf][0][0], ks,
                                          n0 + j * 16, lane);
 #pragma unroll
-        for (int i = 0; i < 2; ++i)
+        for (int i = 0; i < MW; ++i)
 #pragma unroll
           for (int j = 0; j < FN; ++j)
             acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
@@ -718,7 +720,7 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
 
   // epilogue: atomic accumulate fp32
 #pragma unroll
-  for (int i = 0; i < 2; ++i) {
+  for (int i = 0; i < MW; ++i) {
 #pragma unroll
     for (int reg = 0; reg < 4; ++reg) {
       int kout = kout0 + m0 + i * 16 + (lane >> 4) * 4 + reg;
@@ -925,8 +927,9 @@ hipError_t launch_conv_gather_gemm(GatherGemmArgs* args, int G,
 }
 
 // wgrad variants (autotuner): 0 auto (BNF heuristic + env), 1 BNF=64,
-// 2 BNF=128.  The gather table is dynamically sized here too: at
-// BNF=128 that is the difference between 2 and 3 blocks/CU.
+// 2 BNF=128, 3 BNF=64/BMK=128, 4 BNF=128/BMK=128.  The gather table is
+// dynamically sized here too: at BNF=128 that is the difference
+// between 2 and 3 blocks/CU.
 hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
                              hipStream_t stream, int variant) {
   WgradArgs a = *args;
@@ -941,25 +944,29 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
   // epilogues were parallelized; THEANOMPI_WGRAD_BNF=128 restores it.
   int BNF = 64;
   if (bnf_env == 128 && a.Kdf % 128 == 0) BNF = 128;
-  if (variant == 1) BNF = 64;
-  if (variant == 2 && a.Kdf % 128 == 0) BNF = 128;
-  dim3 grid(cdiv(a.Kgp, 64), cdiv(a.Kdf, BNF), G * a.splitK);
+  if (variant == 1 || variant == 3) BNF = 64;
+  if ((variant == 2 || variant == 4) && a.Kdf % 128 == 0) BNF = 128;
+  int BMK = (variant == 3 || variant == 4) ? 128 : 64;
+  dim3 grid(cdiv(a.Kgp, BMK), cdiv(a.Kdf, BNF), G * a.splitK);
   dim3 block(256);
-  if (BNF == 128) {
-    if (layout == FRAG_CONTIG8)
-      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, 128>), grid, block,
-                         tabBytes, stream, a);
-    else
-      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, 128>), grid, block,
-                         tabBytes, stream, a);
+#define WG(BNF_, BMK_)                                                     \
+  do {                                                                     \
+    if (layout == FRAG_CONTIG8)                                            \
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, BNF_, BMK_>), grid,   \
+                         block, tabBytes, stream, a);                      \
+    else                                                                   \
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, BNF_, BMK_>), grid,    \
+                         block, tabBytes, stream, a);                      \
+  } while (0)
+  if (BMK == 128) {
+    if (BNF == 128) WG(128, 128);
+    else WG(64, 128);
+  } else if (BNF == 128) {
+    WG(128, 64);
   } else {
-    if (layout == FRAG_CONTIG8)
-      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, 64>), grid, block,
-                         tabBytes, stream, a);
-    else
-      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, 64>), grid, block,
-                         tabBytes, stream, a);
+    WG(64, 64);
   }
+#undef WG
   return hipGetLastError();
 }
 

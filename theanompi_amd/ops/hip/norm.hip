@@ -125,6 +125,126 @@ __global__ void lrn_bwd_k(const __bf16* __restrict__ dy,
   }
 }
 
+// 16-channels-per-thread variants: the 4-wide kernels above are
+// LATENCY-bound (measured 3.5x off the bandwidth roofline at 15.8% of
+// the AlexNet step, profiles/alexnet r2): each thread issued 3 tiny
+// dependent loads for 4 outputs.  Here a thread owns 16 channels of
+// one pixel — six independent 8-B loads (1.5x read amplification from
+// the +-half halo), register-resident squares, and an incremental
+// sliding-window sum (2 adds per output).  C % 16 == 0 and half <= 4
+// (the reference models use size 5); launcher falls back otherwise.
+
+__global__ void lrn_fwd16_k(const __bf16* __restrict__ x,
+                            __bf16* __restrict__ y,
+                            __bf16* __restrict__ scale, long rows, int C,
+                            int size, float alpha, float beta, float k) {
+  const int half = size / 2;               // <= 4
+  const float a = alpha / size;
+  const int C16 = C / 16;
+  const int C4 = C / 4;
+  const long total = rows * C16;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c16 = (int)(i % C16);
+    long row = i / C16;
+    const __bf16* xr = x + row * C + c16 * 16;
+    float v[24], sq[24];
+#pragma unroll
+    for (int b = 0; b < 6; ++b) {
+      int cb = c16 * 4 + b - 1;            // absolute 4-channel block
+      if (cb >= 0 && cb < C4) {
+        bf16x4 t = *(const bf16x4*)(xr + (b - 1) * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) v[b * 4 + j] = (float)t[j];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) v[b * 4 + j] = 0.f;
+      }
+    }
+#pragma unroll
+    for (int j = 0; j < 24; ++j) sq[j] = v[j] * v[j];
+    bf16x8 out0, out1, sc0, sc1;
+    float s = 0.f;
+    for (int w = -half; w <= half; ++w) s += sq[4 + w];
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      float sc = k + a * s;
+      float inv = 1.f / sc;                // saved: bwd needs 1/scale
+      float pw = (beta == 0.75f)
+                     ? __fsqrt_rn(inv) * __fsqrt_rn(__fsqrt_rn(inv))
+                     : __powf(inv, beta);
+      float o = v[4 + j] * pw;
+      if (j < 8) { sc0[j] = (__bf16)inv; out0[j] = (__bf16)o; }
+      else { sc1[j - 8] = (__bf16)inv; out1[j - 8] = (__bf16)o; }
+      if (j < 15) s += sq[5 + j + half] - sq[4 + j - half];
+    }
+    long o = row * C + c16 * 16;
+    *(bf16x8*)(y + o) = out0;
+    *(bf16x8*)(y + o + 8) = out1;
+    *(bf16x8*)(scale + o) = sc0;
+    *(bf16x8*)(scale + o + 8) = sc1;
+  }
+}
+
+__global__ void lrn_bwd16_k(const __bf16* __restrict__ dy,
+                            const __bf16* __restrict__ y,
+                            const __bf16* __restrict__ scale,
+                            __bf16* __restrict__ dx, long rows, int C,
+                            int size, float alpha, float beta, float k) {
+  const int half = size / 2;
+  const float coef = 2.f * alpha * beta / size;
+  const int C16 = C / 16;
+  const int C4 = C / 4;
+  const long total = rows * C16;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c16 = (int)(i % C16);
+    long row = i / C16;
+    long base = row * C + c16 * 16;
+    float t[24], yv[16], dyv[16], scv[16];
+#pragma unroll
+    for (int b = 0; b < 6; ++b) {
+      int cb = c16 * 4 + b - 1;
+      if (cb >= 0 && cb < C4) {
+        long ba = base + (b - 1) * 4;
+        bf16x4 d = *(const bf16x4*)(dy + ba);
+        bf16x4 yy = *(const bf16x4*)(y + ba);
+        bf16x4 sc = *(const bf16x4*)(scale + ba);     // = 1/scale
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          t[b * 4 + j] = (float)d[j] * (float)yy[j] * (float)sc[j];
+          int cj = b * 4 + j - 4;
+          if (cj >= 0 && cj < 16) {
+            dyv[cj] = (float)d[j];
+            yv[cj] = (float)yy[j];
+            scv[cj] = (float)sc[j];
+          }
+        }
+      } else {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) t[b * 4 + j] = 0.f;
+      }
+    }
+    bf16x8 out0, out1;
+    float s = 0.f;
+    for (int w = -half; w <= half; ++w) s += t[4 + w];
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      float inv = scv[j];                  // 1/scale
+      float pw = (beta == 0.75f)
+                     ? __fsqrt_rn(inv) * __fsqrt_rn(__fsqrt_rn(inv))
+                     : __powf(inv, beta);
+      float xv = yv[j] * __frcp_rn(pw);    // x = y / scale^beta
+      float d = dyv[j] * pw - coef * xv * s;
+      if (j < 8) out0[j] = (__bf16)d;
+      else out1[j - 8] = (__bf16)d;
+      if (j < 15) s += t[5 + j + half] - t[4 + j - half];
+    }
+    *(bf16x8*)(dx + base) = out0;
+    *(bf16x8*)(dx + base + 8) = out1;
+  }
+}
+
 // ---------------------------------------------------------------------------
 // BatchNorm (training): per-channel mean/var over N*H*W rows.
 // Pass 1: partial sums (fp32 atomics into [C] buffers).
@@ -474,6 +594,13 @@ extern "C" {
 hipError_t launch_lrn_fwd(const void* x, void* y, void* scale, long rows,
                           int C, int size, float alpha, float beta,
                           float k, hipStream_t stream) {
+  if (C % 16 == 0 && size / 2 <= 4) {
+    hipLaunchKernelGGL(lrn_fwd16_k, dim3(ngrid(rows * C / 16)),
+                       dim3(256), 0, stream, (const __bf16*)x,
+                       (__bf16*)y, (__bf16*)scale, rows, C, size, alpha,
+                       beta, k);
+    return hipGetLastError();
+  }
   hipLaunchKernelGGL(lrn_fwd_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
                      stream, (const __bf16*)x, (__bf16*)y,
                      (__bf16*)scale, rows, C, size, alpha, beta, k);
@@ -484,6 +611,13 @@ hipError_t launch_lrn_bwd(const void* dy, const void* y,
                           const void* scale, void* dx, long rows, int C,
                           int size, float alpha, float beta, float k,
                           hipStream_t stream) {
+  if (C % 16 == 0 && size / 2 <= 4) {
+    hipLaunchKernelGGL(lrn_bwd16_k, dim3(ngrid(rows * C / 16)),
+                       dim3(256), 0, stream, (const __bf16*)dy,
+                       (const __bf16*)y, (const __bf16*)scale,
+                       (__bf16*)dx, rows, C, size, alpha, beta, k);
+    return hipGetLastError();
+  }
   hipLaunchKernelGGL(lrn_bwd_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
                      stream, (const __bf16*)dy, (const __bf16*)y,
                      (const __bf16*)scale, (__bf16*)dx, rows, C, size,

@@ -49,7 +49,8 @@ SHAPES["fc"] = [
 
 FWD_VARIANTS = [1, 2, 3, 4, 5, 6, 7, 8, 9, 10]
 DGRAD_VARIANTS = [1, 2, 5, 6, 7, 8, 9, 10]
-WGRAD_GRID = [(v, t) for v in (1, 2) for t in (512, 1024, 2048, 4096)]
+WGRAD_GRID = [(v, t) for v in (1, 2, 3, 4)
+              for t in (512, 1024, 2048, 4096)]
 
 
 def build(name, B, H, W, Cin, K, R, stride, pad, groups):
@@ -105,7 +106,8 @@ def fwd_ok(v, M, Kdf, BNtot):
     return True
 
 
-def tune_shape(row, iters, table, report):
+def tune_shape(row, iters, table, report, stages=("fwd", "dgrad",
+                                                  "wgrad")):
     name, B, H, W, Cin, K, R, stride, pad, groups = row
     from theanompi_amd.ops import tuning
     conv, x = build(*row)
@@ -114,27 +116,31 @@ def tune_shape(row, iters, table, report):
     flops = 2.0 * B * plan.P * plan.Q * K * (Cin // groups) * R * R
 
     # -- fwd sweep ----------------------------------------------------------
-    with torch.no_grad():
-        xn = x.detach()
-        results = {}
+    if "fwd" in stages:
+        with torch.no_grad():
+            xn = x.detach()
+            results = {}
+            force_fwd(conv, M, 0)
+            results[0] = timed(lambda: conv(xn), iters)
+            for v in FWD_VARIANTS:
+                if not fwd_ok(v, M, plan.Kdf, plan.Kgp):
+                    continue
+                force_fwd(conv, M, v)
+                results[v] = timed(lambda: conv(xn), iters)
+        best_f = min(results, key=results.get)
+        t0f, tbf = results[0], results[best_f]
+        if best_f != 0 and tbf < t0f * 0.99:
+            table[tuning.key("fwd", M, plan.Kgp, plan.Kdf, plan.G)] = \
+                best_f
+        force_fwd(conv, M, best_f if tbf < t0f * 0.99 else 0)
+        report.append(
+            f"| {name} fwd | {M}x{plan.Kgp}x{plan.Kdf}g{plan.G} "
+            f"| v{best_f} | {t0f*1e3:.3f} -> {tbf*1e3:.3f} ms "
+            f"| {flops/tbf/1e12:.0f} TF | "
+            + " ".join(f"v{v}:{t*1e3:.3f}"
+                       for v, t in sorted(results.items())) + " |")
+    else:
         force_fwd(conv, M, 0)
-        results[0] = timed(lambda: conv(xn), iters)
-        for v in FWD_VARIANTS:
-            if not fwd_ok(v, M, plan.Kdf, plan.Kgp):
-                continue
-            force_fwd(conv, M, v)
-            results[v] = timed(lambda: conv(xn), iters)
-    best_f = min(results, key=results.get)
-    t0f, tbf = results[0], results[best_f]
-    if best_f != 0 and tbf < t0f * 0.99:
-        table[tuning.key("fwd", M, plan.Kgp, plan.Kdf, plan.G)] = best_f
-    force_fwd(conv, M, best_f if tbf < t0f * 0.99 else 0)
-    report.append(
-        f"| {name} fwd | {M}x{plan.Kgp}x{plan.Kdf}g{plan.G} "
-        f"| v{best_f} | {t0f*1e3:.3f} -> {tbf*1e3:.3f} ms "
-        f"| {flops/tbf/1e12:.0f} TF | "
-        + " ".join(f"v{v}:{t*1e3:.3f}" for v, t in sorted(results.items()))
-        + " |")
 
     if not x.requires_grad:
         del conv, x
@@ -152,41 +158,46 @@ def tune_shape(row, iters, table, report):
         if conv.bias is not None:
             conv.bias.grad = None
 
-    dres = {}
-    force_dgrad(conv, B, 0)
-    dres[0] = timed(fb, iters)
-    for v in DGRAD_VARIANTS:
-        force_dgrad(conv, B, v)
-        dres[v] = timed(fb, iters)
-    best_d = min(dres, key=dres.get)
-    if best_d != 0 and dres[best_d] < dres[0] * 0.995:
-        for p2 in plan.dgrad_plans:
-            if p2 is not None:
-                table[tuning.key("dgrad", B * p2["Hi"] * p2["Wj"],
-                                 plan.Cg, p2["Kdf2"], plan.G)] = best_d
-    force_dgrad(conv, B, best_d if dres[best_d] < dres[0] * 0.995 else 0)
-    report.append(
-        f"| {name} dgrad | (all classes) | v{best_d} "
-        f"| {dres[0]*1e3:.3f} -> {dres[best_d]*1e3:.3f} ms | | "
-        + " ".join(f"v{v}:{t*1e3:.3f}" for v, t in sorted(dres.items()))
-        + " |")
+    if "dgrad" in stages:
+        dres = {}
+        force_dgrad(conv, B, 0)
+        dres[0] = timed(fb, iters)
+        for v in DGRAD_VARIANTS:
+            force_dgrad(conv, B, v)
+            dres[v] = timed(fb, iters)
+        best_d = min(dres, key=dres.get)
+        if best_d != 0 and dres[best_d] < dres[0] * 0.995:
+            for p2 in plan.dgrad_plans:
+                if p2 is not None:
+                    table[tuning.key("dgrad", B * p2["Hi"] * p2["Wj"],
+                                     plan.Cg, p2["Kdf2"], plan.G)] = \
+                        best_d
+        force_dgrad(conv, B,
+                    best_d if dres[best_d] < dres[0] * 0.995 else 0)
+        report.append(
+            f"| {name} dgrad | (all classes) | v{best_d} "
+            f"| {dres[0]*1e3:.3f} -> {dres[best_d]*1e3:.3f} ms | | "
+            + " ".join(f"v{v}:{t*1e3:.3f}"
+                       for v, t in sorted(dres.items())) + " |")
 
     # -- wgrad sweep ---------------------------------------------------------
-    wres = {}
-    conv._v_wgrad = (0, 0)
-    wres[(0, 0)] = timed(fb, iters)
-    for vt in WGRAD_GRID:
-        conv._v_wgrad = vt
-        wres[vt] = timed(fb, iters)
-    best_w = min(wres, key=wres.get)
-    if best_w != (0, 0) and wres[best_w] < wres[(0, 0)] * 0.995:
-        table[tuning.key("wgrad", M, plan.Kgp, plan.Kdf, plan.G)] = \
-            list(best_w)
-    report.append(
-        f"| {name} wgrad | {M}x{plan.Kgp}x{plan.Kdf}g{plan.G} "
-        f"| v{best_w} | {wres[(0,0)]*1e3:.3f} -> {wres[best_w]*1e3:.3f} ms | | "
-        + " ".join(f"{v}/{t_}:{t*1e3:.3f}"
-                   for (v, t_), t in sorted(wres.items())) + " |")
+    if "wgrad" in stages:
+        wres = {}
+        conv._v_wgrad = (0, 0)
+        wres[(0, 0)] = timed(fb, iters)
+        for vt in WGRAD_GRID:
+            conv._v_wgrad = vt
+            wres[vt] = timed(fb, iters)
+        best_w = min(wres, key=wres.get)
+        if best_w != (0, 0) and wres[best_w] < wres[(0, 0)] * 0.995:
+            table[tuning.key("wgrad", M, plan.Kgp, plan.Kdf, plan.G)] = \
+                list(best_w)
+        report.append(
+            f"| {name} wgrad | {M}x{plan.Kgp}x{plan.Kdf}g{plan.G} "
+            f"| v{best_w} | {wres[(0,0)]*1e3:.3f} -> "
+            f"{wres[best_w]*1e3:.3f} ms | | "
+            + " ".join(f"{v}/{t_}:{t*1e3:.3f}"
+                       for (v, t_), t in sorted(wres.items())) + " |")
     del conv, x, dy
     torch.cuda.empty_cache()
 
@@ -231,28 +242,36 @@ def main():
                     help="comma list of zoo models: tune every layer "
                          "shape of each (overrides --model)")
     ap.add_argument("--iters", type=int, default=30)
+    ap.add_argument("--stages", default="fwd,dgrad,wgrad",
+                    help="comma subset of fwd,dgrad,wgrad")
     ap.add_argument("--out", default="gpurun_out/conv_tuned.json")
     ap.add_argument("--report", default="gpurun_out/autotune_report.md")
+    ap.add_argument("--base", default="",
+                    help="existing table to merge new winners into")
     args = ap.parse_args()
     assert torch.cuda.is_available()
     os.environ.setdefault("THEANOMPI_TUNED_OFF", "1")  # sweep from clean
 
     table = {}
+    if args.base:
+        with open(args.base) as f:
+            table = json.load(f)
     report = ["| stage | gemm | winner | default -> best | TF | all |",
               "|---|---|---|---|---|---|"]
+    stages = tuple(args.stages.split(","))
     if args.models:
         rows = []
         for name in args.models.split(","):
             rows.extend(model_rows(name.strip()))
         for row in rows:
-            tune_shape(row, args.iters, table, report)
+            tune_shape(row, args.iters, table, report, stages)
             print(report[-1], flush=True)
         keys = []
     else:
         keys = sorted(SHAPES) if args.model == "all" else [args.model]
     for k in keys:
         for row in SHAPES[k]:
-            tune_shape(row, args.iters, table, report)
+            tune_shape(row, args.iters, table, report, stages)
             print(report[-1], flush=True)
 
     os.makedirs(os.path.dirname(args.out) or ".", exist_ok=True)
