@@ -118,6 +118,8 @@ hipError_t launch_mfma_probe(const void*, const void*, float*, int,
 hipError_t launch_sgd_multi(const void*, int, long, float, int,
                             hipStream_t);
 hipError_t launch_gather_multi(const void*, int, long, hipStream_t);
+hipError_t launch_transpose_multi(const void*, int, int, int,
+                                  hipStream_t);
 hipError_t launch_tr_probe(const int*, float*, int, hipStream_t);
 hipError_t launch_splitk_finalize(float*, const float*, void*, long,
                                   int, int, hipStream_t);
@@ -156,6 +158,13 @@ struct GatherSegH {
   const int* map;
   void* dst;
   long n;
+};
+
+struct TransSegH {
+  const void* src;
+  void* dst;
+  int rows, cols;
+  long sstride, dstride;
 };
 
 // ---- helpers ---------------------------------------------------------------
@@ -906,6 +915,36 @@ static void gather_multi_run(torch::Tensor desc, long nseg, long max_n) {
                                 cur_stream()));
 }
 
+static std::vector<long> transpose_multi_make(
+    std::vector<torch::Tensor> srcs, std::vector<torch::Tensor> dsts,
+    std::vector<long> rows, std::vector<long> cols,
+    std::vector<long> sstrides, std::vector<long> dstrides,
+    torch::Tensor buf) {
+  // buf: caller-held uint8 tensor of n*sizeof(TransSegH); returns
+  // [max_rows, max_cols] for the run call
+  int n = (int)srcs.size();
+  std::vector<TransSegH> segs(n);
+  long mr = 1, mc = 1;
+  for (int i = 0; i < n; ++i) {
+    segs[i] = {srcs[i].data_ptr(), dsts[i].data_ptr(), (int)rows[i],
+               (int)cols[i], sstrides[i], dstrides[i]};
+    if (rows[i] > mr) mr = rows[i];
+    if (cols[i] > mc) mc = cols[i];
+  }
+  TORCH_CHECK(buf.numel() >= (long)(n * sizeof(TransSegH)));
+  hipMemcpyAsync(buf.data_ptr(), segs.data(), n * sizeof(TransSegH),
+                 hipMemcpyHostToDevice, cur_stream());
+  hipStreamSynchronize(cur_stream());
+  return {mr, mc};
+}
+
+static void transpose_multi_run(torch::Tensor desc, long nseg,
+                                long max_rows, long max_cols) {
+  CHECK_HIP(launch_transpose_multi(desc.data_ptr(), (int)nseg,
+                                   (int)max_rows, (int)max_cols,
+                                   cur_stream()));
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("halo_pad", &halo_pad_sym);
   m.def("halo_pad_lrtb", &halo_pad);
@@ -961,4 +1000,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_multi_run", &sgd_multi_run);
   m.def("gather_multi_make", &gather_multi_make);
   m.def("gather_multi_run", &gather_multi_run);
+  m.def("transpose_multi_make", &transpose_multi_make);
+  m.def("transpose_multi_run", &transpose_multi_run);
 }

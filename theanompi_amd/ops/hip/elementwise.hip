@@ -190,13 +190,43 @@ __global__ void u8_finish_k(const unsigned char* __restrict__ src,
 // dst[c][k] = src[k][c] for k < rows, c < cols; strides in elements.
 // 64(col)x32(row) tiles; 8-element vector loads, column gathers from LDS.
 
+__device__ __forceinline__ void transpose_tile_body(
+    const __bf16* __restrict__ src, __bf16* __restrict__ dst, int rows,
+    int cols, long sstride, long dstride, int r0, int c0, int t,
+    __bf16 (*tile)[72]);
+
 __global__ void transpose_bf16_k(const __bf16* __restrict__ src,
                                  __bf16* __restrict__ dst, int rows,
                                  int cols, long sstride, long dstride) {
   __shared__ __bf16 tile[32][72];    // +8 pad kills column-read conflicts
+  transpose_tile_body(src, dst, rows, cols, sstride, dstride,
+                      blockIdx.y * 32, blockIdx.x * 64, threadIdx.x,
+                      tile);
+}
+
+// one launch covers every dgrad-repack transpose of the step (the
+// per-layer launches were 46/step on GoogLeNet — pure launch latency)
+struct TransSeg {
+  const __bf16* src;
+  __bf16* dst;
+  int rows, cols;
+  long sstride, dstride;
+};
+
+__global__ void transpose_multi_k(const TransSeg* __restrict__ segs) {
+  __shared__ __bf16 tile[32][72];
+  const TransSeg s = segs[blockIdx.z];
   const int r0 = blockIdx.y * 32;
   const int c0 = blockIdx.x * 64;
-  const int t = threadIdx.x;
+  if (r0 >= s.rows || c0 >= s.cols) return;
+  transpose_tile_body(s.src, s.dst, s.rows, s.cols, s.sstride,
+                      s.dstride, r0, c0, threadIdx.x, tile);
+}
+
+__device__ __forceinline__ void transpose_tile_body(
+    const __bf16* __restrict__ src, __bf16* __restrict__ dst, int rows,
+    int cols, long sstride, long dstride, int r0, int c0, int t,
+    __bf16 (*tile)[72]) {
   {
     const int row = t >> 3;          // 0..31
     const int cp = (t & 7) * 8;      // 0..56
@@ -367,6 +397,15 @@ hipError_t launch_u8_finish(const unsigned char* src, void* dst,
   long t4 = total / 4;
   hipLaunchKernelGGL(u8_finish_k, dim3(ew_grid(t4)), dim3(256), 0, stream,
                      src, (__bf16*)dst, meanimg, t4, perImg, C, nLogC);
+  return hipGetLastError();
+}
+
+hipError_t launch_transpose_multi(const void* segs_dev, int nseg,
+                                  int max_rows, int max_cols,
+                                  hipStream_t stream) {
+  dim3 grid((max_cols + 63) / 64, (max_rows + 31) / 32, nseg);
+  hipLaunchKernelGGL(transpose_multi_k, grid, dim3(256), 0, stream,
+                     (const TransSeg*)segs_dev);
   return hipGetLastError();
 }
 
