@@ -55,7 +55,7 @@ hipError_t launch_conv_wgrad(const WgradArgs*, int G, int layout,
                              hipStream_t, int variant);
 hipError_t launch_bias_wgrad(const void*, float*, long, int, hipStream_t);
 hipError_t launch_halo_pad(const void*, void*, int, int, int, int, int, int,
-                           int, int, float, hipStream_t);
+                           int, int, float, long, hipStream_t);
 hipError_t launch_relu_fwd(const void*, void*, long, hipStream_t);
 hipError_t launch_relu_bwd_s(const void*, const void*, void*, long, int,
                              long, long, hipStream_t);
@@ -257,13 +257,23 @@ static int detect_layout() {
 
 static torch::Tensor halo_pad(torch::Tensor x, long t, long b, long l,
                               long r, double fill) {
-  check_bf16(x, "x");
+  // x may be a CHANNEL SLICE of a wider NHWC buffer (merged-branch
+  // outputs): inner dim contiguous, pixels a uniform srcC apart
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == torch::kBFloat16);
   TORCH_CHECK(x.dim() == 4, "x must be NHWC");
   auto N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   TORCH_CHECK(C % 4 == 0, "C must be a multiple of 4");
+  long srcC = C;
+  if (!x.is_contiguous()) {
+    TORCH_CHECK(x.stride(3) == 1 && x.stride(2) >= C
+                && x.stride(1) == x.stride(2) * W
+                && x.stride(0) == x.stride(1) * H,
+                "halo_pad: need a channel-slice layout");
+    srcC = x.stride(2);
+  }
   auto y = torch::empty({N, H + t + b, W + l + r, C}, x.options());
   CHECK_HIP(launch_halo_pad(x.data_ptr(), y.data_ptr(), N, H, W, C, t, b,
-                            l, r, (float)fill, cur_stream()));
+                            l, r, (float)fill, srcC, cur_stream()));
   return y;
 }
 

@@ -12,7 +12,9 @@
 __global__ void halo_pad_k(const __bf16* __restrict__ x,
                            __bf16* __restrict__ y, int N, int H, int W,
                            int C, int t, int b, int l, int r, float fill,
-                           long total4) {
+                           long srcC, long total4) {
+  // srcC: pixel stride of the source (== C, or wider when x is a
+  // channel slice of a merged-branch buffer)
   const int Hp = H + t + b, Wp = W + l + r;
   const __bf16 f = (__bf16)fill;
   for (long i4 = blockIdx.x * blockDim.x + threadIdx.x; i4 < total4;
@@ -27,7 +29,7 @@ __global__ void halo_pad_k(const __bf16* __restrict__ x,
     int h = hp - t, w = wp - l;
     pack8 v;
     if (h >= 0 && h < H && w >= 0 && w < W) {
-      v = *(const pack8*)(x + (((long)n * H + h) * W + w) * C + c);
+      v = *(const pack8*)(x + (((long)n * H + h) * W + w) * srcC + c);
     } else {
       __bf16* e = (__bf16*)&v;
       e[0] = f; e[1] = f; e[2] = f; e[3] = f;
@@ -332,11 +334,11 @@ extern "C" {
 
 hipError_t launch_halo_pad(const void* x, void* y, int N, int H, int W,
                            int C, int t, int b, int l, int r, float fill,
-                           hipStream_t stream) {
+                           long srcC, hipStream_t stream) {
   long total4 = (long)N * (H + t + b) * (W + l + r) * C / 4;
   hipLaunchKernelGGL(halo_pad_k, dim3(ew_grid(total4)), dim3(256), 0,
                      stream, (const __bf16*)x, (__bf16*)y, N, H, W, C, t,
-                     b, l, r, fill, total4);
+                     b, l, r, fill, srcC, total4);
   return hipGetLastError();
 }
 
