@@ -457,6 +457,39 @@ def test_conv_variants_numerics():
         assert rel_err(dw, dw0) < 5e-3, (vw, rel_err(dw, dw0))
 
 
+def test_inception_merged_equivalence():
+    """GPU inception (fused 3-way 1x1 GEMM + concat-writing branches)
+    matches the unmerged CPU reference given identical weights loaded
+    through the logical checkpoint mapping; branch weight grads
+    match too."""
+    from theanompi_amd.models.googlenet import _Inception
+    torch.manual_seed(3)
+    B, hw, cin = 4, 14, 64          # widths all multiples of 16: no pad
+    args = (hw, cin, 32, 48, 64, 16, 32, 32)
+    gi = _Inception(*args, dev(), B)
+    ci = _Inception(*args, "cpu", B)
+    assert gi.merged and not ci.merged
+    gp, cp = gi.logical_params(), ci.logical_params()
+    assert len(gp) == len(cp)
+    for (_n1, g1, _s1), (_n2, _g2, s2) in zip(gp, cp):
+        s2(g1().float().cpu())       # GPU weights -> CPU model
+    x = q(torch.randn(B, hw, hw, cin))
+    yg = gi(x.to(dev()).to(torch.bfloat16).requires_grad_())
+    yc = ci(x.clone().requires_grad_())
+    assert yg.shape == yc.shape
+    assert rel_err(yg, yc) < 2e-2
+    dy = q(torch.randn_like(yc))
+    yg.backward(dy.to(dev()).to(torch.bfloat16))
+    yc.backward(dy)
+    # merged flat weight rows map to branch [k, cin] 1x1 weights
+    for bi, branch in enumerate([ci.b1, ci.b3a, ci.b5a]):
+        off = gi.b135.branch_off[bi]
+        k = gi.b135.branch_k[bi]
+        g_merged = gi.b135.weight.grad[0, off:off + k, :cin].cpu()
+        g_ref = branch.weight.grad[:, :, 0, 0]
+        assert rel_err(g_merged, g_ref) < 2e-2, bi
+
+
 def test_sgd_cached_descriptor_lr_invalidate():
     """adjust_hyperp mid-run: the persistent-grads cached descriptor
     must be rebuilt after opt.invalidate() so the new LR takes effect

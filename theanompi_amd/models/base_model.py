@@ -298,9 +298,20 @@ class BaseModel:
         return (20, 40, 60)
 
     def _logical_param_list(self):
+        """Checkpoint entries in reference order.  A module with
+        ``owns_logical_params`` composes its whole subtree's entries
+        itself (e.g. the inception block keeps the unmerged per-branch
+        order even when its 1x1s run as one fused GEMM)."""
         out = []
-        for _name, mod in self.module.named_modules():
-            if isinstance(mod, L._Base):
+        owned = []
+        for name, mod in self.module.named_modules():
+            if any(name == p or name.startswith(p + ".")
+                   for p in owned):
+                continue
+            if getattr(mod, "owns_logical_params", False):
+                owned.append(name)
+                out.extend(mod.logical_params())
+            elif isinstance(mod, L._Base):
                 out.extend(mod.logical_params())
         return out
 

@@ -17,39 +17,68 @@ from theanompi_amd.ops import functional as OF
 
 
 class _Inception(nn.Module):
-    """branches: 1x1 | 1x1->3x3 | 1x1->5x5 | pool->1x1, concat on C."""
+    """branches: 1x1 | 1x1->3x3 | 1x1->5x5 | pool->1x1, concat on C.
+
+    GPU: the three same-input 1x1s (b1, 3x3-reduce, 5x5-reduce) run as
+    ONE fused GEMM (L.MultiConv1x1) — GoogLeNet-32b is launch-latency
+    bound, and this removes ~8 launches per module per step.  The block
+    owns its checkpoint order (owns_logical_params) so the entries stay
+    in the reference's unmerged per-branch sequence."""
+
+    owns_logical_params = True
 
     def __init__(self, hw, cin, c1, c3r, c3, c5r, c5, cp, dev, B):
         super().__init__()
         kw = dict(device=dev, batch=B, init=("he",), relu=True)
-        self.b1 = L.Conv2d((hw, hw, cin), c1, 1, **kw)
-        self.b3a = L.Conv2d((hw, hw, cin), c3r, 1, **kw)
-        self.b3b = L.Conv2d((hw, hw, self.b3a.out_shape[2]
-                             if hasattr(self.b3a, 'out_shape') else c3r),
-                            c3, 3, pad=1, **kw)
-        self.b5a = L.Conv2d((hw, hw, cin), c5r, 1, **kw)
-        self.b5b = L.Conv2d((hw, hw, self.b5a.out_shape[2]
-                             if hasattr(self.b5a, 'out_shape') else c5r),
-                            c5, 5, pad=2, **kw)
+        self.merged = str(dev).startswith("cuda")
+        if self.merged:
+            self.b135 = L.MultiConv1x1((hw, hw, cin), [c1, c3r, c5r],
+                                       **kw)
+            kp = self.b135.branch_kp
+            w1, w3r, w5r = kp
+        else:
+            self.b1 = L.Conv2d((hw, hw, cin), c1, 1, **kw)
+            self.b3a = L.Conv2d((hw, hw, cin), c3r, 1, **kw)
+            self.b5a = L.Conv2d((hw, hw, cin), c5r, 1, **kw)
+            w1 = self.b1.out_shape[2]
+            w3r = self.b3a.out_shape[2]
+            w5r = self.b5a.out_shape[2]
+        self.b3b = L.Conv2d((hw, hw, w3r), c3, 3, pad=1, **kw)
+        self.b5b = L.Conv2d((hw, hw, w5r), c5, 5, pad=2, **kw)
         self.pool = L.MaxPool2d(3, 1, 1)
         self.bp = L.Conv2d((hw, hw, cin), cp, 1, **kw)
-        self.out_c = (self.b1.out_shape[2] + self.b3b.out_shape[2]
+        self._w1 = w1
+        self.out_c = (w1 + self.b3b.out_shape[2]
                       + self.b5b.out_shape[2] + self.bp.out_shape[2])
 
+    def logical_params(self):
+        """Reference per-branch order: b1, b3a, b3b, b5a, b5b, bp."""
+        if self.merged:
+            br = self.b135.branch_logical_params()
+            return (br[0] + br[1] + self.b3b.logical_params()
+                    + br[2] + self.b5b.logical_params()
+                    + self.bp.logical_params())
+        return (self.b1.logical_params() + self.b3a.logical_params()
+                + self.b3b.logical_params() + self.b5a.logical_params()
+                + self.b5b.logical_params() + self.bp.logical_params())
+
     def forward(self, x):
-        if self.b1.gpu and x.is_cuda:
-            # concat-writing branches: every branch's final conv writes
-            # its channel slice of the concat buffer in place
+        if self.merged and x.is_cuda:
+            # one fused GEMM for the three 1x1s, then concat-writing
+            # branch convs: every final output lands in the concat
+            # buffer in place
             N = x.shape[0]
-            P = self.b1.out_shape[0]
+            P = self.b3b.out_shape[0]
+            m = self.b135(x)
+            s1, s3, s5 = self.b135.slices(m)
             buf = torch.empty((N, P, P, self.out_c),
                               dtype=torch.bfloat16, device=x.device)
-            ws = [self.b1.out_shape[2], self.b3b.out_shape[2],
+            ws = [self._w1, self.b3b.out_shape[2],
                   self.b5b.out_shape[2], self.bp.out_shape[2]]
             o = [0, ws[0], ws[0] + ws[1], ws[0] + ws[1] + ws[2]]
-            y1 = self.b1(x, out=buf[..., o[0]:o[1]])
-            y3 = self.b3b(self.b3a(x), out=buf[..., o[1]:o[2]])
-            y5 = self.b5b(self.b5a(x), out=buf[..., o[2]:o[3]])
+            y1 = L.copy_slice(buf[..., o[0]:o[1]], s1)
+            y3 = self.b3b(s3, out=buf[..., o[1]:o[2]])
+            y5 = self.b5b(s5, out=buf[..., o[2]:o[3]])
             yp = self.bp(self.pool(x), out=buf[..., o[3]:])
             return L.join_views(buf, y1, y3, y5, yp)
         y1 = self.b1(x)

@@ -655,6 +655,104 @@ class Conv2d(_Base):
         self.post_step()
 
 
+class MultiConv1x1(Conv2d):
+    """Several 1x1 convs over the SAME input fused into one GEMM
+    (inception branch batching, GPU only): output channels are the
+    branches' padded blocks ``[pad(k0) | pad(k1) | ...]``; callers
+    slice per branch.  One fwd GEMM, one dgrad (which also fuses the
+    branches' dx sum), one wgrad, one bias pass — GoogLeNet-32b is
+    launch-latency bound (615 dispatches/step measured), so fewer,
+    fatter launches are the lever multi-stream could not be.
+
+    Numerics match the unmerged branches exactly: each branch's pad
+    rows start at zero (like an unmerged conv's Kg..Kgp rows) and train
+    under the same gradients.  ``branch_logical_params()`` exposes each
+    branch's logical ``[k_i, Cin]``-shaped weight/bias so checkpoints
+    keep the reference's per-branch entries (weights are cross-loadable
+    with an unmerged model; momentum files remain layout-positional)."""
+
+    def __init__(self, in_shape, ks, **kw):
+        from theanompi_amd.ops.functional import cpad_for
+        self.branch_k = list(ks)
+        self.branch_kp = [cpad_for(k) for k in ks]
+        self.branch_off = [sum(self.branch_kp[:i])
+                           for i in range(len(ks))]
+        super().__init__(in_shape, sum(self.branch_kp), 1, **kw)
+        assert self.gpu and self.groups == 1
+        assert self.plan.Kg == self.plan.Kgp == sum(self.branch_kp)
+        # branch pad rows start at zero, exactly like an unmerged
+        # conv's physical pad rows
+        with torch.no_grad():
+            for k, kp, off in zip(self.branch_k, self.branch_kp,
+                                  self.branch_off):
+                if kp > k:
+                    self.weight.data[0, off + k:off + kp].zero_()
+                    self.bias.data[off + k:off + kp].zero_()
+        self.post_step()
+
+    def slices(self, y):
+        return [y[..., o:o + kp]
+                for o, kp in zip(self.branch_off, self.branch_kp)]
+
+    def branch_logical_params(self):
+        """Per-branch [(W, get, set), (b, get, set)] lists in branch
+        order; shapes match the equivalent standalone Conv2d."""
+        outs = []
+        for i, (k, off) in enumerate(zip(self.branch_k,
+                                         self.branch_off)):
+            def get_w(k=k, off=off):
+                full = self.plan.flat_to_w(self.weight.data)
+                return full[off:off + k, :self.Cg_logical]
+
+            def set_w(v, k=k, off=off):
+                plan = self.plan
+                wl = self.plan.flat_to_w(self.weight.data).cpu()
+                wl[off:off + k, :self.Cg_logical] = v.cpu()
+                plan.w_to_flat(wl.to(self.device_t),
+                               out=self.weight.data)
+                self.post_step()
+
+            def get_b(k=k, off=off):
+                return self.bias.data[off:off + k].clone()
+
+            def set_b(v, k=k, off=off):
+                self.bias.data[off:off + k] = v.to(self.device_t)
+                self.post_step()
+
+            outs.append([(f"W{i}", get_w, set_w),
+                         (f"b{i}", get_b, set_b)])
+        return outs
+
+    def logical_params(self):
+        # flattened branch order (used when the owner does not compose
+        # a custom order)
+        out = []
+        for entry in self.branch_logical_params():
+            out.extend(entry)
+        return out
+
+
+class _CopySlice(torch.autograd.Function):
+    """Write src into a channel slice of a (non-tracked) concat buffer
+    and return the slice as an autograd output: gradient passes
+    straight through to src."""
+
+    @staticmethod
+    def forward(ctx, dst_slice, src):
+        with torch.no_grad():
+            dst_slice.copy_(src)
+        ctx.mark_dirty(dst_slice)
+        return dst_slice
+
+    @staticmethod
+    def backward(ctx, dy):
+        return None, dy
+
+
+def copy_slice(dst_slice, src):
+    return _CopySlice.apply(dst_slice, src)
+
+
 class Linear(_Base):
     """FC layer; logical weight [IN, OUT] (Theano layout).  GPU path is a
     1x1x1 image conv through the same implicit-GEMM kernel (N11)."""
