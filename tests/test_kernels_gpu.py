@@ -478,16 +478,38 @@ def test_inception_merged_equivalence():
     yc = ci(x.clone().requires_grad_())
     assert yg.shape == yc.shape
     assert rel_err(yg, yc) < 2e-2
-    dy = q(torch.randn_like(yc))
-    yg.backward(dy.to(dev()).to(torch.bfloat16))
-    yc.backward(dy)
-    # merged flat weight rows map to branch [k, cin] 1x1 weights
-    for bi, branch in enumerate([ci.b1, ci.b3a, ci.b5a]):
+
+    # gradient check vs STANDALONE GPU 1x1 convs (identical kernels and
+    # bf16 relu decisions — the CPU fp32 baseline flips borderline relu
+    # masks and diverges in wgrad; forward above pins CPU parity)
+    from theanompi_amd.models import layers as L
+    ks = [32, 48, 16]
+    singles = []
+    for bi, k in enumerate(ks):
+        c = L.Conv2d((hw, hw, cin), k, 1, relu=True, device=str(dev()),
+                     batch=B)
         off = gi.b135.branch_off[bi]
-        k = gi.b135.branch_k[bi]
-        g_merged = gi.b135.weight.grad[0, off:off + k, :cin].cpu()
-        g_ref = branch.weight.grad[:, :, 0, 0]
-        assert rel_err(g_merged, g_ref) < 2e-2, bi
+        with torch.no_grad():
+            c.weight.data.copy_(
+                gi.b135.weight.data[:, off:off + k, :])
+            c.bias.data.copy_(gi.b135.bias.data[off:off + k])
+        c.post_step()
+        singles.append(c)
+    xg2 = x.to(dev()).to(torch.bfloat16).requires_grad_()
+    m = gi.b135(xg2)
+    dy = torch.randn_like(m)
+    m.backward(dy)
+    xg3 = x.to(dev()).to(torch.bfloat16).requires_grad_()
+    for bi, c in enumerate(singles):
+        off = gi.b135.branch_off[bi]
+        kp = gi.b135.branch_kp[bi]
+        ys = c(xg3)
+        ys.backward(dy[..., off:off + kp])
+        gm = gi.b135.weight.grad[0, off:off + kp, :]
+        assert rel_err(gm, c.weight.grad[0]) < 1e-3, bi
+        bm = gi.b135.bias.grad[off:off + kp]
+        assert rel_err(bm, c.bias.grad) < 1e-3, bi
+    assert rel_err(xg2.grad, xg3.grad) < 1e-3   # fused dx sum
 
 
 def test_sgd_cached_descriptor_lr_invalidate():
@@ -628,7 +650,9 @@ def test_inception_concat_write_equivalence():
         return y.detach().float(), xi.grad.detach().float(), grads
 
     def old_path(xi):
-        return T.cat([m.b1(xi), m.b3b(m.b3a(xi)), m.b5b(m.b5a(xi)),
+        mm = m.b135(xi)
+        s1, s3, s5 = m.b135.slices(mm)
+        return T.cat([s1, m.b3b(s3), m.b5b(s5),
                       m.bp(m.pool(xi))], dim=-1)
 
     y_new, dx_new, g_new = run(m)               # concat-write path

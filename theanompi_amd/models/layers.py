@@ -393,8 +393,8 @@ class _HipConvFn(torch.autograd.Function):
             db = _arena_alias(db) if (db is not None
                                       and layer.bias.grad is None) else \
                 torch.zeros_like(layer.b_f32)
-            C.bias_wgrad(dy, db)
-        # weight grad (fp32, flat layout)
+        # weight grad (fp32, flat layout); bias grad rides the same
+        # kernel (fTile==0 blocks reduce their staged dy tiles)
         dw = getattr(layer, "_dw_buf", None)
         dw = _arena_alias(dw) if (dw is not None
                                   and layer.weight.grad is None) else \
@@ -404,7 +404,7 @@ class _HipConvFn(torch.autograd.Function):
         C.conv_wgrad(xp, dy, plan.table, dw, plan.kdf_mask,
                      plan.Wp * plan.Cphys, plan.Cphys,
                      plan.stride[0], plan.stride[1],
-                     plan.Cg if plan.G > 1 else 0, wv, wt)
+                     plan.Cg if plan.G > 1 else 0, wv, wt, db)
         dx = None
         if ctx.needs_input_grad[0]:
             dx = layer._dgrad(dy)

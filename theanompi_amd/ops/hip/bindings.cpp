@@ -431,7 +431,8 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
                        torch::Tensor kdfMask,
                        long rowStride, long colStride, long sh, long sw,
                        long groupInOff, long variant = 0,
-                       long split_target = 0) {
+                       long split_target = 0,
+                       c10::optional<torch::Tensor> db = c10::nullopt) {
   check_bf16(xp, "xp");
   check_bf16(dy, "dy");
   check_f32(dw, "dw");
@@ -458,6 +459,13 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
   a.Kgp = (int)Kgp;
   a.Kphys = (int)Kphys;
   a.xTotal = xp.numel();
+  a.db = nullptr;
+  if (db.has_value()) {
+    auto& dbt = db.value();
+    check_f32(dbt, "db");
+    TORCH_CHECK(dbt.numel() == G * Kgp, "db must be [G*Kgp]");
+    a.db = dbt.data_ptr<float>();
+  }
   long base_blocks = ((Kgp + 63) / 64) * ((Kdf + 63) / 64) * G;
   static const long wgrad_env = [] {
     const char* e = getenv("THEANOMPI_WGRAD_T");
@@ -972,7 +980,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("table"), py::arg("dw"), py::arg("kdfMask"),
         py::arg("rowStride"), py::arg("colStride"), py::arg("sh"),
         py::arg("sw"), py::arg("groupInOff"), py::arg("variant") = 0,
-        py::arg("split_target") = 0);
+        py::arg("split_target") = 0, py::arg("db") = py::none());
   m.def("bias_wgrad", &bias_wgrad);
   m.def("relu_fwd", &relu_fwd);
   m.def("relu_bwd", &relu_bwd);

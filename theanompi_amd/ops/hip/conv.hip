@@ -582,6 +582,10 @@ struct WgradArgs {
   int Kdf, Kgp, Kphys;
   int splitK;
   long xTotal;                      // X elements (gather tail clamp)
+  // optional fused bias grad: db[kout] += sum_pix dy — the fTile==0
+  // blocks re-read their staged dy tiles from LDS (the separate
+  // bias_wgrad kernel was ~51 launches/step on GoogLeNet)
+  float* __restrict__ db;
 };
 
 template <int LAYOUT, int BNF, int BMK = 64>
@@ -682,6 +686,12 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
 #pragma unroll
     for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
+  // fused bias grad: thread j < BMK owns kout column j of this block's
+  // dy tiles (only fTile==0 blocks participate; split-K slices cover
+  // disjoint pixel sets, atomics combine)
+  const bool doBias = (a.db != nullptr) && (fTile == 0);
+  float dbAcc = 0.f;
+
   const int nchunks = (a.M + BKP - 1) / BKP;
   const int nci = (nchunks - slice + a.splitK - 1) / a.splitK;
   if (slice < nchunks) {
@@ -696,6 +706,16 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
         asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
       const int buf = ci & 1;
+      if (doBias && tid < BMK) {
+        // logical col j of row r sits at the staging-swizzled slot
+        const int j = tid;
+        const int jg = j & ~7, jo = j & 7;
+        for (int r2 = 0; r2 < BKP; ++r2) {
+          int slot = (jg ^ ((r2 & 7) * 8) ^ (((r2 >> 3) & 3)
+                                            * (BMK / 4))) + jo;
+          dbAcc += (float)DYt[buf][r2][slot];
+        }
+      }
 #pragma unroll
       for (int ks = 0; ks < BKP; ks += 32) {
         bf16x8 af[MW], bfr[FN];
@@ -716,6 +736,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
       }
       __builtin_amdgcn_s_barrier();
     }
+  }
+
+  if (doBias && tid < BMK) {
+    int kout = kout0 + tid;
+    if (kout < a.Kgp) atomicAdd(&a.db[g * a.Kgp + kout], dbAcc);
   }
 
   // epilogue: atomic accumulate fp32
