@@ -46,6 +46,7 @@ struct WgradArgs {
   int Kdf, Kgp, Kphys;
   int splitK;
   long xTotal;
+  float* db;          // optional fused bias grad
 };
 
 extern "C" {
