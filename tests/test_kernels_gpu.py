@@ -509,7 +509,11 @@ def test_inception_merged_equivalence():
         assert rel_err(gm, c.weight.grad[0]) < 1e-3, bi
         bm = gi.b135.bias.grad[off:off + kp]
         assert rel_err(bm, c.bias.grad) < 1e-3, bi
-    assert rel_err(xg2.grad, xg3.grad) < 1e-3   # fused dx sum
+    # fused dx sum: the STANDALONE baseline rounds each branch's dx to
+    # bf16 and accumulates in bf16 (3 roundings + bf16 adds); the
+    # merged GEMM accumulates everything in fp32 with one rounding, so
+    # they differ by baseline rounding noise (~5e-3 measured)
+    assert rel_err(xg2.grad, xg3.grad) < 2e-2
 
 
 def test_sgd_cached_descriptor_lr_invalidate():
