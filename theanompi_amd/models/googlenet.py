@@ -70,7 +70,8 @@ class _Inception(nn.Module):
             N = x.shape[0]
             P = self.b3b.out_shape[0]
             m = self.b135(x)
-            s1, s3, s5 = self.b135.slices(m)
+            # one-cat backward instead of per-slice scatter+add glue
+            s1, s3, s5 = L.split_channels(m, self.b135.branch_kp)
             buf = torch.empty((N, P, P, self.out_c),
                               dtype=torch.bfloat16, device=x.device)
             ws = [self._w1, self.b3b.out_shape[2],

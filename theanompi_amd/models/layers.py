@@ -753,6 +753,39 @@ def copy_slice(dst_slice, src):
     return _CopySlice.apply(dst_slice, src)
 
 
+class _SplitChannels(torch.autograd.Function):
+    """Channel split whose backward is ONE cat: plain slicing would
+    give each consumer a SliceBackward (zeros_like + scatter) plus adds
+    to combine them — ~8 full-size kernels per inception module."""
+
+    @staticmethod
+    def forward(ctx, m, *widths):
+        ctx.widths = widths
+        ctx.shape = m.shape
+        ctx.dev = m.device
+        ctx.dt = m.dtype
+        outs, o = [], 0
+        for w in widths:
+            outs.append(m[..., o:o + w])
+            o += w
+        assert o == m.shape[-1], "widths must partition the channels"
+        return tuple(outs)
+
+    @staticmethod
+    def backward(ctx, *ds):
+        parts = []
+        for d, w in zip(ds, ctx.widths):
+            if d is None:
+                d = torch.zeros(*ctx.shape[:-1], w, dtype=ctx.dt,
+                                device=ctx.dev)
+            parts.append(d)
+        return (torch.cat(parts, dim=-1), *([None] * len(ctx.widths)))
+
+
+def split_channels(m, widths):
+    return _SplitChannels.apply(m, *widths)
+
+
 class Linear(_Base):
     """FC layer; logical weight [IN, OUT] (Theano layout).  GPU path is a
     1x1x1 image conv through the same implicit-GEMM kernel (N11)."""
