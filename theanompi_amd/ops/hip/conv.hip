@@ -588,21 +588,21 @@ struct WgradArgs {
   float* __restrict__ db;
 };
 
-template <int LAYOUT, int BNF, int BMK = 64>
+template <int LAYOUT, int BNF, int BMK = 64, int DEPTH = 2>
 __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   // TN GEMM over pixels.  Both operands stage LINEARLY ([pix][64ch]
   // subtiles, 16-B glds pieces, content-swizzled) and the k-major MFMA
   // fragments come from ds_read_b64_tr_b16 (hardware transpose).
-  // Counted-vmcnt double buffering, LDS-resident gather table.
-  // BMK (kout tile) 64 or 128: the wider tile does 2x the MFMA work
-  // per staged dy byte for wide-Kgp layers (autotuned per shape).
+  // Counted-vmcnt DEPTH-deep ring (like the fwd gather), LDS-resident
+  // gather table.  BMK (kout tile) 64 or 128: the wider tile does 2x
+  // the MFMA work per staged dy byte for wide-Kgp layers (autotuned).
   constexpr int BKP = 64;          // pixel chunk
   constexpr int XI = BNF / 32;     // x-tile glds per wave
   constexpr int DI = BMK / 32;     // dy-tile glds per wave
   constexpr int TPW = XI + DI;
 
-  __shared__ __align__(16) __bf16 DYt[2][BKP][BMK];       // [pix][kout]
-  __shared__ __align__(16) __bf16 XT[2][BKP][BNF];        // [pix][fd]
+  __shared__ __align__(16) __bf16 DYt[DEPTH][BKP][BMK];   // [pix][kout]
+  __shared__ __align__(16) __bf16 XT[DEPTH][BKP][BNF];    // [pix][fd]
   extern __shared__ int Tab[];
 
   const int tid = threadIdx.x;
@@ -695,17 +695,26 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   const int nchunks = (a.M + BKP - 1) / BKP;
   const int nci = (nchunks - slice + a.splitK - 1) / a.splitK;
   if (slice < nchunks) {
-    stage(0, slice * BKP);
+#pragma unroll
+    for (int d = 0; d < DEPTH - 1; ++d)
+      if (d < nci) stage(d % DEPTH, (slice + (long)d * a.splitK) * BKP);
     for (int ci = 0; ci < nci; ++ci) {
-      if (ci + 1 < nci)
-        stage((ci + 1) & 1, (slice + (long)(ci + 1) * a.splitK) * BKP);
-      if (ci + 1 < nci)
+      const int tnext = ci + DEPTH - 1;
+      if (tnext < nci)
+        stage(tnext % DEPTH,
+              (slice + (long)tnext * a.splitK) * BKP);
+      const int newer = (nci - 1 - ci) < (DEPTH - 1) ? (nci - 1 - ci)
+                                                     : (DEPTH - 1);
+      if (newer == 2)
+        asm volatile("s_waitcnt vmcnt(%0) lgkmcnt(0)" ::"i"(2 * TPW)
+                     : "memory");
+      else if (newer == 1)
         asm volatile("s_waitcnt vmcnt(%0) lgkmcnt(0)" ::"i"(TPW)
                      : "memory");
       else
         asm volatile("s_waitcnt vmcnt(0) lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_s_barrier();
-      const int buf = ci & 1;
+      const int buf = ci % DEPTH;
       if (doBias && tid < BMK) {
         // logical col j of row r sits at the staging-swizzled slot
         const int j = tid;
@@ -952,9 +961,10 @@ hipError_t launch_conv_gather_gemm(GatherGemmArgs* args, int G,
 }
 
 // wgrad variants (autotuner): 0 auto (BNF heuristic + env), 1 BNF=64,
-// 2 BNF=128, 3 BNF=64/BMK=128, 4 BNF=128/BMK=128.  The gather table is
-// dynamically sized here too: at BNF=128 that is the difference
-// between 2 and 3 blocks/CU.
+// 2 BNF=128, 3 BNF=64/BMK=128, 4 BNF=128/BMK=128, and 3-deep-ring
+// versions 5 (64/64/D3), 6 (128/64/D3), 7 (64/128/D3).  The gather
+// table is dynamically sized here too: at BNF=128 that is the
+// difference between 2 and 3 blocks/CU.
 hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
                              hipStream_t stream, int variant) {
   WgradArgs a = *args;
@@ -969,27 +979,35 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
   // epilogues were parallelized; THEANOMPI_WGRAD_BNF=128 restores it.
   int BNF = 64;
   if (bnf_env == 128 && a.Kdf % 128 == 0) BNF = 128;
-  if (variant == 1 || variant == 3) BNF = 64;
-  if ((variant == 2 || variant == 4) && a.Kdf % 128 == 0) BNF = 128;
-  int BMK = (variant == 3 || variant == 4) ? 128 : 64;
+  if (variant == 1 || variant == 3 || variant == 5 || variant == 7)
+    BNF = 64;
+  if ((variant == 2 || variant == 4 || variant == 6)
+      && a.Kdf % 128 == 0)
+    BNF = 128;
+  int BMK = (variant == 3 || variant == 4 || variant == 7) ? 128 : 64;
+  int D = (variant >= 5) ? 3 : 2;
   dim3 grid(cdiv(a.Kgp, BMK), cdiv(a.Kdf, BNF), G * a.splitK);
   dim3 block(256);
-#define WG(BNF_, BMK_)                                                     \
+#define WG(BNF_, BMK_, D_)                                                 \
   do {                                                                     \
     if (layout == FRAG_CONTIG8)                                            \
-      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, BNF_, BMK_>), grid,   \
-                         block, tabBytes, stream, a);                      \
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, BNF_, BMK_, D_>),     \
+                         grid, block, tabBytes, stream, a);                \
     else                                                                   \
-      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, BNF_, BMK_>), grid,    \
-                         block, tabBytes, stream, a);                      \
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, BNF_, BMK_, D_>),      \
+                         grid, block, tabBytes, stream, a);                \
   } while (0)
-  if (BMK == 128) {
-    if (BNF == 128) WG(128, 128);
-    else WG(64, 128);
+  if (D == 3) {
+    if (BMK == 128) WG(64, 128, 3);
+    else if (BNF == 128) WG(128, 64, 3);
+    else WG(64, 64, 3);
+  } else if (BMK == 128) {
+    if (BNF == 128) WG(128, 128, 2);
+    else WG(64, 128, 2);
   } else if (BNF == 128) {
-    WG(128, 64);
+    WG(128, 64, 2);
   } else {
-    WG(64, 64);
+    WG(64, 64, 2);
   }
 #undef WG
   return hipGetLastError();
