@@ -8,8 +8,11 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <atomic>
 #include <cstdlib>
 #include <cstring>
+#include <mutex>
+#include <unordered_map>
 #include <vector>
 
 // ---- extern "C" launchers from the .hip TUs --------------------------------
@@ -107,7 +110,7 @@ hipError_t launch_bn_fwd_eval(const void*, void*, float*, float*,
 hipError_t launch_bn_bwd(const void*, const void*, const void*, const float*,
                          const float*, const float*, const float*, float*,
                          int*, float*, float*, void*, long, int, int,
-                         hipStream_t);
+                         int, int, hipStream_t);
 hipError_t launch_softmax_xent_fwd(const void*, const long*, void*, float*,
                                    int, int, int, float*, hipStream_t);
 hipError_t launch_softmax_xent_bwd(const void*, const long*, void*, int, int,
@@ -751,21 +754,38 @@ static void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma,
   check_bf16(dy, "dy");
   long C = x.size(-1);
   long rows = x.numel() / C;
-  auto sdy = torch::empty_like(dgamma);     // zeroed by the prep kernel
-  auto sdyx = torch::empty_like(dgamma);
+  // the per-channel accumulators ARE the grad outputs: sdy == dbeta
+  // (sum dy), sdyx == dgamma (sum dy*xhat); prep zeroes them, the sums
+  // pass accumulates, nothing to copy afterwards
+  auto& sdy = dbeta;
+  auto& sdyx = dgamma;
   auto rgamma = torch::empty_like(gamma);
-  // THEANOMPI_BN_YPATH=0 forces the exact x path (A/B knob): allok
-  // starts at 0, the prep kernel only ever clears it, so the y-path
-  // instantiation always exits.
+  // THEANOMPI_BN_YPATH=0 forces the exact x path (A/B knob).  allok
+  // is a PERSISTENT per-device cell holding the token of the last
+  // call whose y-path was unusable — no per-call reset fill (that was
+  // one FillFunctor launch per BN backward, 53/step on ResNet-50).
   static const bool ypath_on = [] {
     const char* e = getenv("THEANOMPI_BN_YPATH");
     return !(e && e[0] == '0');
   }();
-  auto allok = ypath_on
-      ? torch::ones({1}, torch::dtype(torch::kInt32)
-                             .device(gamma.device()))
-      : torch::zeros({1}, torch::dtype(torch::kInt32)
-                              .device(gamma.device()));
+  static std::mutex allok_mu;
+  static std::unordered_map<int, torch::Tensor> allok_cells;
+  torch::Tensor allok;
+  {
+    std::lock_guard<std::mutex> lk(allok_mu);
+    auto it = allok_cells.find((int)gamma.device().index());
+    if (it == allok_cells.end()) {
+      it = allok_cells
+               .emplace((int)gamma.device().index(),
+                        torch::zeros({1}, torch::dtype(torch::kInt32)
+                                              .device(gamma.device())))
+               .first;
+    }
+    allok = it->second;
+  }
+  static std::atomic<int> token_ctr{1};
+  int token = token_ctr.fetch_add(1);
+  if (token <= 0) { token_ctr.store(2); token = 1; }  // wrap safety
   CHECK_HIP(launch_bn_bwd(dy.data_ptr(), x.data_ptr(), y.data_ptr(),
                           save_mean.data_ptr<float>(),
                           save_invstd.data_ptr<float>(),
@@ -774,9 +794,8 @@ static void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma,
                           rgamma.data_ptr<float>(),
                           allok.data_ptr<int>(), sdy.data_ptr<float>(),
                           sdyx.data_ptr<float>(), dx.data_ptr(), rows,
-                          (int)C, relu ? 1 : 0, cur_stream()));
-  dgamma.copy_(sdyx);
-  dbeta.copy_(sdy);
+                          (int)C, relu ? 1 : 0, token,
+                          ypath_on ? 0 : 1, cur_stream()));
 }
 
 // ---- softmax / sgd ----------------------------------------------------------

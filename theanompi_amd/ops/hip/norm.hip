@@ -397,11 +397,15 @@ __global__ void bn_apply_k(const __bf16* __restrict__ x,
 // bn_bwd_sums / bn_bwd_apply, which are memory-bound (ResNet-50: BN is
 // ~31% of the step).  A tiny |gamma| would amplify y's bf16 rounding,
 // so that case keeps the exact x path (uniform per-layer branch).
+// allok is a persistent per-device "bad token" cell (no per-call
+// reset fill — that was one FillFunctor launch per BN backward, 53 a
+// step on ResNet-50): the y-path is usable iff *allok != token.
 __global__ void bn_gamma_prep_k(const float* __restrict__ gamma,
                                 float* __restrict__ rgamma,
                                 int* __restrict__ allok,
                                 float* __restrict__ sdy,
-                                float* __restrict__ sdyx, int C) {
+                                float* __restrict__ sdyx, int C,
+                                int token, int forceX) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
   sdy[c] = 0.f;                 // accumulators for the sums pass
@@ -409,7 +413,7 @@ __global__ void bn_gamma_prep_k(const float* __restrict__ gamma,
   float g = gamma[c];
   bool ok = fabsf(g) > 1e-2f;
   rgamma[c] = ok ? 1.f / g : 0.f;
-  if (!ok) atomicAnd(allok, 0);
+  if (!ok || forceX) atomicExch(allok, token);
 }
 
 // backward pass 1: per-channel  sum(dy)  and  sum(dy * xhat)
@@ -429,8 +433,8 @@ __global__ void bn_bwd_sums_k(const __bf16* __restrict__ dy,
                               const int* __restrict__ allok,
                               float* __restrict__ sdy,
                               float* __restrict__ sdyx, long rows,
-                              int C) {
-  if ((*allok != 0) != (USEY != 0)) return;   // other path owns this op
+                              int C, int token) {
+  if ((*allok != token) != (USEY != 0)) return;  // other path owns this
   constexpr int RLN = 256 / CLW;
   const int c8n = C / 8;
   const int cl = threadIdx.x & (CLW - 1);
@@ -512,14 +516,14 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
                                const float* __restrict__ sdy,
                                const float* __restrict__ sdyx,
                                __bf16* __restrict__ dx, long rows, int C,
-                               int relu) {
+                               int relu, int token) {
   const float invM = 1.f / rows;
   const long t4 = rows * C / 4;
   // xhat from y — skip the x read.  Only valid without fused relu:
   // the mean-subtraction terms need the true xhat at MASKED positions
   // too, and y=0 there cannot reconstruct it.  (bn_bwd_sums_k has no
   // such restriction: its xhat is always multiplied by the masked g.)
-  const bool useY = (*allok != 0) && !relu;
+  const bool useY = (*allok != token) && !relu;
   if (useY) {           // non-relu only: dy + y loads, no x traffic
     for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < t4;
          i4 += (long)gridDim.x * blockDim.x) {
@@ -682,9 +686,11 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
                          const float* gamma, const float* beta,
                          float* rgamma, int* allok, float* sdy,
                          float* sdyx, void* dx, long rows, int C,
-                         int relu, hipStream_t stream) {
+                         int relu, int token, int forceX,
+                         hipStream_t stream) {
   hipLaunchKernelGGL(bn_gamma_prep_k, dim3(cdiv(C, 256)), dim3(256), 0,
-                     stream, gamma, rgamma, allok, sdy, sdyx, C);
+                     stream, gamma, rgamma, allok, sdy, sdyx, C, token,
+                     forceX);
   {
     int c8n = C / 8;
     int clw = c8n >= 32 ? 32 : (c8n >= 16 ? 16 : 8);
@@ -704,11 +710,11 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
       hipLaunchKernelGGL(k1, g, dim3(256), 0, stream, (const __bf16*)dy,
                          (const __bf16*)x, (const __bf16*)y, mean,
                          invstd, beta, rgamma, allok, sdy, sdyx, rows,
-                         C);
+                         C, token);
       hipLaunchKernelGGL(k0, g, dim3(256), 0, stream, (const __bf16*)dy,
                          (const __bf16*)x, (const __bf16*)y, mean,
                          invstd, beta, rgamma, allok, sdy, sdyx, rows,
-                         C);
+                         C, token);
     };
     auto pick = [&](auto tag) {
       constexpr int W = decltype(tag)::value;
@@ -724,7 +730,8 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
   hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C / 4)), dim3(256),
                      0, stream, (const __bf16*)dy, (const __bf16*)x,
                      (const __bf16*)y, mean, invstd, gamma, beta, rgamma,
-                     allok, sdy, sdyx, (__bf16*)dx, rows, C, relu);
+                     allok, sdy, sdyx, (__bf16*)dx, rows, C, relu,
+                     token);
   return hipGetLastError();
 }
 

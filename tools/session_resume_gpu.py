@@ -2,7 +2,12 @@
 checkpoint workflow + end-to-end RESUME on hardware."""
 import glob
 import os
-from theanompi_amd import BSP
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+
+from theanompi_amd import BSP  # noqa: E402
 
 ck = "gpurun_out/sess_ck"
 rule = BSP()
