@@ -389,6 +389,52 @@ __global__ void bn_apply_k(const __bf16* __restrict__ x,
   }
 }
 
+// 16-B, 2-deep-unrolled normalize (the 8-B kernel above is
+// latency-bound on the big early-layer tensors; C % 8 == 0 always
+// holds for the channel-padded layouts)
+__global__ void bn_apply_k16(const __bf16* __restrict__ x,
+                             __bf16* __restrict__ y,
+                             const float* __restrict__ mean,
+                             const float* __restrict__ invstd,
+                             const float* __restrict__ gamma,
+                             const float* __restrict__ beta, long rows,
+                             int C, int relu) {
+  const long t8 = rows * C / 8;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i8 = blockIdx.x * (long)blockDim.x + threadIdx.x; i8 < t8;
+       i8 += 2 * stride) {
+    long j8 = i8 + stride;
+    long i = i8 * 8;
+    long j = j8 < t8 ? j8 * 8 : i;
+    bf16x8 xv0 = *(const bf16x8*)(x + i);
+    bf16x8 xv1 = *(const bf16x8*)(x + j);
+    int c0 = (int)(i % C), c1 = (int)(j % C);
+    bf16x8 o0, o1;
+#pragma unroll
+    for (int h = 0; h < 2; ++h) {
+      int c = (h ? c1 : c0) + h * 0;
+      const bf16x8& xv = h ? xv1 : xv0;
+      bf16x8& o = h ? o1 : o0;
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        f32x4 mv = *(const f32x4*)(mean + c + q * 4);
+        f32x4 iv = *(const f32x4*)(invstd + c + q * 4);
+        f32x4 gv = *(const f32x4*)(gamma + c + q * 4);
+        f32x4 bv = *(const f32x4*)(beta + c + q * 4);
+#pragma unroll
+        for (int k2 = 0; k2 < 4; ++k2) {
+          float v = ((float)xv[q * 4 + k2] - mv[k2]) * iv[k2] * gv[k2]
+                    + bv[k2];
+          if (relu && v < 0.f) v = 0.f;
+          o[q * 4 + k2] = (__bf16)v;
+        }
+      }
+    }
+    *(bf16x8*)(y + i) = o0;
+    if (j8 < t8) *(bf16x8*)(y + j8 * 8) = o1;
+  }
+}
+
 // Backward prep: per-channel 1/gamma plus a single "all channels safe"
 // flag.  When every |gamma| is comfortably away from 0 the backward
 // kernels reconstruct xhat from the SAVED bf16 OUTPUT
@@ -575,6 +621,82 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
   }
 }
 
+// 16-B variant (C % 8 == 0): same math, half the memory transactions
+__global__ void bn_bwd_apply_k16(const __bf16* __restrict__ dy,
+                                 const __bf16* __restrict__ x,
+                                 const __bf16* __restrict__ y,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ invstd,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ beta,
+                                 const float* __restrict__ rgamma,
+                                 const int* __restrict__ allok,
+                                 const float* __restrict__ sdy,
+                                 const float* __restrict__ sdyx,
+                                 __bf16* __restrict__ dx, long rows,
+                                 int C, int relu, int token) {
+  const float invM = 1.f / rows;
+  const long t8 = rows * C / 8;
+  const bool useY = (*allok != token) && !relu;
+  if (useY) {
+    for (long i8 = blockIdx.x * (long)blockDim.x + threadIdx.x; i8 < t8;
+         i8 += (long)gridDim.x * blockDim.x) {
+      long i = i8 * 8;
+      int c = (int)(i % C);
+      bf16x8 gv8 = *(const bf16x8*)(dy + i);
+      bf16x8 yv8 = *(const bf16x8*)(y + i);
+      bf16x8 o;
+#pragma unroll
+      for (int q = 0; q < 2; ++q) {
+        f32x4 bv = *(const f32x4*)(beta + c + q * 4);
+        f32x4 rg = *(const f32x4*)(rgamma + c + q * 4);
+        f32x4 ist = *(const f32x4*)(invstd + c + q * 4);
+        f32x4 ga = *(const f32x4*)(gamma + c + q * 4);
+        f32x4 s1 = *(const f32x4*)(sdy + c + q * 4);
+        f32x4 s2 = *(const f32x4*)(sdyx + c + q * 4);
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          float g = (float)gv8[q * 4 + j];
+          float xh = ((float)yv8[q * 4 + j] - bv[j]) * rg[j];
+          o[q * 4 + j] = (__bf16)(ist[j] * ga[j]
+                                  * (g - s1[j] * invM
+                                     - xh * s2[j] * invM));
+        }
+      }
+      *(bf16x8*)(dx + i) = o;
+    }
+    return;
+  }
+  for (long i8 = blockIdx.x * (long)blockDim.x + threadIdx.x; i8 < t8;
+       i8 += (long)gridDim.x * blockDim.x) {
+    long i = i8 * 8;
+    int c = (int)(i % C);
+    bf16x8 gv8 = *(const bf16x8*)(dy + i);
+    bf16x8 xv8 = *(const bf16x8*)(x + i);
+    bf16x8 yv8;
+    if (relu) yv8 = *(const bf16x8*)(y + i);
+    bf16x8 o;
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      f32x4 mv = *(const f32x4*)(mean + c + q * 4);
+      f32x4 iv = *(const f32x4*)(invstd + c + q * 4);
+      f32x4 ga = *(const f32x4*)(gamma + c + q * 4);
+      f32x4 s1 = *(const f32x4*)(sdy + c + q * 4);
+      f32x4 s2 = *(const f32x4*)(sdyx + c + q * 4);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float g = (float)gv8[q * 4 + j];
+        if (relu && (float)yv8[q * 4 + j] <= 0.f) g = 0.f;
+        float xh = ((float)xv8[q * 4 + j] - mv[j]) * iv[j];
+        o[q * 4 + j] = (__bf16)(iv[j] * ga[j]
+                                * (g - s1[j] * invM
+                                   - xh * s2[j] * invM));
+      }
+    }
+    *(bf16x8*)(dx + i) = o;
+  }
+}
+
 static inline int ngrid(long work) {
   long g = (work + 255) / 256;
   if (g > 4096) g = 4096;
@@ -662,6 +784,13 @@ hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
   hipLaunchKernelGGL(bn_finalize_k, dim3(cdiv(C, 256)), dim3(256), 0,
                      stream, s1, s2, mean, invstd, rmean, rvar, rows, C,
                      momentum, eps);
+  if (C % 8 == 0) {
+    hipLaunchKernelGGL(bn_apply_k16, dim3(ngrid(rows * C / 16)),
+                       dim3(256), 0, stream, (const __bf16*)x,
+                       (__bf16*)y, mean, invstd, gamma, beta, rows, C,
+                       relu);
+    return hipGetLastError();
+  }
   hipLaunchKernelGGL(bn_apply_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
                      stream, (const __bf16*)x, (__bf16*)y, mean, invstd,
                      gamma, beta, rows, C, relu);
@@ -727,11 +856,19 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
     else if (clw == 16) pick(std::integral_constant<int, 16>{});
     else                pick(std::integral_constant<int, 8>{});
   }
-  hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C / 4)), dim3(256),
-                     0, stream, (const __bf16*)dy, (const __bf16*)x,
-                     (const __bf16*)y, mean, invstd, gamma, beta, rgamma,
-                     allok, sdy, sdyx, (__bf16*)dx, rows, C, relu,
-                     token);
+  if (C % 8 == 0) {
+    hipLaunchKernelGGL(bn_bwd_apply_k16, dim3(ngrid(rows * C / 8)),
+                       dim3(256), 0, stream, (const __bf16*)dy,
+                       (const __bf16*)x, (const __bf16*)y, mean, invstd,
+                       gamma, beta, rgamma, allok, sdy, sdyx,
+                       (__bf16*)dx, rows, C, relu, token);
+  } else {
+    hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C / 4)),
+                       dim3(256), 0, stream, (const __bf16*)dy,
+                       (const __bf16*)x, (const __bf16*)y, mean, invstd,
+                       gamma, beta, rgamma, allok, sdy, sdyx,
+                       (__bf16*)dx, rows, C, relu, token);
+  }
   return hipGetLastError();
 }
 
