@@ -452,7 +452,7 @@ def test_conv_variants_numerics():
         y, _dw, dx = run(vf, 0)
         assert rel_err(y, y0) < 5e-3, (vf, rel_err(y, y0))
         assert rel_err(dx, dx0) < 5e-3, (vf, rel_err(dx, dx0))
-    for vw in (1, 2, 3, 4, 5, 6, 7):
+    for vw in (1, 2, 3, 4, 5, 6, 7, 8):
         _y, dw, _dx = run(0, vw)
         assert rel_err(dw, dw0) < 5e-3, (vw, rel_err(dw, dw0))
 

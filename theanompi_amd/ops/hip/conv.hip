@@ -588,15 +588,17 @@ struct WgradArgs {
   float* __restrict__ db;
 };
 
-template <int LAYOUT, int BNF, int BMK = 64, int DEPTH = 2>
+template <int LAYOUT, int BNF, int BMK = 64, int DEPTH = 2,
+          int BKP = 64>
 __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   // TN GEMM over pixels.  Both operands stage LINEARLY ([pix][64ch]
   // subtiles, 16-B glds pieces, content-swizzled) and the k-major MFMA
   // fragments come from ds_read_b64_tr_b16 (hardware transpose).
   // Counted-vmcnt DEPTH-deep ring (like the fwd gather), LDS-resident
   // gather table.  BMK (kout tile) 64 or 128: the wider tile does 2x
-  // the MFMA work per staged dy byte for wide-Kgp layers (autotuned).
-  constexpr int BKP = 64;          // pixel chunk
+  // the MFMA work per staged dy byte for wide-Kgp layers; BKP (pixel
+  // chunk) 64 or 128 — the swizzles are 32-row periodic, so the wider
+  // chunk halves barrier/wait overhead per pixel (all autotuned).
   constexpr int XI = BNF / 32;     // x-tile glds per wave
   constexpr int DI = BMK / 32;     // dy-tile glds per wave
   constexpr int TPW = XI + DI;
@@ -979,35 +981,41 @@ hipError_t launch_conv_wgrad(const WgradArgs* args, int G, int layout,
   // epilogues were parallelized; THEANOMPI_WGRAD_BNF=128 restores it.
   int BNF = 64;
   if (bnf_env == 128 && a.Kdf % 128 == 0) BNF = 128;
-  if (variant == 1 || variant == 3 || variant == 5 || variant == 7)
+  if (variant == 1 || variant == 3 || variant == 5 || variant == 7
+      || variant == 8)
     BNF = 64;
   if ((variant == 2 || variant == 4 || variant == 6)
       && a.Kdf % 128 == 0)
     BNF = 128;
   int BMK = (variant == 3 || variant == 4 || variant == 7) ? 128 : 64;
-  int D = (variant >= 5) ? 3 : 2;
+  int D = (variant >= 5 && variant <= 7) ? 3 : 2;
+  int BKP = (variant == 8) ? 128 : 64;
   dim3 grid(cdiv(a.Kgp, BMK), cdiv(a.Kdf, BNF), G * a.splitK);
   dim3 block(256);
-#define WG(BNF_, BMK_, D_)                                                 \
+#define WG(BNF_, BMK_, D_, BKP_)                                           \
   do {                                                                     \
     if (layout == FRAG_CONTIG8)                                            \
-      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, BNF_, BMK_, D_>),     \
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_CONTIG8, BNF_, BMK_, D_,       \
+                                       BKP_>),                             \
                          grid, block, tabBytes, stream, a);                \
     else                                                                   \
-      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, BNF_, BMK_, D_>),      \
+      hipLaunchKernelGGL((conv_wgrad_k<FRAG_HALVES, BNF_, BMK_, D_,        \
+                                       BKP_>),                             \
                          grid, block, tabBytes, stream, a);                \
   } while (0)
-  if (D == 3) {
-    if (BMK == 128) WG(64, 128, 3);
-    else if (BNF == 128) WG(128, 64, 3);
-    else WG(64, 64, 3);
+  if (BKP == 128) {
+    WG(64, 64, 2, 128);
+  } else if (D == 3) {
+    if (BMK == 128) WG(64, 128, 3, 64);
+    else if (BNF == 128) WG(128, 64, 3, 64);
+    else WG(64, 64, 3, 64);
   } else if (BMK == 128) {
-    if (BNF == 128) WG(128, 128, 2);
-    else WG(64, 128, 2);
+    if (BNF == 128) WG(128, 128, 2, 64);
+    else WG(64, 128, 2, 64);
   } else if (BNF == 128) {
-    WG(128, 64, 2);
+    WG(128, 64, 2, 64);
   } else {
-    WG(64, 64, 2);
+    WG(64, 64, 2, 64);
   }
 #undef WG
   return hipGetLastError();

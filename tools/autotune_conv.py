@@ -49,7 +49,7 @@ SHAPES["fc"] = [
 
 FWD_VARIANTS = [1, 2, 3, 4, 5, 6, 7, 8, 9, 10]
 DGRAD_VARIANTS = [1, 2, 5, 6, 7, 8, 9, 10]
-WGRAD_GRID = [(v, t) for v in (1, 2, 3, 4, 5, 6, 7)
+WGRAD_GRID = [(v, t) for v in (1, 2, 3, 4, 5, 6, 7, 8)
               for t in (512, 1024, 2048, 4096)]
 
 
