@@ -599,8 +599,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
   // the MFMA work per staged dy byte for wide-Kgp layers; BKP (pixel
   // chunk) 64 or 128 — the swizzles are 32-row periodic, so the wider
   // chunk halves barrier/wait overhead per pixel (all autotuned).
-  constexpr int XI = BNF / 32;     // x-tile glds per wave
-  constexpr int DI = BMK / 32;     // dy-tile glds per wave
+  constexpr int XI = BKP * BNF / 2048;   // x-tile glds per wave
+  constexpr int DI = BKP * BMK / 2048;   // dy-tile glds per wave
   constexpr int TPW = XI + DI;
 
   __shared__ __align__(16) __bf16 DYt[DEPTH][BKP][BMK];   // [pix][kout]
