@@ -457,6 +457,37 @@ def test_conv_variants_numerics():
         assert rel_err(dw, dw0) < 5e-3, (vw, rel_err(dw, dw0))
 
 
+def test_batchnorm_fused_residual():
+    """y = relu(bn(x) + res) in one pass vs the separate composition —
+    forward and ALL grads (dx, dgamma, dbeta, dres)."""
+    from theanompi_amd.models import layers as L
+    from theanompi_amd.ops import functional as OF
+    torch.manual_seed(9)
+    d = dev()
+    N, H, W, C = 8, 14, 14, 64
+    bn1 = L.BatchNorm2d(C, relu=False, device=str(d))
+    bn2 = L.BatchNorm2d(C, relu=False, device=str(d))
+    x0 = torch.randn(N, H, W, C, device=d).to(torch.bfloat16)
+    r0 = torch.randn(N, H, W, C, device=d).to(torch.bfloat16)
+    dy = torch.randn(N, H, W, C, device=d).to(torch.bfloat16)
+
+    x1 = x0.clone().requires_grad_()
+    r1 = r0.clone().requires_grad_()
+    y1 = bn1(x1, residual=r1, relu=True)          # fused
+    y1.backward(dy)
+
+    x2 = x0.clone().requires_grad_()
+    r2 = r0.clone().requires_grad_()
+    y2 = OF.add(bn2(x2), r2, relu=True)           # separate
+    y2.backward(dy)
+
+    assert rel_err(y1, y2) < 5e-3
+    assert rel_err(x1.grad, x2.grad) < 1e-2
+    assert rel_err(r1.grad, r2.grad) < 5e-3
+    assert rel_err(bn1.gamma.grad, bn2.gamma.grad) < 1e-2
+    assert rel_err(bn1.beta.grad, bn2.beta.grad) < 1e-2
+
+
 def test_inception_merged_equivalence():
     """GPU inception (fused 3-way 1x1 GEMM + concat-writing branches)
     matches the unmerged CPU reference given identical weights loaded

@@ -905,10 +905,12 @@ class BatchNorm2d(_Base):
                              torch.ones(c_phys, device=dev))
         self.eps, self.momentum, self.relu = eps, momentum, relu
 
-    def forward(self, x, prestats=None):
+    def forward(self, x, prestats=None, residual=None, relu=None):
         return OF.batchnorm(x, self.gamma, self.beta, self.running_mean,
                             self.running_var, self.training, self.momentum,
-                            self.eps, self.relu, prestats=prestats)
+                            self.eps,
+                            self.relu if relu is None else relu,
+                            prestats=prestats, residual=residual)
 
     def logical_params(self):
         return [("gamma", lambda: self.gamma.data,

@@ -196,7 +196,11 @@ class _LRNFn(torch.autograd.Function):
 class _BatchNormFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, gamma, beta, running_mean, running_var,
-                training, momentum, eps, relu, prestats=None):
+                training, momentum, eps, relu, prestats=None,
+                residual=None):
+        # residual: fused residual join y = [relu](bn(x) + residual) —
+        # one pass instead of bn_apply + add_relu (ResNet bottleneck
+        # exits; the separate adds were ~8% of its step)
         C = _ops.ext()
         y = torch.empty_like(x)
         c = x.shape[-1]
@@ -204,10 +208,12 @@ class _BatchNormFn(torch.autograd.Function):
         save_invstd = torch.empty(c, dtype=torch.float32, device=x.device)
         C.bn_fwd(x, gamma, beta, y, save_mean, save_invstd,
                  running_mean, running_var, float(momentum), float(eps),
-                 bool(training), bool(relu), prestats)
+                 bool(training), bool(relu), prestats,
+                 residual.contiguous() if residual is not None else None)
         ctx.save_for_backward(x, gamma, beta, save_mean, save_invstd, y)
         ctx.relu = relu
         ctx.training = training
+        ctx.has_res = residual is not None
         return y
 
     @staticmethod
@@ -217,10 +223,18 @@ class _BatchNormFn(torch.autograd.Function):
         dx = torch.empty_like(x)
         dgamma = torch.empty_like(gamma)
         dbeta = torch.empty_like(gamma)
-        C.bn_bwd(dy.contiguous(), x, gamma, beta, save_mean,
-                 save_invstd, y, dx, dgamma, dbeta, bool(ctx.relu))
+        dy = dy.contiguous()
+        dres = None
+        if ctx.has_res and ctx.relu:
+            # masked dy comes out of the same kernel pass
+            dres = torch.empty_like(dy)
+        C.bn_bwd(dy, x, gamma, beta, save_mean,
+                 save_invstd, y, dx, dgamma, dbeta, bool(ctx.relu),
+                 dres, bool(ctx.has_res))
+        if ctx.has_res and not ctx.relu:
+            dres = dy              # pass-through, no kernel needed
         return (dx, dgamma, dbeta, None, None, None, None, None, None,
-                None)
+                None, dres)
 
 
 class _ReluFn(torch.autograd.Function):
@@ -361,13 +375,17 @@ def lrn(x, size=5, alpha=1e-4, beta=0.75, k=1.0):
 
 
 def batchnorm(x, gamma, beta, running_mean, running_var, training,
-              momentum=0.1, eps=1e-5, relu=False, prestats=None):
+              momentum=0.1, eps=1e-5, relu=False, prestats=None,
+              residual=None):
     if x.is_cuda:
         return _BatchNormFn.apply(x, gamma, beta, running_mean,
                                   running_var, training, momentum, eps,
-                                  relu, prestats)
-    return batchnorm_ref(x, gamma.float(), beta.float(), running_mean,
-                         running_var, training, momentum, eps, relu)
+                                  relu, prestats, residual)
+    y = batchnorm_ref(x, gamma.float(), beta.float(), running_mean,
+                      running_var, training, momentum, eps, relu=False)
+    if residual is not None:
+        y = y + residual
+    return F.relu(y) if relu else y
 
 
 def relu(x):

@@ -102,15 +102,15 @@ hipError_t launch_bn_fwd_train(const void*, void*, float*, float*,
                                float*, float*, float*, float*,
                                const float*, const float*, float*, int,
                                long, int, float, float, int,
-                               hipStream_t);
+                               const void*, hipStream_t);
 hipError_t launch_bn_fwd_eval(const void*, void*, float*, float*,
                               const float*, const float*, const float*,
                               const float*, long, int, float, int,
-                              hipStream_t);
+                              const void*, hipStream_t);
 hipError_t launch_bn_bwd(const void*, const void*, const void*, const float*,
                          const float*, const float*, const float*, float*,
                          int*, float*, float*, void*, long, int, int,
-                         int, int, hipStream_t);
+                         int, int, void*, hipStream_t);
 hipError_t launch_softmax_xent_fwd(const void*, const long*, void*, float*,
                                    int, int, int, float*, hipStream_t);
 hipError_t launch_softmax_xent_bwd(const void*, const long*, void*, int, int,
@@ -713,11 +713,18 @@ static void bn_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
                    torch::Tensor save_invstd, torch::Tensor rmean,
                    torch::Tensor rvar, double momentum, double eps,
                    bool training, bool relu,
-                   c10::optional<torch::Tensor> prestats = c10::nullopt) {
+                   c10::optional<torch::Tensor> prestats = c10::nullopt,
+                   c10::optional<torch::Tensor> res = c10::nullopt) {
   check_bf16(x, "x");
   check_f32(gamma, "gamma");
   long C = x.size(-1);
   long rows = x.numel() / C;
+  const void* resp = nullptr;
+  if (res.has_value()) {
+    check_bf16(res.value(), "res");
+    TORCH_CHECK(res.value().numel() == x.numel());
+    resp = res.value().data_ptr();
+  }
   if (training) {
     auto s1 = torch::zeros_like(save_mean);
     auto s2 = torch::zeros_like(save_mean);
@@ -735,14 +742,14 @@ static void bn_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
         rvar.data_ptr<float>(), gamma.data_ptr<float>(),
         beta.data_ptr<float>(), pre, STATS_NB, rows, (int)C,
         (float)momentum, (float)eps,
-        relu ? 1 : 0, cur_stream()));
+        relu ? 1 : 0, resp, cur_stream()));
   } else {
     CHECK_HIP(launch_bn_fwd_eval(
         x.data_ptr(), y.data_ptr(), save_mean.data_ptr<float>(),
         save_invstd.data_ptr<float>(), rmean.data_ptr<float>(),
         rvar.data_ptr<float>(), gamma.data_ptr<float>(),
         beta.data_ptr<float>(), rows, (int)C, (float)eps, relu ? 1 : 0,
-        cur_stream()));
+        resp, cur_stream()));
   }
 }
 
@@ -750,7 +757,9 @@ static void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma,
                    torch::Tensor beta, torch::Tensor save_mean,
                    torch::Tensor save_invstd, torch::Tensor y,
                    torch::Tensor dx, torch::Tensor dgamma,
-                   torch::Tensor dbeta, bool relu) {
+                   torch::Tensor dbeta, bool relu,
+                   c10::optional<torch::Tensor> dres = c10::nullopt,
+                   bool has_res = false) {
   check_bf16(dy, "dy");
   long C = x.size(-1);
   long rows = x.numel() / C;
@@ -795,7 +804,13 @@ static void bn_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma,
                           allok.data_ptr<int>(), sdy.data_ptr<float>(),
                           sdyx.data_ptr<float>(), dx.data_ptr(), rows,
                           (int)C, relu ? 1 : 0, token,
-                          ypath_on ? 0 : 1, cur_stream()));
+                          // a fused residual means the saved y is NOT
+                          // bn(x): the xhat-from-y reconstruction is
+                          // invalid, force the exact x path
+                          (!ypath_on || has_res) ? 1 : 0,
+                          dres.has_value() ? dres.value().data_ptr()
+                                           : nullptr,
+                          cur_stream()));
 }
 
 // ---- softmax / sgd ----------------------------------------------------------
@@ -1024,8 +1039,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("beta"), py::arg("y"), py::arg("save_mean"),
         py::arg("save_invstd"), py::arg("rmean"), py::arg("rvar"),
         py::arg("momentum"), py::arg("eps"), py::arg("training"),
-        py::arg("relu"), py::arg("prestats") = py::none());
-  m.def("bn_bwd", &bn_bwd);
+        py::arg("relu"), py::arg("prestats") = py::none(),
+        py::arg("res") = py::none());
+  m.def("bn_bwd", &bn_bwd, py::arg("dy"), py::arg("x"),
+        py::arg("gamma"), py::arg("beta"), py::arg("save_mean"),
+        py::arg("save_invstd"), py::arg("y"), py::arg("dx"),
+        py::arg("dgamma"), py::arg("dbeta"), py::arg("relu"),
+        py::arg("dres") = py::none(), py::arg("has_res") = false);
   m.def("softmax_xent_fwd", &softmax_xent_fwd, py::arg("logits"),
         py::arg("labels"), py::arg("probs"), py::arg("loss"),
         py::arg("nCls"), py::arg("stats") = py::none());

@@ -367,13 +367,19 @@ __global__ void bn_apply_k(const __bf16* __restrict__ x,
                            const float* __restrict__ invstd,
                            const float* __restrict__ gamma,
                            const float* __restrict__ beta, long rows,
-                           int C, int relu) {
+                           int C, int relu,
+                           const __bf16* __restrict__ res) {
+  // res (optional): fused residual join — y = relu(bn(x) + res), one
+  // pass instead of bn_apply + add_relu (the ResNet residual adds
+  // were ~8% of the step)
   const long t4 = rows * C / 4;
   for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < t4;
        i4 += (long)gridDim.x * blockDim.x) {
     long i = i4 * 4;
     int c = (int)(i % C);
     bf16x4 xv = *(const bf16x4*)(x + i);
+    bf16x4 rv;
+    if (res) rv = *(const bf16x4*)(res + i);
     f32x4 mv = *(const f32x4*)(mean + c);
     f32x4 iv = *(const f32x4*)(invstd + c);
     f32x4 gv = *(const f32x4*)(gamma + c);
@@ -382,6 +388,7 @@ __global__ void bn_apply_k(const __bf16* __restrict__ x,
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       float v = ((float)xv[j] - mv[j]) * iv[j] * gv[j] + bv[j];
+      if (res) v += (float)rv[j];
       if (relu && v < 0.f) v = 0.f;
       o[j] = (__bf16)v;
     }
@@ -398,7 +405,8 @@ __global__ void bn_apply_k16(const __bf16* __restrict__ x,
                              const float* __restrict__ invstd,
                              const float* __restrict__ gamma,
                              const float* __restrict__ beta, long rows,
-                             int C, int relu) {
+                             int C, int relu,
+                             const __bf16* __restrict__ res) {
   const long t8 = rows * C / 8;
   const long stride = (long)gridDim.x * blockDim.x;
   for (long i8 = blockIdx.x * (long)blockDim.x + threadIdx.x; i8 < t8;
@@ -408,12 +416,18 @@ __global__ void bn_apply_k16(const __bf16* __restrict__ x,
     long j = j8 < t8 ? j8 * 8 : i;
     bf16x8 xv0 = *(const bf16x8*)(x + i);
     bf16x8 xv1 = *(const bf16x8*)(x + j);
+    bf16x8 rv0, rv1;
+    if (res) {
+      rv0 = *(const bf16x8*)(res + i);
+      rv1 = *(const bf16x8*)(res + j);
+    }
     int c0 = (int)(i % C), c1 = (int)(j % C);
     bf16x8 o0, o1;
 #pragma unroll
     for (int h = 0; h < 2; ++h) {
-      int c = (h ? c1 : c0) + h * 0;
+      int c = h ? c1 : c0;
       const bf16x8& xv = h ? xv1 : xv0;
+      const bf16x8& rvv = h ? rv1 : rv0;
       bf16x8& o = h ? o1 : o0;
 #pragma unroll
       for (int q = 0; q < 2; ++q) {
@@ -425,6 +439,7 @@ __global__ void bn_apply_k16(const __bf16* __restrict__ x,
         for (int k2 = 0; k2 < 4; ++k2) {
           float v = ((float)xv[q * 4 + k2] - mv[k2]) * iv[k2] * gv[k2]
                     + bv[k2];
+          if (res) v += (float)rvv[q * 4 + k2];
           if (relu && v < 0.f) v = 0.f;
           o[q * 4 + k2] = (__bf16)v;
         }
@@ -562,7 +577,10 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
                                const float* __restrict__ sdy,
                                const float* __restrict__ sdyx,
                                __bf16* __restrict__ dx, long rows, int C,
-                               int relu, int token) {
+                               int relu, int token,
+                               __bf16* __restrict__ dres) {
+  // dres (optional, relu path only): the relu-masked dy — the
+  // gradient of a residual fused into this BN's apply
   const float invM = 1.f / rows;
   const long t4 = rows * C / 4;
   // xhat from y — skip the x read.  Only valid without fused relu:
@@ -608,16 +626,18 @@ __global__ void bn_bwd_apply_k(const __bf16* __restrict__ dy,
     f32x4 ga = *(const f32x4*)(gamma + c);
     f32x4 s1 = *(const f32x4*)(sdy + c);
     f32x4 s2 = *(const f32x4*)(sdyx + c);
-    bf16x4 o;
+    bf16x4 o, dr;
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       float g = (float)gv4[j];
       if (relu && (float)yv4[j] <= 0.f) g = 0.f;
+      if (dres) dr[j] = (__bf16)g;
       float xh = ((float)xv4[j] - mv[j]) * iv[j];
       o[j] = (__bf16)(iv[j] * ga[j]
                       * (g - s1[j] * invM - xh * s2[j] * invM));
     }
     *(bf16x4*)(dx + i) = o;
+    if (dres) *(bf16x4*)(dres + i) = dr;
   }
 }
 
@@ -634,7 +654,8 @@ __global__ void bn_bwd_apply_k16(const __bf16* __restrict__ dy,
                                  const float* __restrict__ sdy,
                                  const float* __restrict__ sdyx,
                                  __bf16* __restrict__ dx, long rows,
-                                 int C, int relu, int token) {
+                                 int C, int relu, int token,
+                                 __bf16* __restrict__ dres) {
   const float invM = 1.f / rows;
   const long t8 = rows * C / 8;
   const bool useY = (*allok != token) && !relu;
@@ -675,7 +696,7 @@ __global__ void bn_bwd_apply_k16(const __bf16* __restrict__ dy,
     bf16x8 xv8 = *(const bf16x8*)(x + i);
     bf16x8 yv8;
     if (relu) yv8 = *(const bf16x8*)(y + i);
-    bf16x8 o;
+    bf16x8 o, dr;
 #pragma unroll
     for (int q = 0; q < 2; ++q) {
       f32x4 mv = *(const f32x4*)(mean + c + q * 4);
@@ -687,6 +708,7 @@ __global__ void bn_bwd_apply_k16(const __bf16* __restrict__ dy,
       for (int j = 0; j < 4; ++j) {
         float g = (float)gv8[q * 4 + j];
         if (relu && (float)yv8[q * 4 + j] <= 0.f) g = 0.f;
+        if (dres) dr[q * 4 + j] = (__bf16)g;
         float xh = ((float)xv8[q * 4 + j] - mv[j]) * iv[j];
         o[q * 4 + j] = (__bf16)(iv[j] * ga[j]
                                 * (g - s1[j] * invM
@@ -694,6 +716,7 @@ __global__ void bn_bwd_apply_k16(const __bf16* __restrict__ dy,
       }
     }
     *(bf16x8*)(dx + i) = o;
+    if (dres) *(bf16x8*)(dres + i) = dr;
   }
 }
 
@@ -757,7 +780,8 @@ hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
                                const float* gamma, const float* beta,
                                float* prestats, int nbuckets,
                                long rows, int C, float momentum,
-                               float eps, int relu, hipStream_t stream) {
+                               float eps, int relu, const void* res,
+                               hipStream_t stream) {
   if (prestats) {
     hipLaunchKernelGGL(bn_bucket_reduce_k, dim3(cdiv(C, 256)), dim3(256),
                        0, stream, prestats, s1, s2, C, nbuckets);
@@ -788,12 +812,12 @@ hipError_t launch_bn_fwd_train(const void* x, void* y, float* s1,
     hipLaunchKernelGGL(bn_apply_k16, dim3(ngrid(rows * C / 16)),
                        dim3(256), 0, stream, (const __bf16*)x,
                        (__bf16*)y, mean, invstd, gamma, beta, rows, C,
-                       relu);
+                       relu, (const __bf16*)res);
     return hipGetLastError();
   }
   hipLaunchKernelGGL(bn_apply_k, dim3(ngrid(rows * C / 4)), dim3(256), 0,
                      stream, (const __bf16*)x, (__bf16*)y, mean, invstd,
-                     gamma, beta, rows, C, relu);
+                     gamma, beta, rows, C, relu, (const __bf16*)res);
   return hipGetLastError();
 }
 
@@ -801,12 +825,14 @@ hipError_t launch_bn_fwd_eval(const void* x, void* y, float* mean,
                                float* invstd, const float* rmean,
                                const float* rvar, const float* gamma,
                                const float* beta, long rows, int C,
-                               float eps, int relu, hipStream_t stream) {
+                               float eps, int relu, const void* res,
+                               hipStream_t stream) {
   hipLaunchKernelGGL(bn_eval_stats_k, dim3(cdiv(C, 256)), dim3(256), 0,
                      stream, rmean, rvar, mean, invstd, C, eps);
   hipLaunchKernelGGL(bn_apply_k, dim3(ngrid(rows * C)), dim3(256), 0,
                      stream, (const __bf16*)x, (__bf16*)y, mean, invstd,
-                     gamma, beta, rows, C, relu);
+                     gamma, beta, rows, C, relu,
+                     (const __bf16*)res);
   return hipGetLastError();
 }
 
@@ -815,7 +841,7 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
                          const float* gamma, const float* beta,
                          float* rgamma, int* allok, float* sdy,
                          float* sdyx, void* dx, long rows, int C,
-                         int relu, int token, int forceX,
+                         int relu, int token, int forceX, void* dres,
                          hipStream_t stream) {
   hipLaunchKernelGGL(bn_gamma_prep_k, dim3(cdiv(C, 256)), dim3(256), 0,
                      stream, gamma, rgamma, allok, sdy, sdyx, C, token,
@@ -861,13 +887,15 @@ hipError_t launch_bn_bwd(const void* dy, const void* x, const void* y,
                        dim3(256), 0, stream, (const __bf16*)dy,
                        (const __bf16*)x, (const __bf16*)y, mean, invstd,
                        gamma, beta, rgamma, allok, sdy, sdyx,
-                       (__bf16*)dx, rows, C, relu, token);
+                       (__bf16*)dx, rows, C, relu, token,
+                       (__bf16*)dres);
   } else {
     hipLaunchKernelGGL(bn_bwd_apply_k, dim3(ngrid(rows * C / 4)),
                        dim3(256), 0, stream, (const __bf16*)dy,
                        (const __bf16*)x, (const __bf16*)y, mean, invstd,
                        gamma, beta, rgamma, allok, sdy, sdyx,
-                       (__bf16*)dx, rows, C, relu, token);
+                       (__bf16*)dx, rows, C, relu, token,
+                       (__bf16*)dres);
   }
   return hipGetLastError();
 }
