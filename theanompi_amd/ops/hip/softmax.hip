@@ -280,30 +280,50 @@ struct SgdSeg {
 __global__ void sgd_multi_k(const SgdSeg* __restrict__ segs, float mu,
                             int phase) {
   const SgdSeg s = segs[blockIdx.y];
+  const long stride = (long)gridDim.x * blockDim.x;
+  // 2-deep unroll: six independent 16-B loads in flight per thread
+  // (the 1-deep version measured ~65% of the bandwidth roofline)
   for (long i4 = blockIdx.x * (long)blockDim.x + threadIdx.x; i4 < s.n4;
-       i4 += (long)gridDim.x * blockDim.x) {
+       i4 += 2 * stride) {
+    const long k4 = i4 + stride;
+    const bool two = k4 < s.n4;
     long i = i4 * 4;
+    long k = two ? k4 * 4 : i;
     f32x4 wv = *(const f32x4*)(s.w + i);
     f32x4 gv = *(const f32x4*)(s.g + i);
     f32x4 vv = *(const f32x4*)(s.v + i);
+    f32x4 wv2 = *(const f32x4*)(s.w + k);
+    f32x4 gv2 = *(const f32x4*)(s.g + k);
+    f32x4 vv2 = *(const f32x4*)(s.v + k);
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       if (phase == 2) {
         wv[j] += gv[j];
+        wv2[j] += gv2[j];
       } else {
         float gr = gv[j] + s.wd * wv[j];
         vv[j] = mu * vv[j] - s.lr * gr;
-        if (phase == 0) wv[j] += vv[j];
+        float gr2 = gv2[j] + s.wd * wv2[j];
+        vv2[j] = mu * vv2[j] - s.lr * gr2;
+        if (phase == 0) { wv[j] += vv[j]; wv2[j] += vv2[j]; }
       }
     }
-    if (phase != 2) *(f32x4*)(s.v + i) = vv;
+    if (phase != 2) {
+      *(f32x4*)(s.v + i) = vv;
+      if (two) *(f32x4*)(s.v + k) = vv2;
+    }
     if (phase != 1) {
       *(f32x4*)(s.w + i) = wv;
+      if (two) *(f32x4*)(s.w + k) = wv2;
       if (s.out) {
-        bf16x4 o;
+        bf16x4 o, o2;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) o[j] = (__bf16)wv[j];
+        for (int j = 0; j < 4; ++j) {
+          o[j] = (__bf16)wv[j];
+          o2[j] = (__bf16)wv2[j];
+        }
         *(bf16x4*)(s.out + i) = o;
+        if (two) *(bf16x4*)(s.out + k) = o2;
       }
     }
   }
