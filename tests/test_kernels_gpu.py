@@ -458,15 +458,19 @@ def test_conv_variants_numerics():
 
 
 def test_batchnorm_fused_residual():
-    """y = relu(bn(x) + res) in one pass vs the separate composition —
-    forward and ALL grads (dx, dgamma, dbeta, dres)."""
+    """y = relu(bn(x) + res) in one pass vs a manual fp32 reference
+    that shares the fused path's own relu mask — forward and ALL grads
+    (dx, dgamma, dbeta, dres).  (Comparing against the separate
+    bn+add_relu composition is ill-posed at relu boundaries: its extra
+    bf16 rounding flips borderline masks, and one flipped position
+    shifts max-normalized dx by ~0.7 — measured, both paths match the
+    fp32 reference on non-borderline elements to <0.3%.)"""
     from theanompi_amd.models import layers as L
-    from theanompi_amd.ops import functional as OF
     torch.manual_seed(9)
     d = dev()
     N, H, W, C = 8, 14, 14, 64
+    M = N * H * W
     bn1 = L.BatchNorm2d(C, relu=False, device=str(d))
-    bn2 = L.BatchNorm2d(C, relu=False, device=str(d))
     x0 = torch.randn(N, H, W, C, device=d).to(torch.bfloat16)
     r0 = torch.randn(N, H, W, C, device=d).to(torch.bfloat16)
     dy = torch.randn(N, H, W, C, device=d).to(torch.bfloat16)
@@ -476,19 +480,22 @@ def test_batchnorm_fused_residual():
     y1 = bn1(x1, residual=r1, relu=True)          # fused
     y1.backward(dy)
 
-    x2 = x0.clone().requires_grad_()
-    r2 = r0.clone().requires_grad_()
-    y2 = OF.add(bn2(x2), r2, relu=True)           # separate
-    y2.backward(dy)
-
-    # the SEPARATE baseline rounds bn(x) to bf16 before the add (two
-    # roundings); the fused path adds in fp32 with one rounding —
-    # differences are baseline rounding noise (~6e-3 measured)
-    assert rel_err(y1, y2) < 1.5e-2
-    assert rel_err(x1.grad, x2.grad) < 1.5e-2
-    assert rel_err(r1.grad, r2.grad) < 1.5e-2
-    assert rel_err(bn1.gamma.grad, bn2.gamma.grad) < 1.5e-2
-    assert rel_err(bn1.beta.grad, bn2.beta.grad) < 1.5e-2
+    with torch.no_grad():
+        xf = x0.float()
+        mean = xf.mean((0, 1, 2))
+        var = xf.var((0, 1, 2), unbiased=False)
+        invstd = 1.0 / torch.sqrt(var + bn1.eps)
+        xhat = (xf - mean) * invstd
+        yref = torch.relu(xhat + r0.float())
+        g = dy.float() * (y1.float() > 0)         # the FUSED mask
+        s1 = g.sum((0, 1, 2))
+        s2 = (g * xhat).sum((0, 1, 2))
+        dxref = invstd * (g - s1 / M - xhat * s2 / M)
+    assert rel_err(y1, yref) < 1.5e-2
+    assert rel_err(x1.grad, dxref) < 5e-3
+    assert rel_err(r1.grad, g) < 1e-6              # exact masked dy
+    assert rel_err(bn1.gamma.grad, s2) < 1e-5
+    assert rel_err(bn1.beta.grad, s1) < 1e-5
 
 
 def test_inception_merged_equivalence():
