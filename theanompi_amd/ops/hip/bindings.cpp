@@ -726,8 +726,26 @@ static void bn_fwd(torch::Tensor x, torch::Tensor gamma, torch::Tensor beta,
     resp = res.value().data_ptr();
   }
   if (training) {
-    auto s1 = torch::zeros_like(save_mean);
-    auto s2 = torch::zeros_like(save_mean);
+    // per-(device,C) cached accumulators, re-zeroed by bn_finalize_k
+    // after each consume — no per-call fill launches
+    static std::mutex bn_mu;
+    static std::unordered_map<long, std::pair<torch::Tensor,
+                                              torch::Tensor>> bn_accs;
+    torch::Tensor s1, s2;
+    {
+      long key = ((long)x.device().index() << 32) | C;
+      std::lock_guard<std::mutex> lk(bn_mu);
+      auto it = bn_accs.find(key);
+      if (it == bn_accs.end()) {
+        it = bn_accs
+                 .emplace(key,
+                          std::make_pair(torch::zeros_like(save_mean),
+                                         torch::zeros_like(save_mean)))
+                 .first;
+      }
+      s1 = it->second.first;
+      s2 = it->second.second;
+    }
     float* pre = nullptr;
     if (prestats.has_value()) {
       auto& st = prestats.value();
