@@ -6,6 +6,7 @@ the duck-type contract also works (the workers only use the contract).
 """
 
 import os
+from contextlib import contextmanager as _contextmanager
 
 import torch
 
@@ -315,11 +316,34 @@ class BaseModel:
                 out.extend(mod.logical_params())
         return out
 
+    @_contextmanager
+    def _vels_as_params(self):
+        """Temporarily swap every rec's master .data with its velocity
+        buffer so the LOGICAL getters/setters address the momenta: the
+        V_<i>_<epoch>.npy files get the same per-parameter shapes and
+        reference order as the W files (SURVEY §5.4 — true SGD-state
+        resume in the reference's logical layout), including through
+        merged layers.  Callers must refresh derived copies after a
+        swap that wrote (load does)."""
+        saved = []
+        try:
+            for r in self.recs:
+                saved.append(r.master.data)
+                r.master.data = r.vel
+            yield
+        finally:
+            for r, d in zip(self.recs, saved):
+                r.master.data = d
+
     def save(self, directory, epoch=None):
         epoch = self.epoch if epoch is None else epoch
-        logical = [g() for (_n, g, _s) in self._logical_param_list()]
+        plist = self._logical_param_list()
+        logical = [g() for (_n, g, _s) in plist]
         helper_funcs.save_weights(logical, directory, epoch)
-        helper_funcs.save_momentums(self.vels, directory, epoch)
+        with self._vels_as_params():
+            vlogical = [g() for (_n, g, _s) in plist]
+        helper_funcs.save_weights(vlogical, directory, epoch,
+                                  prefix="V")
 
     def load(self, directory, epoch=None):
         epoch = self.epoch if epoch is None else epoch
@@ -329,8 +353,15 @@ class BaseModel:
         for i, (_n, _g, setter) in enumerate(plist):
             arr = np.load(os.path.join(directory, f"W_{i}_{epoch}.npy"))
             setter(torch.from_numpy(arr))
-        helper_funcs.load_momentums(self.vels, directory, epoch)
-        self._refresh_layers()
+        # refresh bf16/dgrad copies from the restored weights BEFORE the
+        # vel swap (the setters' post_step inside the swap refreshes
+        # from the velocities; redo it after)
+        with self._vels_as_params():
+            for i, (_n, _g, setter) in enumerate(plist):
+                arr = np.load(
+                    os.path.join(directory, f"V_{i}_{epoch}.npy"))
+                setter(torch.from_numpy(arr))
+        self.refresh_all()
 
     def imagenet_or_synthetic(self, crop, synthetic_factory):
         """Real shard pipeline when the session passed
