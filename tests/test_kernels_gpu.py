@@ -457,6 +457,39 @@ def test_conv_variants_numerics():
         assert rel_err(dw, dw0) < 5e-3, (vw, rel_err(dw, dw0))
 
 
+def test_googlenet_checkpoint_roundtrip(tmp_path):
+    """GPU GoogLeNet (merged inception) save -> clobber -> load
+    restores every logical weight AND velocity bit-exactly — the
+    reference checkpoint layout survives the merged layers."""
+    from theanompi_amd.models.googlenet import GoogLeNet
+    torch.manual_seed(2)
+    m = GoogLeNet({"device": "cuda:0", "batch_size": 4, "n_iters": 1,
+                   "val_iters": 1})
+    with torch.no_grad():
+        for v in m.vels:
+            v.normal_()                       # nonzero SGD state
+    plist = m._logical_param_list()
+    before = [g().detach().float().cpu().clone() for (_n, g, _s) in
+              plist]
+    with m._vels_as_params():
+        vbefore = [g().detach().float().cpu().clone()
+                   for (_n, g, _s) in plist]
+    m.save(str(tmp_path), 1)
+    with torch.no_grad():
+        for p in m.params:
+            p.normal_()
+        for v in m.vels:
+            v.zero_()
+    m.load(str(tmp_path), 1)
+    after = [g().detach().float().cpu() for (_n, g, _s) in plist]
+    with m._vels_as_params():
+        vafter = [g().detach().float().cpu() for (_n, g, _s) in plist]
+    for i, (a, b) in enumerate(zip(before, after)):
+        assert torch.equal(a, b), f"weight {i}"
+    for i, (a, b) in enumerate(zip(vbefore, vafter)):
+        assert torch.equal(a, b), f"vel {i}"
+
+
 def test_batchnorm_fused_residual():
     """y = relu(bn(x) + res) in one pass vs a manual fp32 reference
     that shares the fused path's own relu mask — forward and ALL grads
