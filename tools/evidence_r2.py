@@ -31,7 +31,7 @@ train(AlexNet, {"device": "cuda:0", "batch_size": 128, "n_iters": 16},
 # ImageNet shard pipeline (sub-batch semantics) on hardware
 from theanompi_amd.models.data.imagenet import (ImageNetData,
                                                 write_synthetic_shards)
-paths = write_synthetic_shards("gpurun_out/shards", n_shards=6,
+paths = write_synthetic_shards("/tmp/shards_r2", n_shards=6,
                                images_per_shard=256, hw=256)
 d = ImageNetData(paths, paths[:1], batch_size=128, crop=227,
                  n_classes=1000, device="cuda:0")
