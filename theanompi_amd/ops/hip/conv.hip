@@ -754,7 +754,10 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
     if (kout < a.Kgp) atomicAdd(&a.db[g * a.Kgp + kout], dbAcc);
   }
 
-  // epilogue: atomic accumulate fp32
+  // epilogue: fp32 accumulate.  With splitK == 1 each dW element has
+  // exactly ONE writer block, so plain stores replace the atomics —
+  // the FC-shape wgrads (151 MB of dW at M=128) were ATOMICS-bound.
+  const bool plain = (a.splitK == 1);
 #pragma unroll
   for (int i = 0; i < MW; ++i) {
 #pragma unroll
@@ -767,8 +770,11 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
         int fd = n0 + j * 16 + (lane & 15);
         if (fd0 + fd >= a.Kdf) continue;
         // fold the pad-slot mask in here (saves a full dw pass)
-        atomicAdd(&a.DW[base + fd],
-                  acc[i][j][reg] * a.kdfMask[fd0 + fd]);
+        float v = acc[i][j][reg] * a.kdfMask[fd0 + fd];
+        if (plain)
+          a.DW[base + fd] = v;
+        else
+          atomicAdd(&a.DW[base + fd], v);
       }
     }
   }
