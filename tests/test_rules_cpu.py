@@ -344,7 +344,8 @@ def test_resume_bit_equal(tmp_path):
         np.testing.assert_array_equal(a[k], b[k])
 
 
-@pytest.mark.parametrize("world,port", [(2, 29771), (4, 29773)])
+@pytest.mark.parametrize("world,port", [(2, 29771), (4, 29773),
+                                        (8, 29775)])
 def test_bench_multirank_cpu(tmp_path, world, port):
     """The bench contract under torch.distributed.run at world sizes 2
     and 4 (the driver's multi-GPU invocation shape), CPU/gloo — exactly
