@@ -50,6 +50,7 @@ struct WgradArgs {
   int splitK;
   long xTotal;
   float* db;          // optional fused bias grad
+  float* WS;          // split-K plain-store workspace (or nullptr)
 };
 
 extern "C" {
@@ -486,8 +487,31 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
   if (splitK < 1) splitK = 1;
   if (splitK > maxSplit) splitK = maxSplit;
   a.splitK = (int)splitK;
+  a.WS = nullptr;
+  torch::Tensor ws;
+  if (a.splitK > 1) {
+    // plain-store workspace + segsum reduce instead of dW atomics
+    // (up to ~455 split-K writers serialized per cache line on the
+    // VGG shapes); fully overwritten every call, so reused unzeroed
+    static std::mutex ws_mu;
+    static std::unordered_map<int, torch::Tensor> ws_cache;
+    long need = (long)a.splitK * G * Kgp * Kdf;
+    {
+      std::lock_guard<std::mutex> lk(ws_mu);
+      auto& slot = ws_cache[(int)dw.device().index()];
+      if (!slot.defined() || slot.numel() < need)
+        slot = torch::empty({need}, dw.options());
+      ws = slot;
+    }
+    a.WS = ws.data_ptr<float>();
+  }
   CHECK_HIP(launch_conv_wgrad(&a, (int)G, detect_layout(), cur_stream(),
                               (int)variant));
+  if (a.WS) {
+    CHECK_HIP(launch_segsum(a.WS, dw.data_ptr<float>(),
+                            (long)G * Kgp * Kdf, a.splitK, 0,
+                            cur_stream()));
+  }
 }
 
 static void bias_wgrad(torch::Tensor dy, torch::Tensor db) {

@@ -586,6 +586,11 @@ struct WgradArgs {
   // blocks re-read their staged dy tiles from LDS (the separate
   // bias_wgrad kernel was ~51 launches/step on GoogLeNet)
   float* __restrict__ db;
+  // split-K workspace [splitK][G*Kgp*Kdf]: slices write their partial
+  // tiles PLAIN (each block owns a disjoint region) and a segsum pass
+  // reduces into DW — split-K atomics serialized up to ~455 writers
+  // per dW cache line on the VGG shapes.  nullptr = atomic fallback.
+  float* __restrict__ WS;
 };
 
 template <int LAYOUT, int BNF, int BMK = 64, int DEPTH = 2,
@@ -754,10 +759,14 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
     if (kout < a.Kgp) atomicAdd(&a.db[g * a.Kgp + kout], dbAcc);
   }
 
-  // epilogue: fp32 accumulate.  With splitK == 1 each dW element has
-  // exactly ONE writer block, so plain stores replace the atomics —
-  // the FC-shape wgrads (151 MB of dW at M=128) were ATOMICS-bound.
+  // epilogue: fp32 accumulate.  splitK == 1: ONE writer per dW element
+  // -> plain stores (the FC wgrads were atomics-bound).  splitK > 1
+  // with WS: each slice stores plain into its own workspace region and
+  // segsum reduces afterwards.  Otherwise: atomics.
   const bool plain = (a.splitK == 1);
+  float* out = a.DW;
+  if (!plain && a.WS)
+    out = a.WS + (long)slice * a.Kgp * a.Kdf * gridDim.z / a.splitK;
 #pragma unroll
   for (int i = 0; i < MW; ++i) {
 #pragma unroll
@@ -771,8 +780,8 @@ __global__ __launch_bounds__(256) void conv_wgrad_k(WgradArgs a) {
         if (fd0 + fd >= a.Kdf) continue;
         // fold the pad-slot mask in here (saves a full dw pass)
         float v = acc[i][j][reg] * a.kdfMask[fd0 + fd];
-        if (plain)
-          a.DW[base + fd] = v;
+        if (plain || a.WS)
+          out[base + fd] = v;
         else
           atomicAdd(&a.DW[base + fd], v);
       }
