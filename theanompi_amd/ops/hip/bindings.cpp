@@ -489,7 +489,11 @@ static void conv_wgrad(torch::Tensor xp, torch::Tensor dy,
   a.splitK = (int)splitK;
   a.WS = nullptr;
   torch::Tensor ws;
-  if (a.splitK > 1) {
+  static const bool force_atomic = [] {
+    const char* e = getenv("THEANOMPI_WGRAD_ATOMIC");
+    return e && e[0] == '1';
+  }();
+  if (a.splitK > 1 && !force_atomic) {
     // plain-store workspace + segsum reduce instead of dW atomics
     // (up to ~455 split-K writers serialized per cache line on the
     // VGG shapes); fully overwritten every call, so reused unzeroed
