@@ -201,6 +201,75 @@ __global__ void avgpool_fwd_k(const __bf16* __restrict__ xp,
   }
 }
 
+
+// 8-channel avgpool variants (16-B transactions; C % 8 == 0)
+__global__ void avgpool_fwd_k8(const __bf16* __restrict__ xp,
+                               __bf16* __restrict__ y, int N, int Hp,
+                               int Wp, int C, int P, int Q, int kh,
+                               int kw, int sh, int sw, long total8) {
+  const float inv = 1.f / (kh * kw);
+  for (long i8 = blockIdx.x * blockDim.x + threadIdx.x; i8 < total8;
+       i8 += (long)gridDim.x * blockDim.x) {
+    long i = i8 * 8;
+    int c = (int)(i % C);
+    long pix = i / C;
+    int q = (int)(pix % Q);
+    long tmp = pix / Q;
+    int p = (int)(tmp % P);
+    int n = (int)(tmp / P);
+    const __bf16* base = xp
+        + (((long)n * Hp + p * sh) * Wp + q * sw) * C + c;
+    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    for (int r = 0; r < kh; ++r)
+      for (int s = 0; s < kw; ++s) {
+        bf16x8 v = *(const bf16x8*)(base + ((long)r * Wp + s) * C);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += (float)v[j];
+      }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = (__bf16)(acc[j] * inv);
+    *(bf16x8*)(y + i) = o;
+  }
+}
+
+__global__ void avgpool_bwd_k8(const __bf16* __restrict__ dy,
+                               __bf16* __restrict__ dx, int N, int H,
+                               int W, int C, int P, int Q, int kh,
+                               int kw, int sh, int sw, int ph, int pw,
+                               long total8) {
+  const float inv = 1.f / (kh * kw);
+  for (long i8 = blockIdx.x * blockDim.x + threadIdx.x; i8 < total8;
+       i8 += (long)gridDim.x * blockDim.x) {
+    long i = i8 * 8;
+    int c = (int)(i % C);
+    long pix = i / C;
+    int w = (int)(pix % W);
+    long tmp = pix / W;
+    int h = (int)(tmp % H);
+    int n = (int)(tmp / H);
+    int hp = h + ph, wp = w + pw;
+    float acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
+    int pLo = (hp - kh + sh) / sh; if (pLo < 0) pLo = 0;
+    int pHi = hp / sh; if (pHi > P - 1) pHi = P - 1;
+    int qLo = (wp - kw + sw) / sw; if (qLo < 0) qLo = 0;
+    int qHi = wp / sw; if (qHi > Q - 1) qHi = Q - 1;
+    for (int p = pLo; p <= pHi; ++p)
+      for (int q = qLo; q <= qHi; ++q) {
+        int r = hp - p * sh, s = wp - q * sw;
+        if (r < 0 || r >= kh || s < 0 || s >= kw) continue;
+        bf16x8 g = *(const bf16x8*)(dy
+            + (((long)n * P + p) * Q + q) * C + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[j] += (float)g[j];
+      }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) o[j] = (__bf16)(acc[j] * inv);
+    *(bf16x8*)(dx + i) = o;
+  }
+}
+
 __global__ void avgpool_bwd_k(const __bf16* __restrict__ dy,
                               __bf16* __restrict__ dx, int N, int H,
                               int W, int C, int P, int Q, int kh, int kw,
@@ -280,6 +349,14 @@ hipError_t launch_maxpool_bwd(const void* dy, const unsigned char* am,
 hipError_t launch_avgpool_fwd(const void* xp, void* y, int N, int Hp,
                               int Wp, int C, int P, int Q, int kh, int kw,
                               int sh, int sw, hipStream_t stream) {
+  if (C % 8 == 0) {
+    long total8 = (long)N * P * Q * C / 8;
+    hipLaunchKernelGGL(avgpool_fwd_k8, dim3(pool_grid(total8)),
+                       dim3(256), 0, stream, (const __bf16*)xp,
+                       (__bf16*)y, N, Hp, Wp, C, P, Q, kh, kw, sh, sw,
+                       total8);
+    return hipGetLastError();
+  }
   long total4 = (long)N * P * Q * C / 4;
   hipLaunchKernelGGL(avgpool_fwd_k, dim3(pool_grid(total4)), dim3(256), 0,
                      stream, (const __bf16*)xp, (__bf16*)y, N, Hp, Wp, C,
@@ -291,6 +368,14 @@ hipError_t launch_avgpool_bwd(const void* dy, void* dx, int N, int H,
                               int W, int C, int P, int Q, int kh, int kw,
                               int sh, int sw, int ph, int pw,
                               hipStream_t stream) {
+  if (C % 8 == 0) {
+    long total8 = (long)N * H * W * C / 8;
+    hipLaunchKernelGGL(avgpool_bwd_k8, dim3(pool_grid(total8)),
+                       dim3(256), 0, stream, (const __bf16*)dy,
+                       (__bf16*)dx, N, H, W, C, P, Q, kh, kw, sh, sw,
+                       ph, pw, total8);
+    return hipGetLastError();
+  }
   long total = (long)N * H * W * C;
   hipLaunchKernelGGL(avgpool_bwd_k, dim3(pool_grid(total)), dim3(256), 0,
                      stream, (const __bf16*)dy, (__bf16*)dx, N, H, W, C,
