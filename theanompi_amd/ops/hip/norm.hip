@@ -141,31 +141,32 @@ __global__ void lrn_fwd16_k(const __bf16* __restrict__ x,
   const int half = size / 2;               // <= 4
   const float a = alpha / size;
   const int C16 = C / 16;
-  const int C4 = C / 4;
+  const int C8 = C / 8;
   const long total = rows * C16;
   for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     int c16 = (int)(i % C16);
     long row = i / C16;
     const __bf16* xr = x + row * C + c16 * 16;
-    float v[24], sq[24];
+    // neighbourhood [c16*16-8, c16*16+24) via four 16-B loads
+    float v[32], sq[32];
 #pragma unroll
-    for (int b = 0; b < 6; ++b) {
-      int cb = c16 * 4 + b - 1;            // absolute 4-channel block
-      if (cb >= 0 && cb < C4) {
-        bf16x4 t = *(const bf16x4*)(xr + (b - 1) * 4);
+    for (int b8 = 0; b8 < 4; ++b8) {
+      int cb = c16 * 2 + b8 - 1;           // absolute 8-channel block
+      if (cb >= 0 && cb < C8) {
+        bf16x8 t = *(const bf16x8*)(xr + (b8 - 1) * 8);
 #pragma unroll
-        for (int j = 0; j < 4; ++j) v[b * 4 + j] = (float)t[j];
+        for (int j = 0; j < 8; ++j) v[b8 * 8 + j] = (float)t[j];
       } else {
 #pragma unroll
-        for (int j = 0; j < 4; ++j) v[b * 4 + j] = 0.f;
+        for (int j = 0; j < 8; ++j) v[b8 * 8 + j] = 0.f;
       }
     }
 #pragma unroll
-    for (int j = 0; j < 24; ++j) sq[j] = v[j] * v[j];
+    for (int j = 0; j < 32; ++j) sq[j] = v[j] * v[j];
     bf16x8 out0, out1, sc0, sc1;
     float s = 0.f;
-    for (int w = -half; w <= half; ++w) s += sq[4 + w];
+    for (int w = -half; w <= half; ++w) s += sq[8 + w];
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       float sc = k + a * s;
@@ -173,10 +174,10 @@ __global__ void lrn_fwd16_k(const __bf16* __restrict__ x,
       float pw = (beta == 0.75f)
                      ? __fsqrt_rn(inv) * __fsqrt_rn(__fsqrt_rn(inv))
                      : __powf(inv, beta);
-      float o = v[4 + j] * pw;
+      float o = v[8 + j] * pw;
       if (j < 8) { sc0[j] = (__bf16)inv; out0[j] = (__bf16)o; }
       else { sc1[j - 8] = (__bf16)inv; out1[j - 8] = (__bf16)o; }
-      if (j < 15) s += sq[5 + j + half] - sq[4 + j - half];
+      if (j < 15) s += sq[9 + j + half] - sq[8 + j - half];
     }
     long o = row * C + c16 * 16;
     *(bf16x8*)(y + o) = out0;
@@ -201,19 +202,20 @@ __global__ void lrn_bwd16_k(const __bf16* __restrict__ dy,
     int c16 = (int)(i % C16);
     long row = i / C16;
     long base = row * C + c16 * 16;
-    float t[24], yv[16], dyv[16], scv[16];
+    // neighbourhood [c16*16-8, +24) via four 16-B loads per tensor
+    float t[32], yv[16], dyv[16], scv[16];
 #pragma unroll
-    for (int b = 0; b < 6; ++b) {
-      int cb = c16 * 4 + b - 1;
-      if (cb >= 0 && cb < C4) {
-        long ba = base + (b - 1) * 4;
-        bf16x4 d = *(const bf16x4*)(dy + ba);
-        bf16x4 yy = *(const bf16x4*)(y + ba);
-        bf16x4 sc = *(const bf16x4*)(scale + ba);     // = 1/scale
+    for (int b8 = 0; b8 < 4; ++b8) {
+      int cb = c16 * 2 + b8 - 1;           // absolute 8-channel block
+      if (cb >= 0 && cb < C / 8) {
+        long ba = base + (b8 - 1) * 8;
+        bf16x8 d = *(const bf16x8*)(dy + ba);
+        bf16x8 yy = *(const bf16x8*)(y + ba);
+        bf16x8 sc = *(const bf16x8*)(scale + ba);     // = 1/scale
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          t[b * 4 + j] = (float)d[j] * (float)yy[j] * (float)sc[j];
-          int cj = b * 4 + j - 4;
+        for (int j = 0; j < 8; ++j) {
+          t[b8 * 8 + j] = (float)d[j] * (float)yy[j] * (float)sc[j];
+          int cj = b8 * 8 + j - 8;
           if (cj >= 0 && cj < 16) {
             dyv[cj] = (float)d[j];
             yv[cj] = (float)yy[j];
@@ -222,12 +224,12 @@ __global__ void lrn_bwd16_k(const __bf16* __restrict__ dy,
         }
       } else {
 #pragma unroll
-        for (int j = 0; j < 4; ++j) t[b * 4 + j] = 0.f;
+        for (int j = 0; j < 8; ++j) t[b8 * 8 + j] = 0.f;
       }
     }
     bf16x8 out0, out1;
     float s = 0.f;
-    for (int w = -half; w <= half; ++w) s += t[4 + w];
+    for (int w = -half; w <= half; ++w) s += t[8 + w];
 #pragma unroll
     for (int j = 0; j < 16; ++j) {
       float inv = scv[j];                  // 1/scale
@@ -238,7 +240,7 @@ __global__ void lrn_bwd16_k(const __bf16* __restrict__ dy,
       float d = dyv[j] * pw - coef * xv * s;
       if (j < 8) out0[j] = (__bf16)d;
       else out1[j - 8] = (__bf16)d;
-      if (j < 15) s += t[5 + j + half] - t[4 + j - half];
+      if (j < 15) s += t[9 + j + half] - t[8 + j - half];
     }
     *(bf16x8*)(dx + base) = out0;
     *(bf16x8*)(dx + base + 8) = out1;
