@@ -380,10 +380,6 @@ class GOSGD_Exchanger:
         dist.recv(buf, src=src, group=self.group, tag=tag)
         self.try_merge_buf(buf)
 
-    def recv_buffer(self):
-        """Allocate one host recv buffer (one per peer in the worker)."""
-        return torch.empty_like(self._host_in)
-
     def try_merge_buf(self, buf):
         """Merge a fully-received push buffer (CPU/gloo wire format)."""
         a_j = float(buf[0])
