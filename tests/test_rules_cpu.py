@@ -56,18 +56,22 @@ def test_bsp_two_workers_cpu(tmp_path):
         np.testing.assert_allclose(a[k], b[k], atol=1e-6)
 
 
-@pytest.mark.parametrize("strategy", ["ar", "asa", "asa_xgmi"])
-def test_bsp_strategies_cpu(tmp_path, strategy):
+@pytest.mark.parametrize("strategy,n", [("ar", 2), ("asa", 2),
+                                        ("asa_xgmi", 2),
+                                        ("asa_xgmi", 3)])
+def test_bsp_strategies_cpu(tmp_path, strategy, n):
+    # the world-3 asa_xgmi case exercises the non-divisible bucket
+    # padding (pad to 4*world) on the explicit all-to-all path
     body = f"""
     import os
     os.environ.setdefault("GLOO_SOCKET_IFNAME", "lo")
     from theanompi_amd import BSP
     rule = BSP()
-    rule.init(devices=['cpu', 'cpu'],
+    rule.init(devices=['cpu'] * {n},
               modelfile='theanompi_amd.models.test_model',
               modelclass='TestModel',
               epochs=1, sync_type='grad', exchange_strategy={strategy!r})
-    assert rule.wait() == [0, 0]
+    assert rule.wait() == [0] * {n}
     """
     _run_session(tmp_path, body)
 
