@@ -314,6 +314,17 @@ def test_lrn():
     y.backward(dy.to(dev()).to(torch.bfloat16))
     yr.backward(dy)
     assert rel_err(xg.grad, xr.grad) < 3e-2
+    # C % 16 != 0 exercises the 4-wide fallback kernels
+    x8 = q(torch.randn(2, 6, 6, 8))
+    xg8 = x8.to(dev()).to(torch.bfloat16).requires_grad_()
+    y8 = OF.lrn(xg8, 5, 1e-4, 0.75, 1.0)
+    xr8 = x8.clone().requires_grad_()
+    yr8 = OF.lrn_ref(xr8, 5, 1e-4, 0.75, 1.0)
+    assert rel_err(y8, yr8) < 2e-2
+    dy8 = q(torch.randn_like(yr8))
+    y8.backward(dy8.to(dev()).to(torch.bfloat16))
+    yr8.backward(dy8)
+    assert rel_err(xg8.grad, xr8.grad) < 3e-2
 
 
 def test_batchnorm_train_and_eval():
