@@ -459,7 +459,7 @@ def test_conv_variants_numerics():
                 x.grad.float().clone())
 
     y0, dw0, dx0 = run(0, 0)
-    for vf in (1, 2, 3, 5, 6, 7, 8, 9, 10):
+    for vf in (1, 2, 3, 5, 6, 7, 8, 9, 10, 11, 12, 13):
         y, _dw, dx = run(vf, 0)
         assert rel_err(y, y0) < 5e-3, (vf, rel_err(y, y0))
         assert rel_err(dx, dx0) < 5e-3, (vf, rel_err(dx, dx0))

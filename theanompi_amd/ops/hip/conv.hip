@@ -51,7 +51,8 @@ struct GatherGemmArgs {
 
 constexpr int STATS_NB = 32;      // stat atomic buckets (contention)
 
-template <int BM, int BN, int LAYOUT, int DEPTH, int BK = 64>
+template <int BM, int BN, int LAYOUT, int DEPTH, int BK = 64,
+          int NFAST = 0>
 __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
   constexpr int WAVES = (BM == 256) ? 8 : 4;     // block = WAVES*64 thr
   constexpr int WAVES_N = (BN == 128) ? 2 : 1;
@@ -79,10 +80,22 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
-  // XCD-aware m-tile remap: give each XCD a contiguous chunk of m-tiles
-  // so neighbouring tiles (overlapping conv receptive fields) share L2.
-  int mTile = blockIdx.x;
-  {
+  // XCD-aware tile remap.  Default: give each XCD a contiguous chunk
+  // of m-tiles so neighbouring tiles (overlapping conv receptive
+  // fields) share L2.  NFAST (launcher requires gridDim.x % 8 == 0
+  // and > 1 n-tile): an XCD walks ALL n-tiles of one m-tile before
+  // advancing, so a wide-output layer's A tile is read from DRAM once
+  // instead of once per n-tile (ResNet's 256-out 1x1s re-read X 4x).
+  int mTile, nTile;
+  if (NFAST) {
+    long lid = (long)blockIdx.y * gridDim.x + blockIdx.x;
+    const int nT = gridDim.y;
+    const int xcd = (int)(lid & 7);
+    long idx = lid >> 3;
+    nTile = (int)(idx % nT);
+    mTile = xcd * ((int)gridDim.x >> 3) + (int)(idx / nT);
+  } else {
+    mTile = blockIdx.x;
     const int nwg = gridDim.x;
     if (nwg >= 16) {
       const int q = nwg >> 3, r = nwg & 7;
@@ -90,8 +103,8 @@ __global__ __launch_bounds__(BM * 2) void conv_gather_gemm(GatherGemmArgs a) {
       mTile = (xcd < r) ? xcd * (q + 1) + idx
                         : r * (q + 1) + (xcd - r) * q + idx;
     }
+    nTile = blockIdx.y;
   }
-  const int nTile = blockIdx.y;
   const int g = blockIdx.z / a.splitK;
   const int slice = blockIdx.z % a.splitK;
 
@@ -938,7 +951,8 @@ hipError_t launch_conv_gather_gemm(GatherGemmArgs* args, int G,
 
   const int BM = (variant == 3 || variant == 4) ? 256 : 128;
   const int BN = (variant == 2 || variant == 6 || variant == 8
-                  || variant == 9) ? 64 : 128;
+                  || variant == 9 || variant == 12
+                  || variant == 13) ? 64 : 128;
   dim3 grid(cdiv(a.M, BM), cdiv(a.BNtot, BN), G * a.splitK);
   dim3 block(BM * 2);
 
@@ -951,6 +965,23 @@ hipError_t launch_conv_gather_gemm(GatherGemmArgs* args, int G,
     else                                                                   \
       hipLaunchKernelGGL((conv_gather_gemm<BM_, BN_, FRAG_HALVES, D_,      \
                                            BK_>),                          \
+                         grid, block, tabBytes, stream, a);                \
+  } while (0)
+  // NFAST variants (11-13) need gx % 8 == 0 and > 1 n-tile; fall back
+  // to their m-fast twins otherwise
+  if (variant >= 11 && variant <= 13
+      && !(grid.x % 8 == 0 && grid.y > 1)) {
+    variant = (variant == 11) ? 1 : (variant == 12) ? 6 : 2;
+  }
+#define GGN(BM_, BN_, D_, BK_)                                             \
+  do {                                                                     \
+    if (layout == FRAG_CONTIG8)                                            \
+      hipLaunchKernelGGL((conv_gather_gemm<BM_, BN_, FRAG_CONTIG8, D_,     \
+                                           BK_, 1>),                       \
+                         grid, block, tabBytes, stream, a);                \
+    else                                                                   \
+      hipLaunchKernelGGL((conv_gather_gemm<BM_, BN_, FRAG_HALVES, D_,      \
+                                           BK_, 1>),                       \
                          grid, block, tabBytes, stream, a);                \
   } while (0)
   switch (variant) {
@@ -970,9 +1001,13 @@ hipError_t launch_conv_gather_gemm(GatherGemmArgs* args, int G,
     case 8: GG(128, 64, 3, 32); break;
     case 9: GG(128, 64, 3, 64); break;
     case 10: GG(128, 128, 3, 64); break;
+    case 11: GGN(128, 128, 2, 64); break;
+    case 12: GGN(128, 64, 2, 32); break;
+    case 13: GGN(128, 64, 2, 64); break;
     case 1:
     default: GG(128, 128, 2, 64); break;
   }
+#undef GGN
 #undef GG
   return hipGetLastError();
 }

@@ -47,8 +47,8 @@ SHAPES["fc"] = [
     ("fc7", 128, 1, 1, 4096, 4096, 1, 1, 0, 1),
 ]
 
-FWD_VARIANTS = [1, 2, 3, 4, 5, 6, 7, 8, 9, 10]
-DGRAD_VARIANTS = [1, 2, 5, 6, 7, 8, 9, 10]
+FWD_VARIANTS = [1, 2, 3, 4, 5, 6, 7, 8, 9, 10, 11, 12, 13]
+DGRAD_VARIANTS = [1, 2, 5, 6, 7, 8, 9, 10, 11, 12, 13]
 WGRAD_GRID = [(v, t) for v in (1, 2, 3, 4, 5, 6, 7, 8)
               for t in (512, 1024, 2048, 4096)]
 
