@@ -436,7 +436,10 @@ def test_conv_variants_numerics():
     from theanompi_amd.models import layers as L
     torch.manual_seed(11)
     d = dev()
-    B, H, W, Cin, K = 8, 14, 14, 128, 128
+    # M=8192 -> 64 m-tiles (%8==0) and 2 n-tiles at BN=128 so the
+    # NFAST variants (11-13) take their real path instead of falling
+    # back to the m-fast twins
+    B, H, W, Cin, K = 8, 32, 32, 128, 256
     conv = L.Conv2d((H, W, Cin), K, 3, stride=1, pad=1, relu=False,
                     device=str(d), batch=B)
     x = torch.randn(B, H, W, Cin, device=d).to(torch.bfloat16)
