@@ -367,6 +367,7 @@ class _HipConvFn(torch.autograd.Function):
         layer = ctx.layer
         plan = layer.plan
         xp, y = ctx.saved_tensors
+        layer._dyp_ready = None
         if layer.relu:
             # the relu mask kernel accepts a channel slice of a concat
             # buffer directly (uniform row stride), so the inception
@@ -375,7 +376,23 @@ class _HipConvFn(torch.autograd.Function):
                 dy = dy.contiguous()
             dym = torch.empty(dy.shape, dtype=dy.dtype,
                               device=dy.device)
-            C.relu_bwd(dy, y, dym)
+            single = layer._dgrad_single_plan()
+            if (single is not None and ctx.needs_input_grad[0]
+                    and dy.shape[-1] % 8 == 0):
+                # dual write: dym (wgrad/bias) + the halo-padded copy
+                # dgrad needs, skipping its halo_pad pass+launch
+                ht, hb, hl, hr = single["halo"]
+                N, P, Q, K = dym.shape
+                buf = getattr(layer, "_dyp_buf", None)
+                if buf is None or buf.shape[0] != N:
+                    buf = torch.zeros((N, P + ht + hb, Q + hl + hr, K),
+                                      dtype=torch.bfloat16,
+                                      device=dy.device)
+                    layer._dyp_buf = buf
+                C.relu_bwd_dual(dy, y, dym, buf, ht, hl)
+                layer._dyp_ready = buf
+            else:
+                C.relu_bwd(dy, y, dym)
             dy = dym
         else:
             dy = dy.contiguous()
@@ -547,6 +564,20 @@ class Conv2d(_Base):
         self._stats_valid = False
         return self._stats_buf
 
+    def _dgrad_single_plan(self):
+        """The sole dgrad plan when this layer has exactly one stride
+        class WITH a halo (the pre-padded dual-relu-backward target);
+        None otherwise."""
+        cached = getattr(self, "_dgrad_single_c", -1)
+        if cached != -1:
+            return cached
+        plans = [p2 for p2 in self.plan.dgrad_plans if p2 is not None]
+        single = None
+        if len(plans) == 1 and any(plans[0]["halo"]):
+            single = plans[0]
+        self._dgrad_single_c = single
+        return single
+
     def _dgrad(self, dy):
         C = _ops.ext()
         plan = self.plan
@@ -555,12 +586,17 @@ class Conv2d(_Base):
         alloc = torch.empty if plan.dgrad_full_cover else torch.zeros
         dx = alloc((N, H, W, plan.Cphys), dtype=torch.bfloat16,
                    device=dy.device)
+        pre = getattr(self, "_dyp_ready", None)
+        self._dyp_ready = None
         for plan2, wdg in zip(plan.dgrad_plans, self._wdg):
             if plan2 is None:
                 continue
             ht, hb, hl, hr = plan2["halo"]
-            dyp = C.halo_pad_lrtb(dy, ht, hb, hl, hr, 0.0) \
-                if (ht or hb or hl or hr) else dy
+            if pre is not None:
+                dyp = pre                 # dual relu backward pre-padded
+            else:
+                dyp = C.halo_pad_lrtb(dy, ht, hb, hl, hr, 0.0) \
+                    if (ht or hb or hl or hr) else dy
             i, j = plan2["cls"]
             Qp = plan.Q + hl + hr
             vd = plan2.get("variant")

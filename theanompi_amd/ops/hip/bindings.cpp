@@ -66,6 +66,9 @@ hipError_t launch_relu_bwd_s(const void*, const void*, void*, long, int,
                              long, long, hipStream_t);
 hipError_t launch_relu_bwd(const void*, const void*, void*, long,
                            hipStream_t);
+hipError_t launch_relu_bwd_dual(const void*, const void*, void*, void*,
+                                long, int, int, int, int, int, int,
+                                int, long, long, hipStream_t);
 hipError_t launch_add_relu(const void*, const void*, void*, long, int,
                            hipStream_t);
 hipError_t launch_dropout_fwd(const void*, void*, unsigned char*, long,
@@ -571,6 +574,30 @@ static void relu_bwd(torch::Tensor dy, torch::Tensor y, torch::Tensor dx) {
                               cur_stream()));
 }
 
+// dual-output relu backward: dx contiguous [N,H,W,C] for wgrad/bias,
+// AND the same masked values into the interior of a halo-padded
+// buffer dxp [N,H+2ht',W+2hl',C] (borders stay zero; cached by the
+// layer) — the single-class dgrad then skips its halo_pad launch.
+static void relu_bwd_dual(torch::Tensor dy, torch::Tensor y,
+                          torch::Tensor dx, torch::Tensor dxp, long ht,
+                          long hl) {
+  TORCH_CHECK(dy.scalar_type() == torch::kBFloat16);
+  check_bf16(dx, "dx");
+  check_bf16(dxp, "dxp");
+  long N = dx.size(0), H = dx.size(1), W = dx.size(2), C = dx.size(3);
+  TORCH_CHECK(dxp.size(0) == N && dxp.size(3) == C);
+  long Hp2 = dxp.size(1), Wp2 = dxp.size(2);
+  TORCH_CHECK(C % 8 == 0, "relu_bwd_dual: C % 8");
+  long ldDy = dy.is_contiguous() ? C : row_ld(dy);
+  long ldY = y.is_contiguous() ? C : row_ld(y);
+  TORCH_CHECK(ldDy > 0 && ldY > 0, "relu_bwd_dual: unsupported layout");
+  CHECK_HIP(launch_relu_bwd_dual(dy.data_ptr(), y.data_ptr(),
+                                 dx.data_ptr(), dxp.data_ptr(),
+                                 dx.numel(), (int)C, (int)H, (int)W,
+                                 (int)Hp2, (int)Wp2, (int)ht, (int)hl,
+                                 ldDy, ldY, cur_stream()));
+}
+
 static void add_relu(torch::Tensor a, torch::Tensor b, torch::Tensor y,
                      bool relu) {
   check_bf16(a, "a");
@@ -1068,6 +1095,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_wgrad", &bias_wgrad);
   m.def("relu_fwd", &relu_fwd);
   m.def("relu_bwd", &relu_bwd);
+  m.def("relu_bwd_dual", &relu_bwd_dual);
   m.def("add_relu", &add_relu);
   m.def("dropout_fwd", &dropout_fwd);
   m.def("dropout_bwd", &dropout_bwd);

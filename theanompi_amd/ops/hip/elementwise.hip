@@ -96,6 +96,37 @@ __global__ void relu_bwd_k16(const __bf16* __restrict__ dy,
   }
 }
 
+// relu backward with DUAL output: the contiguous masked dy (consumed
+// by wgrad/bias) and the same values written into the interior of a
+// halo-padded buffer whose borders stay zero (cached per layer) — the
+// single-class dgrad then needs no separate halo_pad launch + pass.
+// dy/y may be channel slices (uniform per-pixel stride ld*).
+__global__ void relu_bwd_dual_k(const __bf16* __restrict__ dy,
+                                const __bf16* __restrict__ y,
+                                __bf16* __restrict__ dx,
+                                __bf16* __restrict__ dxp, long n8,
+                                int C, int H, int W, int Hp2, int Wp2,
+                                int ht, int hl, long ldDy, long ldY) {
+  const __bf16 z = (__bf16)0.f;
+  for (long i8 = blockIdx.x * blockDim.x + threadIdx.x; i8 < n8;
+       i8 += (long)gridDim.x * blockDim.x) {
+    long i = i8 * 8;
+    int c = (int)(i % C);
+    long pix = i / C;
+    int w = (int)(pix % W);
+    long tmp = pix / W;
+    int h = (int)(tmp % H);
+    int n = (int)(tmp / H);
+    bf16x8 g = *(const bf16x8*)(dy + pix * ldDy + c);
+    bf16x8 v = *(const bf16x8*)(y + pix * ldY + c);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) g[j] = v[j] > z ? g[j] : z;
+    *(bf16x8*)(dx + i) = g;
+    long po = ((((long)n * Hp2) + h + ht) * Wp2 + w + hl) * C + c;
+    *(bf16x8*)(dxp + po) = g;
+  }
+}
+
 __global__ void add_relu_k16(const __bf16* __restrict__ p,
                              const __bf16* __restrict__ q,
                              __bf16* __restrict__ y, long n8, int relu) {
@@ -423,6 +454,18 @@ hipError_t launch_relu_bwd(const void* dy, const void* y, void* dx, long n,
   hipLaunchKernelGGL(relu_bwd_k, dim3(ew_grid(n / 4)), dim3(256), 0,
                      stream, (const __bf16*)dy, (const __bf16*)y,
                      (__bf16*)dx, n / 4);
+  return hipGetLastError();
+}
+
+hipError_t launch_relu_bwd_dual(const void* dy, const void* y, void* dx,
+                                void* dxp, long n, int C, int H, int W,
+                                int Hp2, int Wp2, int ht, int hl,
+                                long ldDy, long ldY,
+                                hipStream_t stream) {
+  hipLaunchKernelGGL(relu_bwd_dual_k, dim3(ew_grid(n / 8)), dim3(256),
+                     0, stream, (const __bf16*)dy, (const __bf16*)y,
+                     (__bf16*)dx, (__bf16*)dxp, n / 8, C, H, W, Hp2,
+                     Wp2, ht, hl, ldDy, ldY);
   return hipGetLastError();
 }
 
