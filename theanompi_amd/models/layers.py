@@ -447,6 +447,22 @@ class _Base(nn.Module):
         reference's per-layer W-then-b order."""
         return []
 
+    def _dgrad_single_plan(self):
+        """The sole dgrad plan when the layer has exactly one stride
+        class WITH a halo (the pre-padded dual-relu-backward target);
+        None otherwise.  Lives on _Base: Linear shares the conv
+        backward path."""
+        cached = getattr(self, "_dgrad_single_c", -1)
+        if cached != -1:
+            return cached
+        plans = [p2 for p2 in getattr(self, "plan").dgrad_plans
+                 if p2 is not None]
+        single = None
+        if len(plans) == 1 and any(plans[0]["halo"]):
+            single = plans[0]
+        self._dgrad_single_c = single
+        return single
+
     def post_step(self):
         """Called after every optimizer step (refresh bf16 copies)."""
 
@@ -563,20 +579,6 @@ class Conv2d(_Base):
             return None
         self._stats_valid = False
         return self._stats_buf
-
-    def _dgrad_single_plan(self):
-        """The sole dgrad plan when this layer has exactly one stride
-        class WITH a halo (the pre-padded dual-relu-backward target);
-        None otherwise."""
-        cached = getattr(self, "_dgrad_single_c", -1)
-        if cached != -1:
-            return cached
-        plans = [p2 for p2 in self.plan.dgrad_plans if p2 is not None]
-        single = None
-        if len(plans) == 1 and any(plans[0]["halo"]):
-            single = plans[0]
-        self._dgrad_single_c = single
-        return single
 
     def _dgrad(self, dy):
         C = _ops.ext()
