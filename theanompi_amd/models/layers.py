@@ -16,6 +16,7 @@ reference ops (ops/functional.py *_ref).
 """
 
 import math
+import os
 
 import numpy as np
 import torch
@@ -29,6 +30,10 @@ from theanompi_amd.ops.functional import cpad_for
 # ---------------------------------------------------------------------------
 # gather-table / repack-map builders (host side, run once per layer)
 # ---------------------------------------------------------------------------
+
+# dual-output relu backward (masked dy + pre-padded dgrad halo in one
+# pass) — THEANOMPI_DUAL_RELU=0 disables for A/B measurement
+DUAL_RELU = os.environ.get("THEANOMPI_DUAL_RELU", "1") != "0"
 
 PIECE = 16   # flat-k granularity in elements (32 B of bf16)
 KPAD = 64    # flat reduction dim padded to one GEMM K-tile
@@ -376,7 +381,7 @@ class _HipConvFn(torch.autograd.Function):
                 dy = dy.contiguous()
             dym = torch.empty(dy.shape, dtype=dy.dtype,
                               device=dy.device)
-            single = layer._dgrad_single_plan()
+            single = layer._dgrad_single_plan() if DUAL_RELU else None
             if (single is not None and ctx.needs_input_grad[0]
                     and dy.shape[-1] % 8 == 0):
                 # dual write: dym (wgrad/bias) + the halo-padded copy
