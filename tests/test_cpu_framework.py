@@ -570,3 +570,46 @@ def test_bench_contract_cpu(tmp_path):
     assert d["n_gpus"] == 1 and d["steps"] == 2
     assert d["data"] == "synthetic"
     assert d["config"]["global_batch"] == 8
+
+
+def test_imagenet_augmentation_semantics(tmp_path):
+    """SURVEY §3.5 augmentation contract: train batches are randomly
+    cropped/mirrored per shard visit (different across epochs), the
+    val path is a DETERMINISTIC center crop with no mirror."""
+    import numpy as np
+    from theanompi_amd.models.data.imagenet import (ImageNetData,
+                                                    write_synthetic_shards)
+    paths = write_synthetic_shards(str(tmp_path), n_shards=1,
+                                   images_per_shard=4, hw=40)
+    d = ImageNetData(paths, paths, batch_size=4, crop=32,
+                     n_classes=1000, device="cpu", shuffle=False)
+    try:
+        # val: bit-identical across calls and equal to the manual
+        # center crop of the raw uint8 shard
+        xv1, yv1 = d.get_val_batch(0)
+        xv2, yv2 = d.get_val_batch(0)
+        assert torch.equal(xv1, xv2) and torch.equal(yv1, yv2)
+        raw = np.load(paths[0])
+        o = (40 - 32) // 2
+        want = torch.from_numpy(
+            raw[:, o:o + 32, o:o + 32, :].astype(np.float32)) \
+            * (1 / 128.0) - 1.0
+        assert torch.equal(xv1, want)
+
+        # train: the random crop/mirror must stay inside the source
+        # image (the batch is a translated window of raw pixel values)
+        # and vary across epoch visits with overwhelming probability
+        seen = set()
+        for epoch in range(6):
+            d.reset("train")
+            x, _ = d.get_train_batch(0)
+            assert x.shape == (4, 32, 32, 3)
+            # every served pixel value exists in the raw shard (crop
+            # never reads out of bounds / stale slot memory)
+            lo = raw.min() / 128.0 - 1.0
+            hi = raw.max() / 128.0 - 1.0
+            assert x.min() >= lo - 1e-6 and x.max() <= hi + 1e-6
+            seen.add(x.numpy().tobytes())
+        assert len(seen) > 1, "train crops never varied"
+    finally:
+        d.close()
