@@ -454,3 +454,30 @@ def test_relaunch_on_rendezvous_port_race(tmp_path):
     assert not rule2._maybe_relaunch([17])
     """
     _run_session(tmp_path, body, timeout=360)
+
+
+def test_torchrun_launched_worker(tmp_path):
+    """README multi-node contract: the worker modules bootstrap from
+    torchrun-style env alone (RANK/LOCAL_RANK/MASTER_* + the
+    THEANOMPI_AMD_* config vars), with no Rule launcher in the loop —
+    this is the documented path to multi-node."""
+    import subprocess
+    import sys
+    env = dict(os.environ)
+    env.update({
+        "GLOO_SOCKET_IFNAME": "lo",
+        "PYTHONPATH": REPO + os.pathsep + env.get("PYTHONPATH", ""),
+        "THEANOMPI_AMD_MODELFILE": "theanompi_amd.models.test_model",
+        "THEANOMPI_AMD_MODELCLASS": "TestModel",
+        "THEANOMPI_AMD_OPT_EPOCHS": "1",
+        "THEANOMPI_AMD_OPT_SEED": "3",
+    })
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29613", "-m", "theanompi_amd.bsp_worker"],
+        env=env, cwd=str(tmp_path), capture_output=True, text=True,
+        timeout=240)
+    assert out.returncode == 0, \
+        f"stdout:\n{out.stdout[-3000:]}\nstderr:\n{out.stderr[-3000:]}"
+    assert "val: cost" in out.stdout
