@@ -28,6 +28,7 @@ class _Attr(TorchDispatchMode):
         super().__init__()
         self.hits = collections.Counter()
         self.bytes = collections.Counter()
+        self.anon = []
 
     def __torch_dispatch__(self, func, types, args=(), kwargs=None):
         name = str(func)
@@ -45,6 +46,15 @@ class _Attr(TorchDispatchMode):
                 else None
             if t is not None:
                 self.bytes[key] += t.numel() * t.element_size()
+            if site == "?" and t is not None:
+                src = args[1] if (len(args) > 1
+                                  and isinstance(args[1], torch.Tensor)) \
+                    else None
+                self.anon.append(
+                    (name.split(".Tensor")[0], tuple(t.shape),
+                     tuple(t.stride()), str(t.dtype).replace(
+                         "torch.", ""),
+                     tuple(src.stride()) if src is not None else None))
         return func(*args, **(kwargs or {}))
 
 
@@ -72,6 +82,11 @@ def main():
                                 key=lambda kv: -kv[1])[:30]:
         mb = mode.bytes[(op, site)] / steps / 1e6
         print(f"{n / steps:7.1f}/step  {mb:9.2f} MB/step  {op:14s} {site}")
+    if mode.anon:
+        print("== engine-internal ops (no repo frame), dst shape/stride:")
+        an = collections.Counter(mode.anon)
+        for sig, n in sorted(an.items(), key=lambda kv: -kv[1])[:20]:
+            print(f"{n / steps:7.1f}/step  {sig}")
 
 
 if __name__ == "__main__":
