@@ -784,7 +784,6 @@ class _CopySlice(torch.autograd.Function):
     def forward(ctx, dst_slice, src):
         with torch.no_grad():
             dst_slice.copy_(src)
-        ctx.mark_dirty(dst_slice)
         return dst_slice
 
     @staticmethod
@@ -792,7 +791,21 @@ class _CopySlice(torch.autograd.Function):
         return None, dy
 
 
+def _alias_strided(v):
+    """Non-view tensor over the same storage/strides as view ``v``.
+    mark_dirty on a VIEW rebases the base's graph onto CopySlices,
+    which materializes TWO engine copies per inception module in
+    backward (the b1 slice grad + the whole concat-buffer grad,
+    ~101 MB/step on GoogLeNet — profiles/glue_attribution_r2.txt);
+    a non-view alias keeps autograd's view machinery out entirely."""
+    t = torch.empty(0, dtype=v.dtype, device=v.device)
+    t.set_(v.untyped_storage(), v.storage_offset(), v.shape, v.stride())
+    return t
+
+
 def copy_slice(dst_slice, src):
+    if dst_slice._base is not None:
+        dst_slice = _alias_strided(dst_slice)
     return _CopySlice.apply(dst_slice, src)
 
 
