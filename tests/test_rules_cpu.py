@@ -481,3 +481,40 @@ def test_torchrun_launched_worker(tmp_path):
     assert out.returncode == 0, \
         f"stdout:\n{out.stdout[-3000:]}\nstderr:\n{out.stderr[-3000:]}"
     assert "val: cost" in out.stdout
+
+
+def test_gosgd_merge_invariants():
+    """Blot et al. GoSGD algebra at unit level (SURVEY §2.2 GOSGD row):
+    the receiver's merge w <- (a*w + a_j*w_j)/(a+a_j), a <- a+a_j must
+    (1) be the alpha-weighted mean, (2) conserve total alpha across the
+    pair given the sender's halving, (3) conserve the alpha-weighted
+    parameter sum — the consensus invariant that makes the gossip
+    average converge to the true average."""
+    import torch
+    from theanompi_amd.lib.exchanger import GOSGD_Exchanger
+
+    w_r = torch.nn.Parameter(torch.tensor([2.0, -1.0]))
+    recv = GOSGD_Exchanger({}, [w_r])
+    assert recv.alpha == 1.0
+
+    # sender side (emulated): alpha 1.0 halves to 0.5, wire carries
+    # [alpha_j, w_j...]
+    a_j = 0.5
+    w_j = torch.tensor([4.0, 3.0])
+    buf = torch.empty(3, dtype=torch.float32)
+    buf[0] = a_j
+    buf[1:] = w_j
+
+    # pair state before the push: both ranks alpha=1; sender holds w_j
+    weighted_before = 1.0 * w_r.detach().clone() + 1.0 * w_j
+
+    recv.try_merge_buf(buf)
+
+    want = (1.0 * torch.tensor([2.0, -1.0]) + a_j * w_j) / (1.0 + a_j)
+    assert torch.allclose(w_r.detach(), want)
+    assert recv.alpha == 1.5
+    # alpha conservation: receiver 1.5 + sender's retained 0.5 == 2.0
+    assert recv.alpha + 0.5 == 2.0
+    # alpha-weighted parameter sum over the pair is conserved
+    weighted_after = recv.alpha * w_r.detach() + 0.5 * w_j
+    assert torch.allclose(weighted_after, weighted_before)
